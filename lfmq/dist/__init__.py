@@ -1,0 +1,78 @@
+"""Process-group plumbing: RCCL over xGMI on GPU, gloo on CPU for tests.
+
+One process per GPU, torch.distributed; the "nccl" backend string IS RCCL
+on ROCm.  The same code path runs under gloo with world_size>1 on a CPU
+host (tests/test_tp_gloo.py) — the only difference is the backend string,
+so multi-rank numerics are testable without a GPU cluster.
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+_TP_GROUP: Optional[dist.ProcessGroup] = None
+
+
+def init_dist(backend: Optional[str] = None) -> tuple[int, int]:
+    """Initialise torch.distributed from torchrun env vars.
+
+    Returns (rank, world_size).  No-op single-rank mode when WORLD_SIZE
+    is absent or 1.
+    """
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world <= 1:
+        return 0, 1
+    if not dist.is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group(backend=backend)
+        if backend == "nccl":
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    return dist.get_rank(), dist.get_world_size()
+
+
+def get_rank() -> int:
+    return dist.get_rank() if dist.is_initialized() else 0
+
+
+def get_world_size() -> int:
+    return dist.get_world_size() if dist.is_initialized() else 1
+
+
+def get_tp_group():
+    """The tensor-parallel group.  Single-node: the whole world."""
+    return _TP_GROUP  # None => default group
+
+
+def all_reduce_sum(t: torch.Tensor) -> torch.Tensor:
+    if get_world_size() > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=get_tp_group())
+    return t
+
+
+def all_gather_cat(t: torch.Tensor, dim: int = -1) -> torch.Tensor:
+    world = get_world_size()
+    if world == 1:
+        return t
+    parts = [torch.empty_like(t) for _ in range(world)]
+    dist.all_gather(parts, t.contiguous(), group=get_tp_group())
+    return torch.cat(parts, dim=dim)
+
+
+def all_to_all_single(out: torch.Tensor, inp: torch.Tensor,
+                      out_splits=None, in_splits=None) -> torch.Tensor:
+    if get_world_size() == 1:
+        out.copy_(inp)
+        return out
+    dist.all_to_all_single(out, inp, out_splits, in_splits, group=get_tp_group())
+    return out
+
+
+def barrier() -> None:
+    if get_world_size() > 1:
+        dist.barrier()

@@ -1,0 +1,226 @@
+"""Llama-3 family architecture, built natively for the MI355X decode path.
+
+Random-init from a local config (no network => no pretrained weights,
+BASELINE.md protocol).  MI355X-first choices:
+  - fused qkv_proj and gate_up_proj Linears (one GEMV launch each at
+    decode instead of 3/2 — launch overhead is the TP=8 risk,
+    SURVEY.md §2.5)
+  - static KV cache + pure-tensor decode step, hipGraph-capturable
+  - attention decode / rmsnorm / rope / silu·mul via the gfx950 HIP
+    kernels (lfmq.ops.functional)
+
+TP sharding (applied by lfmq.dist.tp): qkv/gate_up/lm_head column-split,
+o_proj/down_proj row-split with all-reduce.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field, asdict
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from ..ops import functional as F
+from .kvcache import KVCache
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    n_layers: int = 32
+    n_heads: int = 32
+    n_kv_heads: int = 8
+    head_dim: int = 128
+    rms_eps: float = 1e-5
+    rope_theta: float = 500000.0
+    max_seq: int = 8192
+    tie_embeddings: bool = False
+    arch: str = "llama"
+
+    @classmethod
+    def llama3_8b(cls, **kw):
+        return cls(**{**dict(vocab_size=128256, hidden_size=4096,
+                             intermediate_size=14336, n_layers=32, n_heads=32,
+                             n_kv_heads=8), **kw})
+
+    @classmethod
+    def llama3_70b(cls, **kw):
+        return cls(**{**dict(vocab_size=128256, hidden_size=8192,
+                             intermediate_size=28672, n_layers=80, n_heads=64,
+                             n_kv_heads=8), **kw})
+
+    @classmethod
+    def tiny(cls, **kw):
+        """2-layer 8B-shaped-per-layer test model."""
+        return cls(**{**dict(vocab_size=512, hidden_size=256,
+                             intermediate_size=512, n_layers=2, n_heads=4,
+                             n_kv_heads=2, head_dim=64, max_seq=512), **kw})
+
+    def to_dict(self) -> dict:
+        return asdict(self)
+
+    @classmethod
+    def from_dict(cls, d: dict) -> "LlamaConfig":
+        d = {k: v for k, v in d.items() if k in cls.__dataclass_fields__}
+        return cls(**d)
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim: int, eps: float, dtype=torch.bfloat16):
+        super().__init__()
+        self.eps = eps
+        self.weight = nn.Parameter(torch.ones(dim, dtype=dtype))
+
+    def forward(self, x, residual: Optional[torch.Tensor] = None):
+        return F.rmsnorm(x, self.weight, self.eps, residual)
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype):
+        super().__init__()
+        self.cfg = cfg
+        self.n_heads = cfg.n_heads
+        self.n_kv_heads = cfg.n_kv_heads
+        self.head_dim = cfg.head_dim
+        qkv_out = (cfg.n_heads + 2 * cfg.n_kv_heads) * cfg.head_dim
+        self.qkv_proj = nn.Linear(cfg.hidden_size, qkv_out, bias=False, dtype=dtype)
+        self.o_proj = nn.Linear(cfg.n_heads * cfg.head_dim, cfg.hidden_size,
+                                bias=False, dtype=dtype)
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+
+    def forward(self, x: torch.Tensor, cache_k: torch.Tensor,
+                cache_v: torch.Tensor, positions: torch.Tensor,
+                seq_len_after: int) -> torch.Tensor:
+        b, s, _ = x.shape
+        hq, hkv, d = self.n_heads, self.n_kv_heads, self.head_dim
+        qkv = self.qkv_proj(x)
+        q, k, v = qkv.split([hq * d, hkv * d, hkv * d], dim=-1)
+        q = q.reshape(b, s, hq, d)
+        k = k.reshape(b, s, hkv, d)
+        v = v.reshape(b, s, hkv, d)
+        q, k = F.apply_rope(q, k, positions, self.cfg.rope_theta)
+        pos0 = seq_len_after - s
+        F.kv_append(cache_k, cache_v,
+                    k.transpose(1, 2), v.transpose(1, 2), pos0)
+        qh = q.transpose(1, 2)  # [b, hq, s, d]
+        if s == 1:
+            o = F.attn_decode(qh, cache_k, cache_v, seq_len_after, self.scale)
+        elif pos0 == 0:
+            o = F.attn_prefill(qh, k.transpose(1, 2), v.transpose(1, 2),
+                               causal=True, scale=self.scale)
+        else:  # chunked continuation: attend to full cache, causal inside chunk
+            o = self._chunk_attn(qh, cache_k, cache_v, pos0, seq_len_after)
+        o = o.transpose(1, 2).reshape(b, s, hq * d)
+        return self.o_proj(o)
+
+    def _chunk_attn(self, q, cache_k, cache_v, pos0, seq_len_after):
+        b, hq, s, d = q.shape
+        hkv = cache_k.shape[1]
+        rep = hq // hkv
+        kc = cache_k[:, :, :seq_len_after].float().repeat_interleave(rep, dim=1)
+        vc = cache_v[:, :, :seq_len_after].float().repeat_interleave(rep, dim=1)
+        sc = torch.einsum("bhqd,bhtd->bhqt", q.float(), kc) * self.scale
+        t_idx = torch.arange(seq_len_after, device=q.device)
+        q_idx = pos0 + torch.arange(s, device=q.device)
+        mask = t_idx[None, :] > q_idx[:, None]
+        sc = sc.masked_fill(mask, float("-inf"))
+        p = torch.softmax(sc, dim=-1)
+        return torch.einsum("bhqt,bhtd->bhqd", p, vc).to(q.dtype)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype):
+        super().__init__()
+        self.gate_up_proj = nn.Linear(cfg.hidden_size, 2 * cfg.intermediate_size,
+                                      bias=False, dtype=dtype)
+        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size,
+                                   bias=False, dtype=dtype)
+        self.inter = cfg.intermediate_size
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        gu = self.gate_up_proj(x)
+        gate, up = gu.split([self.inter, self.inter], dim=-1)
+        return self.down_proj(F.silu_mul(gate, up))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype):
+        super().__init__()
+        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_eps, dtype)
+        self.self_attn = LlamaAttention(cfg, dtype)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_eps, dtype)
+        self.mlp = LlamaMLP(cfg, dtype)
+
+    def forward(self, x, cache_k, cache_v, positions, seq_len_after):
+        h = self.input_layernorm(x)
+        h = self.self_attn(h, cache_k, cache_v, positions, seq_len_after)
+        x = x + h
+        h = self.post_attention_layernorm(x)
+        x = x + self.mlp(h)
+        return x
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype=torch.bfloat16):
+        super().__init__()
+        self.cfg = cfg
+        self.dtype = dtype
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size, dtype=dtype)
+        self.layers = nn.ModuleList(LlamaDecoderLayer(cfg, dtype)
+                                    for _ in range(cfg.n_layers))
+        self.norm = RMSNorm(cfg.hidden_size, cfg.rms_eps, dtype)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False, dtype=dtype)
+        if cfg.tie_embeddings:
+            self.lm_head.weight = self.embed_tokens.weight
+
+    @torch.no_grad()
+    def init_random(self, seed: int = 0):
+        g = torch.Generator().manual_seed(seed)
+        for p in self.parameters():
+            p.data.copy_(torch.randn(p.shape, generator=g, dtype=torch.float32)
+                         .mul_(0.02).to(p.dtype))
+        return self
+
+    def make_cache(self, batch: int, max_seq: Optional[int] = None,
+                   device=None) -> KVCache:
+        dev = device or next(self.parameters()).device
+        # TP: layers may hold a KV-head shard; size the cache from the module
+        hkv = self.layers[0].self_attn.n_kv_heads
+        return KVCache(self.cfg.n_layers, batch, hkv,
+                       max_seq or self.cfg.max_seq, self.cfg.head_dim, dev,
+                       dtype=self.dtype)
+
+    def forward(self, input_ids: torch.Tensor, cache: KVCache,
+                positions: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """input_ids [b, s] -> logits [b, s, vocab].  Advances the cache."""
+        b, s = input_ids.shape
+        pos0 = cache.seq_len
+        if positions is None:
+            positions = torch.arange(pos0, pos0 + s, device=input_ids.device)
+        seq_after = pos0 + s
+        x = self.embed_tokens(input_ids)
+        for i, layer in enumerate(self.layers):
+            x = layer(x, cache.k[i], cache.v[i], positions, seq_after)
+        x = self.norm(x)
+        logits = self.lm_head(x)
+        cache.advance(s)
+        return logits
+
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
+                 cache: Optional[KVCache] = None) -> torch.Tensor:
+        b, s = input_ids.shape
+        if cache is None:
+            cache = self.make_cache(b, max_seq=s + max_new_tokens + 8)
+        logits = self.forward(input_ids, cache)
+        out = [input_ids]
+        tok = logits[:, -1:].argmax(dim=-1)
+        for _ in range(max_new_tokens - 1):
+            out.append(tok)
+            logits = self.forward(tok, cache)
+            tok = logits[:, -1:].argmax(dim=-1)
+        out.append(tok)
+        return torch.cat(out, dim=1)

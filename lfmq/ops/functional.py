@@ -1,0 +1,91 @@
+"""Device-typed op entry points used by the model runtime.
+
+On a CUDA (ROCm) device every call goes to the hand-written gfx950 HIP
+kernel — a missing extension raises (lfmq.ops.hip_ext), it never falls
+back to eager torch.  On CPU the exact reference twin runs (unit tests +
+the OPT int8 CPU decode config).  This is a device split, not a backend
+dispatch: per device there is exactly one implementation.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from . import hip_ext, ref
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5,
+            residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if x.is_cuda:
+        return hip_ext().rmsnorm(x, weight, eps, residual)
+    return ref.rmsnorm(x, weight, eps, residual)
+
+
+def rmsnorm_residual(x: torch.Tensor, residual: torch.Tensor,
+                     weight: torch.Tensor, eps: float = 1e-5):
+    """Returns (normed, new_residual = x + residual) in one pass."""
+    if x.is_cuda:
+        return hip_ext().rmsnorm_residual(x, residual, weight, eps)
+    new_res = ref.residual_add(x, residual)
+    return ref.rmsnorm(new_res, weight, eps), new_res
+
+
+def apply_rope(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
+               theta: float = 500000.0):
+    """q: [b, s, hq, d], k: [b, s, hkv, d] — rotated in place on GPU."""
+    if q.is_cuda:
+        hip_ext().rope(q, k, positions, theta)
+        return q, k
+    return ref.apply_rope(q, positions, theta), ref.apply_rope(k, positions, theta)
+
+
+def dq_linear(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,
+              qzeros: torch.Tensor, bits: int, group_size: int,
+              bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Fused dequant-GEMM  y = x @ dequant(W)^T  (+bias)."""
+    if x.is_cuda:
+        x2 = x.reshape(-1, x.shape[-1])
+        y = hip_ext().dq_linear(x2, qweight, scales, qzeros, bits, group_size, bias)
+        return y.reshape(*x.shape[:-1], y.shape[-1])
+    return ref.dq_linear(x, qweight, scales, qzeros, bits, group_size, bias).to(x.dtype)
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if gate.is_cuda:
+        return hip_ext().silu_mul(gate, up)
+    return ref.silu_mul(gate, up)
+
+
+def attn_decode(q, k_cache, v_cache, seq_len: int, scale: Optional[float] = None):
+    if q.is_cuda:
+        return hip_ext().attn_decode(
+            q, k_cache, v_cache, seq_len, scale or 1.0 / math.sqrt(q.shape[-1]))
+    return ref.attn_decode(q, k_cache, v_cache, seq_len, scale)
+
+
+def attn_prefill(q, k, v, causal: bool = True, scale: Optional[float] = None):
+    """Prefill attention. GPU: explicit bf16 matmul chain via rocBLAS (the
+    PyTorch-ROCm layer; never SDPA, which can route to aotriton/Triton)."""
+    if q.is_cuda:
+        b, hq, s, d = q.shape
+        hkv = k.shape[1]
+        rep = hq // hkv
+        scale = scale or 1.0 / math.sqrt(d)
+        kf = k.repeat_interleave(rep, dim=1) if rep > 1 else k
+        vf = v.repeat_interleave(rep, dim=1) if rep > 1 else v
+        sc = torch.matmul(q, kf.transpose(-1, -2)).float() * scale
+        if causal:
+            mask = torch.full((s, s), float("-inf"), device=q.device).triu(1)
+            sc = sc + mask
+        p = torch.softmax(sc, dim=-1).to(q.dtype)
+        return torch.matmul(p, vf)
+    return ref.attn_prefill(q, k, v, causal, scale)
+
+
+def kv_append(k_cache, v_cache, k_new, v_new, pos: int) -> None:
+    if k_cache.is_cuda:
+        hip_ext().kv_append(k_cache, v_cache, k_new, v_new, pos)
+        return
+    ref.kv_append(k_cache, v_cache, k_new, v_new, pos)

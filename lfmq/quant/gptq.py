@@ -1,0 +1,205 @@
+"""GPTQ calibration: per-channel Hessian accumulate, damped Cholesky
+inverse, column-blocked quantize-then-error-propagate loop.
+
+Math (the north star's inventory, BASELINE.json:5):
+  H   = Σ 2·XᵀX                      (ops.hessian_accum — HIP K1 on GPU)
+  H_d = H + λ·mean(diag H)·I          (λ = damp_percent)
+  U   = chol(H_d⁻¹, upper)            (torch.linalg → rocSOLVER on GPU)
+  for each column j (in blocks):      q_j = quant(w_j)
+      err = (w_j − dq_j) / U_jj ;  w_{j+1..} −= err ⊗ U_{j, j+1..}
+Scales/zeros are recomputed at every group boundary from the
+error-compensated weights.  act_order=True quantizes columns in
+descending diag(H) order and records g_idx (per-column group index).
+
+Calibration is sequential per decoder block: only one block's Hessians
+(≈1 GB fp32 for 8B, ≈4 GB for 70B) are live at a time, and the next
+block's inputs are produced by the already-quantized block, so error
+propagates realistically.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from ..config import QuantConfig
+from ..models.qlinear import QLinear
+from ..ops import ref
+from .calib import CalibStream, InputCatcher
+from .common import decoder_blocks, iter_target_linears, replace_module
+
+
+def _hessian_accum(h: torch.Tensor, x: torch.Tensor) -> None:
+    if x.is_cuda:
+        from ..ops import hip_ext
+        hip_ext().hessian_accum(h, x)
+    else:
+        ref.hessian_accum(h, x)
+
+
+@torch.no_grad()
+def gptq_quantize_layer(
+    w: torch.Tensor, h: torch.Tensor, cfg: QuantConfig
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
+    """w fp32 [n, k], h fp32 [k, k] -> (codes int32 [n,k], scales fp32 [n,G],
+    zeros int32 [n,G], g_idx int32 [k] or None).
+
+    codes/scales/zeros are in ORIGINAL column order; with act_order the
+    per-column group assignment is returned as g_idx (scales row-indexed
+    by that group id).
+    """
+    n, k = w.shape
+    g = k if cfg.group_size == -1 else cfg.group_size
+    G = k // g
+    w = w.float().clone()
+    h = h.float().clone()
+    dev = w.device
+
+    dead = torch.diag(h) == 0
+    if dead.any():
+        h[dead, dead] = 1.0
+        w[:, dead] = 0.0
+
+    perm = None
+    if cfg.act_order:
+        perm = torch.argsort(torch.diag(h), descending=True)
+        w = w[:, perm]
+        h = h[perm][:, perm]
+
+    damp = cfg.damp_percent * torch.diag(h).mean()
+    h += torch.eye(k, device=dev) * damp
+
+    # U = chol(H^-1, upper) via the standard two-step (robust on rocSOLVER):
+    L = torch.linalg.cholesky(h)
+    hinv = torch.cholesky_inverse(L)
+    U = torch.linalg.cholesky(hinv, upper=True)
+
+    codes = torch.zeros(n, k, dtype=torch.int32, device=dev)
+    scales = torch.zeros(n, G, dtype=torch.float32, device=dev)
+    zeros = torch.zeros(n, G, dtype=torch.int32, device=dev)
+    qmax = (1 << cfg.bits) - 1
+
+    bs = cfg.block_size
+    scale_cur = torch.ones(n, device=dev)
+    zero_cur = torch.zeros(n, dtype=torch.int32, device=dev)
+    for i1 in range(0, k, bs):
+        i2 = min(i1 + bs, k)
+        wb = w[:, i1:i2].clone()
+        err = torch.zeros_like(wb)
+        ub = U[i1:i2, i1:i2]
+        for j in range(i2 - i1):
+            col = i1 + j
+            if col % g == 0:
+                # (re)fit scale/zero on the error-compensated group slice
+                grp = w[:, col:col + g]
+                _, s_, z_ = ref.quant_groupwise(grp, cfg.bits, g, cfg.sym)
+                gi = col // g
+                scales[:, gi] = s_[:, 0]
+                zeros[:, gi] = z_[:, 0]
+                scale_cur = s_[:, 0].to(dev)
+                zero_cur = z_[:, 0].to(dev)
+            wj = wb[:, j]
+            q = torch.clamp(torch.round(wj / scale_cur) + zero_cur, 0, qmax)
+            codes[:, col] = q.to(torch.int32)
+            dq = (q - zero_cur) * scale_cur
+            e = (wj - dq) / ub[j, j]
+            if j + 1 < i2 - i1:
+                wb[:, j + 1:] -= torch.outer(e, ub[j, j + 1:])
+            err[:, j] = e
+        w[:, i1:i2] = wb
+        if i2 < k:
+            w[:, i2:] -= err @ U[i1:i2, i2:]
+
+    g_idx = None
+    if perm is not None:
+        inv = torch.argsort(perm)
+        codes = codes[:, inv]
+        # group of original column c = (its position in permuted order) // g
+        g_idx = (inv // g).to(torch.int32)
+    return codes, scales, zeros, g_idx
+
+
+class GPTQQuantizer:
+    def __init__(self, cfg: QuantConfig):
+        self.cfg = cfg
+
+    @torch.no_grad()
+    def run(self, model: nn.Module, calib: Optional[CalibStream] = None,
+            device: Optional[torch.device] = None) -> nn.Module:
+        cfg = self.cfg
+        device = device or next(model.parameters()).device
+        if calib is None:
+            vocab = model.cfg.vocab_size
+            calib = CalibStream(vocab, cfg.calib_seq_len, cfg.calib_n_seqs,
+                                cfg.calib_seed)
+
+        from .common import make_block_runner
+        run_block = make_block_runner(model)
+        xs = _capture_block0_inputs(model, calib, device)
+
+        target_names = {n for n, _ in iter_target_linears(model, cfg)}
+        for bname, block in decoder_blocks(model):
+            block.to(device)
+            linears = {n: m for n, m in block.named_modules()
+                       if isinstance(m, nn.Linear) and f"{bname}.{n}" in
+                       {t for t in target_names if t.startswith(bname + ".")}}
+            # accumulate H per linear over all calib batches
+            hess: Dict[str, torch.Tensor] = {
+                n: torch.zeros(m.in_features, m.in_features,
+                               dtype=torch.float32, device=device)
+                for n, m in linears.items()}
+
+            def consume(name, x):
+                _hessian_accum(hess[name], x.to(device))
+
+            with InputCatcher(linears, consume=consume):
+                for x in xs:
+                    run_block(block, x.to(device))
+
+            for lname, lin in linears.items():
+                codes, scales, zeros, g_idx = gptq_quantize_layer(
+                    lin.weight.float(), hess[lname], cfg)
+                q = QLinear(lin.in_features, lin.out_features, cfg.bits,
+                            cfg.group_size if cfg.group_size != -1 else lin.in_features,
+                            bias=lin.bias is not None)
+                q.load_quantized(codes.cpu(), scales.cpu(), zeros.cpu(),
+                                 lin.bias.detach() if lin.bias is not None else None)
+                if g_idx is not None:
+                    q.register_buffer("g_idx", g_idx.cpu())
+                q.to(device)
+                replace_module(block, lname, q)
+                del hess[lname]
+
+            # next block's inputs come from the QUANTIZED block
+            xs = [run_block(block, x.to(device)).to(xs[0].device) for x in xs]
+        return model
+
+
+@torch.no_grad()
+def _capture_block0_inputs(model, calib: CalibStream, device) -> List[torch.Tensor]:
+    """Run the embedding front-end only, catching block-0 inputs."""
+    blocks = decoder_blocks(model)
+    first = blocks[0][1]
+
+    class _Stop(Exception):
+        pass
+
+    caught: List[torch.Tensor] = []
+
+    def hook(m, args, kwargs):
+        caught.append(args[0].detach().cpu())
+        raise _Stop
+
+    h = first.register_forward_pre_hook(hook, with_kwargs=True)
+    try:
+        for ids in calib.batches():
+            try:
+                cache = model.make_cache(ids.shape[0], max_seq=ids.shape[1],
+                                         device=device)
+                model(ids.to(device), cache)
+            except _Stop:
+                pass
+    finally:
+        h.remove()
+    return caught
