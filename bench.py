@@ -1,0 +1,86 @@
+#!/usr/bin/env python3
+"""Driver bench contract: flagship quantized decode on N GPUs of one node.
+
+    python bench.py --gpus N --steps K --warmup W
+    (N>1: launched by the driver via torch.distributed.run, one rank/GPU,
+     RCCL over xGMI)
+
+Measures the BASELINE.json headline metric — Llama-3-8B W4A16 group-size
+128 greedy-decode tokens/sec — on synthetic data and random-init weights
+(no network; BASELINE.md protocol).  value = whole-job tokens/s (batch ×
+steps ÷ max-over-ranks elapsed).  TP keeps total work fixed as N grows
+=> scaling "strong".
+"""
+import argparse
+import json
+import os
+import sys
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=64)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--batch", type=int, default=1)
+    ap.add_argument("--prompt", type=int, default=128)
+    ap.add_argument("--bits", type=int, default=4)
+    ap.add_argument("--group-size", type=int, default=128)
+    ap.add_argument("--layers", type=int, default=0,
+                    help="override layer count (debug only; 0 = full model)")
+    args = ap.parse_args()
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    import lfmq.dist as lfmq_dist
+    from lfmq.bench.decode import bench_decode, build_quantized_llama
+
+    rank, world = lfmq_dist.init_dist()
+    assert world == args.gpus or world == 1, \
+        f"WORLD_SIZE {world} != --gpus {args.gpus}"
+    world = max(world, 1)
+
+    if torch.cuda.is_available():
+        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0")))
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")  # contract smoke-run only
+
+    model, cfg, qcfg = build_quantized_llama(
+        args.model, "rtn", args.bits, args.group_size, device,
+        tp_rank=rank, tp_world=world, seed=0,
+        n_layers=args.layers or None)
+
+    r = bench_decode(model, cfg, args.batch, args.prompt, args.steps,
+                     args.warmup, device)
+
+    if rank == 0:
+        out = {
+            "metric": "decode tok/s",
+            "value": r["tok_per_s"],
+            "unit": "tok/s",
+            "n_gpus": world,
+            "steps": r["steps"],
+            "warmup": r["warmup"],
+            "ms_per_step": r["ms_per_step"],
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "quant": f"w{args.bits}a16-g{args.group_size}-rtn",
+                "global_batch": args.batch,
+                "seq_len": args.prompt,
+                "gen_steps": r["steps"],
+                "parallelism": f"tp{world}",
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

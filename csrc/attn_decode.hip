@@ -1,0 +1,173 @@
+// K8 — GQA decode attention: o = softmax(q·Kᵀ/√d)·V against the KV cache.
+// Memory-bound cache scan: flash-decoding split over T.
+//
+// Phase A: grid (b·hq, ceil(T/256)); each 256-thread workgroup owns a
+//   256-token chunk: one cache row per THREAD for q·K (scattered 16-B
+//   loads that still consume whole lines across the d-loop), then a
+//   per-wave PV pass (one V row per wave-instruction, coalesced 256 B)
+//   with p broadcast from LDS.  Partials (o fp32[d], m, l) go to scratch.
+// Phase B: one workgroup per (b, head) merges chunk partials with the
+//   standard flash rescale and writes bf16.
+// The 4 q-heads of a kv-head re-read the same K/V; chunk-major grid order
+// keeps those reads L2-resident.
+#include "common.h"
+
+namespace {
+
+constexpr int TPB = 256;       // threads (= tokens per chunk)
+constexpr int NWAVE = TPB / LFMQ_WAVE;
+
+template <int D>
+__global__ __launch_bounds__(TPB) void attn_decode_partial(
+    const u32* __restrict__ q, const u32* __restrict__ kc,
+    const u32* __restrict__ vc, float* __restrict__ scratch, int b, int hq,
+    int hkv, int tmax, int T, float scale, int nchunks) {
+  const int bh = blockIdx.x;          // b*hq flat
+  const int chunk = blockIdx.y;
+  const int bb = bh / hq;
+  const int h = bh % hq;
+  const int hk = h / (hq / hkv);
+  const int dw = D / 2;
+
+  const u32* qrow = q + (u64)bh * dw;
+  const u32* krow0 = kc + ((u64)bb * hkv + hk) * tmax * dw;
+  const u32* vrow0 = vc + ((u64)bb * hkv + hk) * tmax * dw;
+
+  __shared__ float p_sh[TPB];
+  __shared__ float red[NWAVE];
+  __shared__ float o_sh[NWAVE][D];
+  __shared__ float m_sh, l_sh;
+
+  // ---- q·K: one cache row per thread --------------------------------
+  const int t = chunk * TPB + threadIdx.x;
+  float s = -INFINITY;
+  if (t < T) {
+    const u32* krow = krow0 + (u64)t * dw;
+    float acc = 0.f;
+#pragma unroll
+    for (int i = 0; i < dw / 4; ++i) {
+      const u32x4 kv4 = *(const u32x4*)(krow + 4 * i);
+      const u32x4 qv4 = *(const u32x4*)(qrow + 4 * i);
+      acc = dot2_bf16(kv4[0], qv4[0], acc);
+      acc = dot2_bf16(kv4[1], qv4[1], acc);
+      acc = dot2_bf16(kv4[2], qv4[2], acc);
+      acc = dot2_bf16(kv4[3], qv4[3], acc);
+    }
+    s = acc * scale;
+  }
+
+  // ---- chunk softmax --------------------------------------------------
+  const int wave = threadIdx.x / LFMQ_WAVE;
+  const int lane = threadIdx.x % LFMQ_WAVE;
+  float wm = wave_reduce_max(s);
+  if (lane == 0) red[wave] = wm;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float m = red[0];
+#pragma unroll
+    for (int i = 1; i < NWAVE; ++i) m = fmaxf(m, red[i]);
+    m_sh = m;
+  }
+  __syncthreads();
+  const float m = m_sh;
+  const float p = (t < T) ? __expf(s - m) : 0.f;
+  p_sh[threadIdx.x] = p;
+  float wl = wave_reduce_sum(p);
+  if (lane == 0) red[wave] = wl;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float l = 0.f;
+#pragma unroll
+    for (int i = 0; i < NWAVE; ++i) l += red[i];
+    l_sh = l;
+  }
+
+  // ---- PV: each wave sums its 64 tokens, one V row per instruction ---
+  float o0 = 0.f, o1 = 0.f;  // lane's dword (2 elems); lanes >= dw idle
+  const int tbase = chunk * TPB + wave * LFMQ_WAVE;
+#pragma unroll 4
+  for (int tt = 0; tt < LFMQ_WAVE; ++tt) {
+    const int tv = tbase + tt;
+    if (tv >= T) break;
+    const float pv = p_sh[wave * LFMQ_WAVE + tt];
+    if (lane < dw) {
+      float v0, v1;
+      bf16x2_to_f32(vrow0[(u64)tv * dw + lane], v0, v1);
+      o0 = fmaf(pv, v0, o0);
+      o1 = fmaf(pv, v1, o1);
+    }
+  }
+  if (lane < dw) {
+    o_sh[wave][2 * lane] = o0;
+    o_sh[wave][2 * lane + 1] = o1;
+  }
+  __syncthreads();
+
+  // ---- write partial: scratch[bh][chunk] = {o[D], m, l} ---------------
+  float* out = scratch + ((u64)bh * nchunks + chunk) * (D + 2);
+  for (int j = threadIdx.x; j < D; j += TPB) {
+    float v = 0.f;
+#pragma unroll
+    for (int w = 0; w < NWAVE; ++w) v += o_sh[w][j];
+    out[j] = v;
+  }
+  if (threadIdx.x == 0) {
+    out[D] = m;
+    out[D + 1] = l_sh;
+  }
+}
+
+template <int D>
+__global__ __launch_bounds__(128) void attn_decode_combine(
+    const float* __restrict__ scratch, u32* __restrict__ o, int nchunks,
+    int used_chunks_stride, int T) {
+  const int bh = blockIdx.x;
+  const int used = (T + TPB - 1) / TPB;
+  const float* base = scratch + (u64)bh * used_chunks_stride * (D + 2);
+
+  float M = -INFINITY;
+  for (int c = 0; c < used; ++c) M = fmaxf(M, base[c * (D + 2) + D]);
+  float L = 0.f;
+  for (int c = 0; c < used; ++c)
+    L += __expf(base[c * (D + 2) + D] - M) * base[c * (D + 2) + D + 1];
+  const float invL = 1.f / L;
+
+  const int j = threadIdx.x;  // two elems per thread via dword packing
+  if (2 * j < D) {
+    float acc0 = 0.f, acc1 = 0.f;
+    for (int c = 0; c < used; ++c) {
+      const float w = __expf(base[c * (D + 2) + D] - M);
+      acc0 = fmaf(w, base[c * (D + 2) + 2 * j], acc0);
+      acc1 = fmaf(w, base[c * (D + 2) + 2 * j + 1], acc1);
+    }
+    o[(u64)bh * (D / 2) + j] = pack_bf16x2(acc0 * invL, acc1 * invL);
+  }
+}
+
+}  // namespace
+
+// scratch must hold b*hq*ceil(T/256)*(d+2) floats
+hipError_t lfmq_attn_decode(const void* q, const void* kc, const void* vc,
+                            void* o, void* scratch, int b, int hq, int hkv,
+                            int tmax, int T, int d, float scale,
+                            hipStream_t stream) {
+  if (d != 128 && d != 64) return hipErrorInvalidValue;
+  if (hq % hkv) return hipErrorInvalidValue;
+  const int nchunks = (T + TPB - 1) / TPB;
+  dim3 gridA(b * hq, nchunks), blockA(TPB);
+  dim3 gridB(b * hq), blockB(128);
+  if (d == 128) {
+    hipLaunchKernelGGL((attn_decode_partial<128>), gridA, blockA, 0, stream,
+                       (const u32*)q, (const u32*)kc, (const u32*)vc,
+                       (float*)scratch, b, hq, hkv, tmax, T, scale, nchunks);
+    hipLaunchKernelGGL((attn_decode_combine<128>), gridB, blockB, 0, stream,
+                       (const float*)scratch, (u32*)o, nchunks, nchunks, T);
+  } else {
+    hipLaunchKernelGGL((attn_decode_partial<64>), gridA, blockA, 0, stream,
+                       (const u32*)q, (const u32*)kc, (const u32*)vc,
+                       (float*)scratch, b, hq, hkv, tmax, T, scale, nchunks);
+    hipLaunchKernelGGL((attn_decode_combine<64>), gridB, blockB, 0, stream,
+                       (const float*)scratch, (u32*)o, nchunks, nchunks, T);
+  }
+  return hipGetLastError();
+}

@@ -1,0 +1,234 @@
+// lfmq_hip — torch-extension bindings for the gfx950 kernels.
+// Every entry validates dtype/contiguity/device and launches on the
+// current HIP stream.  No fallbacks: errors raise.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+
+hipError_t lfmq_pack(const void*, void*, long, int, int, hipStream_t);
+hipError_t lfmq_unpack(const void*, void*, long, int, int, hipStream_t);
+hipError_t lfmq_dq_gemv(const void*, const void*, const void*, const void*,
+                        void*, const void*, int, int, int, int, int,
+                        hipStream_t);
+hipError_t lfmq_dq_gemm(const void*, const void*, const void*, const void*,
+                        void*, const void*, int, int, int, int, int,
+                        hipStream_t);
+hipError_t lfmq_rmsnorm(const void*, const void*, const void*, void*, void*,
+                        int, int, float, hipStream_t);
+hipError_t lfmq_rope(void*, void*, const void*, const void*, int, int, int,
+                     int, int, hipStream_t);
+hipError_t lfmq_silu_mul(const void*, const void*, void*, long, hipStream_t);
+hipError_t lfmq_silu_mul_fused(const void*, void*, long, int, hipStream_t);
+hipError_t lfmq_residual_add(const void*, const void*, void*, long,
+                             hipStream_t);
+hipError_t lfmq_kv_append(void*, void*, const void*, const void*, int, int,
+                          int, int, int, int, hipStream_t);
+hipError_t lfmq_attn_decode(const void*, const void*, const void*, void*,
+                            void*, int, int, int, int, int, int, float,
+                            hipStream_t);
+hipError_t lfmq_hessian_accum(const void*, void*, int, int, hipStream_t);
+
+namespace {
+
+#define CHECK_HIP(call)                                                     \
+  do {                                                                      \
+    hipError_t e = (call);                                                  \
+    TORCH_CHECK(e == hipSuccess, "lfmq HIP error: ", hipGetErrorString(e)); \
+  } while (0)
+
+void check_t(const torch::Tensor& t, torch::ScalarType dt, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == dt, name, " has wrong dtype ",
+              t.scalar_type(), " expected ", dt);
+}
+
+hipStream_t stream() {
+  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+torch::Tensor pack(torch::Tensor codes, long bits) {
+  check_t(codes, torch::kInt32, "codes");
+  const long n = codes.size(0), k = codes.size(1);
+  const long per = 32 / bits;
+  TORCH_CHECK(k % per == 0, "k must be divisible by ", per);
+  auto out = torch::empty({n, k / per}, codes.options());
+  CHECK_HIP(lfmq_pack(codes.data_ptr(), out.data_ptr(), n, (int)k, (int)bits,
+                      stream()));
+  return out;
+}
+
+torch::Tensor unpack(torch::Tensor packed, long bits, long k) {
+  check_t(packed, torch::kInt32, "packed");
+  const long n = packed.size(0);
+  auto out = torch::empty({n, k}, packed.options());
+  CHECK_HIP(lfmq_unpack(packed.data_ptr(), out.data_ptr(), n, (int)k,
+                        (int)bits, stream()));
+  return out;
+}
+
+torch::Tensor dq_linear(torch::Tensor x, torch::Tensor qweight,
+                        torch::Tensor scales, torch::Tensor qzeros, long bits,
+                        long g, c10::optional<torch::Tensor> bias) {
+  check_t(x, torch::kBFloat16, "x");
+  check_t(qweight, torch::kInt32, "qweight");
+  check_t(scales, torch::kBFloat16, "scales");
+  check_t(qzeros, torch::kInt32, "qzeros");
+  TORCH_CHECK(x.dim() == 2, "x must be [m, k]");
+  const int m = (int)x.size(0), k = (int)x.size(1);
+  const int n = (int)qweight.size(0);
+  TORCH_CHECK(qweight.size(1) == (long)k * bits / 32, "qweight/k mismatch");
+  const void* bptr = nullptr;
+  if (bias.has_value()) {
+    check_t(*bias, torch::kBFloat16, "bias");
+    bptr = bias->data_ptr();
+  }
+  auto y = torch::empty({m, n}, x.options());
+  if (m <= 16) {
+    CHECK_HIP(lfmq_dq_gemv(qweight.data_ptr(), scales.data_ptr(),
+                           qzeros.data_ptr(), x.data_ptr(), y.data_ptr(), bptr,
+                           m, n, k, (int)g, (int)bits, stream()));
+  } else {
+    CHECK_HIP(lfmq_dq_gemm(qweight.data_ptr(), scales.data_ptr(),
+                           qzeros.data_ptr(), x.data_ptr(), y.data_ptr(), bptr,
+                           m, n, k, (int)g, (int)bits, stream()));
+  }
+  return y;
+}
+
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps,
+                      c10::optional<torch::Tensor> residual) {
+  check_t(x, torch::kBFloat16, "x");
+  check_t(w, torch::kBFloat16, "w");
+  const int hidden = (int)x.size(-1);
+  const int rows = (int)(x.numel() / hidden);
+  auto y = torch::empty_like(x);
+  const void* rptr = nullptr;
+  if (residual.has_value()) {
+    check_t(*residual, torch::kBFloat16, "residual");
+    rptr = residual->data_ptr();
+  }
+  CHECK_HIP(lfmq_rmsnorm(x.data_ptr(), rptr, w.data_ptr(), y.data_ptr(),
+                         nullptr, rows, hidden, (float)eps, stream()));
+  return y;
+}
+
+std::vector<torch::Tensor> rmsnorm_residual(torch::Tensor x,
+                                            torch::Tensor residual,
+                                            torch::Tensor w, double eps) {
+  check_t(x, torch::kBFloat16, "x");
+  check_t(residual, torch::kBFloat16, "residual");
+  check_t(w, torch::kBFloat16, "w");
+  const int hidden = (int)x.size(-1);
+  const int rows = (int)(x.numel() / hidden);
+  auto y = torch::empty_like(x);
+  auto res_out = torch::empty_like(x);
+  CHECK_HIP(lfmq_rmsnorm(x.data_ptr(), residual.data_ptr(), w.data_ptr(),
+                         y.data_ptr(), res_out.data_ptr(), rows, hidden,
+                         (float)eps, stream()));
+  return {y, res_out};
+}
+
+void rope(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
+          torch::Tensor inv_freq) {
+  check_t(q, torch::kBFloat16, "q");
+  check_t(k, torch::kBFloat16, "k");
+  check_t(positions, torch::kInt32, "positions");
+  check_t(inv_freq, torch::kFloat32, "inv_freq");
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4, "q/k must be [b, s, h, d]");
+  const int b = (int)q.size(0), s = (int)q.size(1);
+  const int hq = (int)q.size(2), d = (int)q.size(3);
+  const int hkv = (int)k.size(2);
+  TORCH_CHECK(positions.numel() == s, "positions must have s entries");
+  CHECK_HIP(lfmq_rope(q.data_ptr(), k.data_ptr(), positions.data_ptr(),
+                      inv_freq.data_ptr(), b * s, s, hq, hkv, d, stream()));
+}
+
+torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up) {
+  check_t(gate, torch::kBFloat16, "gate");
+  check_t(up, torch::kBFloat16, "up");
+  auto y = torch::empty_like(gate);
+  CHECK_HIP(lfmq_silu_mul(gate.data_ptr(), up.data_ptr(), y.data_ptr(),
+                          gate.numel(), stream()));
+  return y;
+}
+
+torch::Tensor silu_mul_fused(torch::Tensor gate_up) {
+  check_t(gate_up, torch::kBFloat16, "gate_up");
+  const long inter = gate_up.size(-1) / 2;
+  const long rows = gate_up.numel() / (2 * inter);
+  auto sizes = gate_up.sizes().vec();
+  sizes.back() = inter;
+  auto y = torch::empty(sizes, gate_up.options());
+  CHECK_HIP(lfmq_silu_mul_fused(gate_up.data_ptr(), y.data_ptr(), rows,
+                                (int)inter, stream()));
+  return y;
+}
+
+torch::Tensor residual_add(torch::Tensor a, torch::Tensor b) {
+  check_t(a, torch::kBFloat16, "a");
+  check_t(b, torch::kBFloat16, "b");
+  auto y = torch::empty_like(a);
+  CHECK_HIP(lfmq_residual_add(a.data_ptr(), b.data_ptr(), y.data_ptr(),
+                              a.numel(), stream()));
+  return y;
+}
+
+void kv_append(torch::Tensor kc, torch::Tensor vc, torch::Tensor kn,
+               torch::Tensor vn, long pos0) {
+  check_t(kc, torch::kBFloat16, "k_cache");
+  check_t(vc, torch::kBFloat16, "v_cache");
+  check_t(kn, torch::kBFloat16, "k_new");
+  check_t(vn, torch::kBFloat16, "v_new");
+  // caches [b, hkv, tmax, d]; new [b, s, hkv, d] (token-major)
+  const int b = (int)kc.size(0), hkv = (int)kc.size(1);
+  const int tmax = (int)kc.size(2), d = (int)kc.size(3);
+  const int s = (int)kn.size(1);
+  TORCH_CHECK(kn.size(0) == b && kn.size(2) == hkv && kn.size(3) == d,
+              "k_new shape mismatch");
+  CHECK_HIP(lfmq_kv_append(kc.data_ptr(), vc.data_ptr(), kn.data_ptr(),
+                           vn.data_ptr(), b, s, hkv, d, tmax, (int)pos0,
+                           stream()));
+}
+
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
+                          long seq_len, double scale) {
+  check_t(q, torch::kBFloat16, "q");
+  check_t(kc, torch::kBFloat16, "k_cache");
+  check_t(vc, torch::kBFloat16, "v_cache");
+  const int b = (int)q.size(0), hq = (int)q.size(1), d = (int)q.size(3);
+  const int hkv = (int)kc.size(1), tmax = (int)kc.size(2);
+  const int nchunks = ((int)seq_len + 255) / 256;
+  auto scratch = torch::empty({(long)b * hq * nchunks * (d + 2)},
+                              q.options().dtype(torch::kFloat32));
+  auto o = torch::empty_like(q);
+  CHECK_HIP(lfmq_attn_decode(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                             o.data_ptr(), scratch.data_ptr(), b, hq, hkv,
+                             tmax, (int)seq_len, d, (float)scale, stream()));
+  return o;
+}
+
+void hessian_accum(torch::Tensor h, torch::Tensor x) {
+  check_t(h, torch::kFloat32, "H");
+  check_t(x, torch::kBFloat16, "x");
+  const int t = (int)x.size(0), k = (int)x.size(1);
+  TORCH_CHECK(h.size(0) == k && h.size(1) == k, "H must be [k, k]");
+  CHECK_HIP(lfmq_hessian_accum(x.data_ptr(), h.data_ptr(), t, k, stream()));
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("pack", &pack, "pack int codes");
+  m.def("unpack", &unpack, "unpack int codes");
+  m.def("dq_linear", &dq_linear, "fused dequant GEMV/GEMM");
+  m.def("rmsnorm", &rmsnorm, "fused rmsnorm");
+  m.def("rmsnorm_residual", &rmsnorm_residual, "fused rmsnorm + residual");
+  m.def("rope", &rope, "fused rope (in-place)");
+  m.def("silu_mul", &silu_mul, "silu(gate) * up");
+  m.def("silu_mul_fused", &silu_mul_fused, "silu-mul on fused gate_up");
+  m.def("residual_add", &residual_add, "bf16 add");
+  m.def("kv_append", &kv_append, "KV cache append");
+  m.def("attn_decode", &attn_decode, "GQA decode attention");
+  m.def("hessian_accum", &hessian_accum, "H += 2 X^T X");
+}

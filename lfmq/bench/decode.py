@@ -1,0 +1,96 @@
+"""Decode tok/s harness (the BASELINE.json headline metric).
+
+Greedy decode on synthetic data / random-init weights (no network => no
+pretrained checkpoints).  Timing protocol: untimed prefill +
+warmup decode steps, then barrier + synchronize, time EXACTLY K decode
+steps, synchronize + barrier, MAX over ranks.  Nothing is skipped inside
+the timed region: every step runs all layers + lm_head + argmax and feeds
+the sampled token back.
+"""
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import torch
+
+from .. import dist as lfmq_dist
+from ..api import build_model, quantize
+from ..config import QuantConfig
+from ..models.llama import LlamaConfig
+
+
+@torch.no_grad()
+def _fast_random_init(model, seed: int):
+    torch.manual_seed(seed)
+    for p in model.parameters():
+        p.data.normal_(0.0, 0.02)
+    return model
+
+
+@torch.no_grad()
+def build_quantized_llama(model_name: str, scheme: str, bits: int, g: int,
+                          device, tp_rank: int = 0, tp_world: int = 1,
+                          seed: int = 0, n_layers: Optional[int] = None):
+    """Random-init on device, quantize (RTN weights for bench speed: the
+    decode kernel path and memory traffic are scheme-independent), shard."""
+    if model_name == "llama3-8b":
+        cfg = LlamaConfig.llama3_8b()
+    elif model_name == "llama3-70b":
+        cfg = LlamaConfig.llama3_70b()
+    else:
+        raise ValueError(model_name)
+    if n_layers:
+        cfg.n_layers = n_layers
+    model = build_model(cfg)
+    model = model.to_empty(device=device) if device != "cpu" else model
+    _fast_random_init(model, seed)
+    qcfg = QuantConfig(scheme=scheme, bits=bits, group_size=g)
+    model = quantize(model, qcfg)
+    if tp_world > 1:
+        from ..dist.tp import shard_llama
+        model = shard_llama(model, tp_rank, tp_world)
+    return model, cfg, qcfg
+
+
+@torch.no_grad()
+def bench_decode(model, cfg, batch: int, prompt: int, steps: int, warmup: int,
+                 device) -> dict:
+    ids = torch.randint(0, cfg.vocab_size, (batch, prompt), device=device)
+    cache = model.make_cache(batch, max_seq=prompt + warmup + steps + 8,
+                             device=device)
+    logits = model(ids, cache)          # prefill (untimed)
+    tok = logits[:, -1:].argmax(dim=-1)
+    for _ in range(warmup):
+        logits = model(tok, cache)
+        tok = logits[:, -1:].argmax(dim=-1)
+
+    use_cuda = torch.cuda.is_available() and str(device) != "cpu"
+    lfmq_dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        logits = model(tok, cache)
+        tok = logits[:, -1:].argmax(dim=-1)
+    if use_cuda:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    lfmq_dist.barrier()
+
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+    if lfmq_dist.get_world_size() > 1:
+        import torch.distributed as dist
+        e = elapsed.to(device) if use_cuda else elapsed
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = e.cpu()
+    el = float(elapsed[0])
+    return {
+        "elapsed_s": el,
+        "ms_per_step": el / steps * 1000.0,
+        "tok_per_s": batch * steps / el,
+        "steps": steps,
+        "warmup": warmup,
+        "batch": batch,
+        "prompt": prompt,
+    }

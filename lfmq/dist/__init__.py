@@ -49,9 +49,21 @@ def get_tp_group():
     return _TP_GROUP  # None => default group
 
 
+def _backend_is_gloo() -> bool:
+    try:
+        return dist.get_backend() == "gloo"
+    except Exception:
+        return False
+
+
 def all_reduce_sum(t: torch.Tensor) -> torch.Tensor:
     if get_world_size() > 1:
-        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=get_tp_group())
+        if t.dtype == torch.bfloat16 and _backend_is_gloo():
+            f = t.float()
+            dist.all_reduce(f, op=dist.ReduceOp.SUM, group=get_tp_group())
+            t.copy_(f.to(t.dtype))
+        else:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM, group=get_tp_group())
     return t
 
 

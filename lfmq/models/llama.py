@@ -103,8 +103,7 @@ class LlamaAttention(nn.Module):
         v = v.reshape(b, s, hkv, d)
         q, k = F.apply_rope(q, k, positions, self.cfg.rope_theta)
         pos0 = seq_len_after - s
-        F.kv_append(cache_k, cache_v,
-                    k.transpose(1, 2), v.transpose(1, 2), pos0)
+        F.kv_append(cache_k, cache_v, k, v, pos0)
         qh = q.transpose(1, 2)  # [b, hq, s, d]
         if s == 1:
             o = F.attn_decode(qh, cache_k, cache_v, seq_len_after, self.scale)

@@ -48,21 +48,22 @@ class QLinear(nn.Module):
     @torch.no_grad()
     def load_quantized(self, codes: torch.Tensor, scales: torch.Tensor,
                        zeros: torch.Tensor, bias: Optional[torch.Tensor] = None):
-        """codes int32 [n,k], scales fp32/bf16 [n,G], zeros int32 [n,G]."""
-        self.qweight.copy_(ref.pack_codes(codes.cpu(), self.bits))
+        """codes int32 [n,k], scales fp32/bf16 [n,G], zeros int32 [n,G].
+        Device-agnostic: pack runs on whatever device the codes live on."""
+        self.qweight.copy_(ref.pack_codes(codes, self.bits))
         self.scales.copy_(scales.to(self.scales.dtype))
-        self.qzeros.copy_(ref.pack_zeros(zeros.cpu().to(torch.int32), self.bits))
+        self.qzeros.copy_(ref.pack_zeros(zeros.to(torch.int32), self.bits))
         if bias is not None:
             self.bias.copy_(bias.to(self.scales.dtype))
 
     @classmethod
     @torch.no_grad()
     def from_rtn(cls, linear: nn.Linear, cfg: QuantConfig) -> "QLinear":
-        w = linear.weight.float().cpu()
+        w = linear.weight.detach().float()
         codes, scales, zeros = ref.quant_groupwise(w, cfg.bits, cfg.group_size, cfg.sym)
         q = cls(linear.in_features, linear.out_features, cfg.bits,
                 cfg.group_size if cfg.group_size != -1 else linear.in_features,
-                bias=linear.bias is not None)
+                bias=linear.bias is not None).to(w.device)
         q.load_quantized(codes, scales, zeros,
                          linear.bias.detach() if linear.bias is not None else None)
         return q

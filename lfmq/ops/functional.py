@@ -32,11 +32,26 @@ def rmsnorm_residual(x: torch.Tensor, residual: torch.Tensor,
     return ref.rmsnorm(new_res, weight, eps), new_res
 
 
+_INV_FREQ_CACHE: dict = {}
+
+
+def _inv_freq(d: int, theta: float, device) -> torch.Tensor:
+    key = (d, theta, str(device))
+    t = _INV_FREQ_CACHE.get(key)
+    if t is None:
+        t = ref.rope_freqs(d, theta).to(device)
+        _INV_FREQ_CACHE[key] = t
+    return t
+
+
 def apply_rope(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
                theta: float = 500000.0):
     """q: [b, s, hq, d], k: [b, s, hkv, d] — rotated in place on GPU."""
     if q.is_cuda:
-        hip_ext().rope(q, k, positions, theta)
+        q = q.contiguous()
+        k = k.contiguous()
+        hip_ext().rope(q, k, positions.to(torch.int32),
+                       _inv_freq(q.shape[-1], theta, q.device))
         return q, k
     return ref.apply_rope(q, positions, theta), ref.apply_rope(k, positions, theta)
 
@@ -85,7 +100,10 @@ def attn_prefill(q, k, v, causal: bool = True, scale: Optional[float] = None):
 
 
 def kv_append(k_cache, v_cache, k_new, v_new, pos: int) -> None:
+    """caches [b, hkv, T_max, d]; k_new/v_new TOKEN-major [b, s, hkv, d]
+    (as the qkv split produces them — saves the transpose copies)."""
     if k_cache.is_cuda:
-        hip_ext().kv_append(k_cache, v_cache, k_new, v_new, pos)
+        hip_ext().kv_append(k_cache, v_cache, k_new.contiguous(),
+                            v_new.contiguous(), pos)
         return
-    ref.kv_append(k_cache, v_cache, k_new, v_new, pos)
+    ref.kv_append(k_cache, v_cache, k_new.transpose(1, 2), v_new.transpose(1, 2), pos)

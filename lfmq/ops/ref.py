@@ -102,7 +102,7 @@ def pack_codes(codes: torch.Tensor, bits: int) -> torch.Tensor:
     if k % per != 0:
         raise ValueError(f"k={k} not divisible by {per}")
     c = codes.to(torch.int64).reshape(n, k // per, per)
-    shifts = torch.arange(per, dtype=torch.int64) * bits
+    shifts = torch.arange(per, dtype=torch.int64, device=codes.device) * bits
     packed = (c << shifts).sum(dim=-1)
     return (packed & 0xFFFFFFFF).to(torch.int32)  # stored as int32, bits are uint32
 
@@ -112,7 +112,7 @@ def unpack_codes(packed: torch.Tensor, bits: int, k: int) -> torch.Tensor:
     n = packed.shape[0]
     per = 32 // bits
     p = packed.to(torch.int64) & 0xFFFFFFFF
-    shifts = torch.arange(per, dtype=torch.int64) * bits
+    shifts = torch.arange(per, dtype=torch.int64, device=packed.device) * bits
     mask = (1 << bits) - 1
     c = (p.unsqueeze(-1) >> shifts) & mask
     return c.reshape(n, -1)[:, :k].to(torch.int32)

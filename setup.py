@@ -1,0 +1,40 @@
+"""In-tree build of the lfmq_hip extension (gfx950 only).
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built lfmq_hip*.so lands at the repo root so it travels with gpurun
+snapshots (a JIT cache under ~/.cache would not).
+"""
+import os
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+SOURCES = [
+    "csrc/ext.hip",
+    "csrc/pack.hip",
+    "csrc/rmsnorm.hip",
+    "csrc/rope.hip",
+    "csrc/elementwise.hip",
+    "csrc/dq_gemv.hip",
+    "csrc/dq_gemm.hip",
+    "csrc/hessian.hip",
+    "csrc/attn_decode.hip",
+]
+
+setup(
+    name="lfmq_hip",
+    ext_modules=[
+        CUDAExtension(
+            name="lfmq_hip",
+            sources=SOURCES,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)

@@ -1,0 +1,200 @@
+"""GPU parity tests: every HIP kernel vs its CPU reference twin.
+
+All run in ONE gpurun call (`pytest -m gpu`).  Tolerances: exact for
+pack/unpack/kv_append; bf16-accumulation tolerances vs the fp32 reference
+for the compute kernels.  Inputs are random and ASYMMETRIC (guide §3:
+symmetric operands mask transposed-output bugs).
+"""
+import math
+
+import pytest
+import torch
+
+from lfmq.ops import ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def ext():
+    from lfmq.ops import hip_ext
+    return hip_ext()
+
+
+def rel_err(a, b):
+    a, b = a.float().cpu(), b.float().cpu()
+    return float((a - b).abs().max() / (b.abs().max() + 1e-8))
+
+
+# ---------------------------------------------------------------- pack
+@pytest.mark.parametrize("bits", [4, 8])
+def test_pack_unpack_gpu(bits):
+    codes = torch.randint(0, 1 << bits, (64, 256), dtype=torch.int32)
+    gpu = ext().pack(codes.to(DEV), bits)
+    assert torch.equal(gpu.cpu(), ref.pack_codes(codes, bits))
+    back = ext().unpack(gpu, bits, 256)
+    assert torch.equal(back.cpu(), codes)
+
+
+# ------------------------------------------------------------- mfma map
+def test_mfma_layout_probe_via_hessian():
+    """hessian_accum is a pure MFMA kernel: if the assumed 16x16x32
+    fragment layout (A/B: lane&15 row/col, 8 contiguous k at (lane>>4)*8;
+    C/D: col=lane&15, row=(lane>>4)*4+reg) is wrong, this fails."""
+    t, k = 64, 128
+    x = (torch.randn(t, k) * torch.linspace(0.5, 2.0, k)).to(torch.bfloat16)
+    h = torch.zeros(k, k, dtype=torch.float32)
+    hg = h.clone().to(DEV)
+    ext().hessian_accum(hg, x.to(DEV))
+    ref.hessian_accum(h, x.float())
+    assert rel_err(hg, h) < 2e-2
+
+
+@pytest.mark.parametrize("t,k", [(300, 256), (64, 4096), (1000, 512)])
+def test_hessian_accum(t, k):
+    x = torch.randn(t, k).to(torch.bfloat16)
+    h0 = torch.randn(k, k).float()
+    hg = h0.clone().to(DEV)
+    ext().hessian_accum(hg, x.to(DEV))
+    hc = h0.clone()
+    ref.hessian_accum(hc, x.float())
+    assert rel_err(hg, hc) < 2e-2
+
+
+# ---------------------------------------------------------------- gemv
+@pytest.mark.parametrize("bits", [4, 8])
+@pytest.mark.parametrize("m", [1, 2, 4, 8, 16])
+@pytest.mark.parametrize("n,k,g", [(256, 2048, 128), (37, 512, 64),
+                                   (129, 4096, 128), (64, 96, 32)])
+def test_dq_gemv(bits, m, n, k, g):
+    torch.manual_seed(bits * 1000 + m * 100 + n)
+    w = torch.randn(n, k)
+    codes, scales, zeros = ref.quant_groupwise(w, bits, g, sym=False)
+    qw = ref.pack_codes(codes, bits)
+    qz = ref.pack_zeros(zeros, bits)
+    x = torch.randn(m, k).to(torch.bfloat16)
+    y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(),
+                          qz, bits, g)
+    y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
+                        qz.to(DEV), bits, g, None)
+    assert y.shape == (m, n)
+    assert rel_err(y, y_ref) < 2e-2, f"rel={rel_err(y, y_ref)}"
+
+
+def test_dq_gemv_bias():
+    n, k, g = 128, 256, 64
+    w = torch.randn(n, k)
+    codes, scales, zeros = ref.quant_groupwise(w, 4, g, False)
+    qw, qz = ref.pack_codes(codes, 4), ref.pack_zeros(zeros, 4)
+    bias = torch.randn(n).to(torch.bfloat16)
+    x = torch.randn(2, k).to(torch.bfloat16)
+    y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(), qz,
+                          4, g, bias.float())
+    y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
+                        qz.to(DEV), 4, g, bias.to(DEV))
+    assert rel_err(y, y_ref) < 2e-2
+
+
+# ---------------------------------------------------------------- gemm
+@pytest.mark.parametrize("bits", [4, 8])
+@pytest.mark.parametrize("m,n,k,g", [(64, 256, 512, 128), (200, 300, 256, 64),
+                                     (128, 128, 128, 64), (513, 100, 1024, 128)])
+def test_dq_gemm(bits, m, n, k, g):
+    torch.manual_seed(m + n)
+    w = torch.randn(n, k)
+    codes, scales, zeros = ref.quant_groupwise(w, bits, g, False)
+    qw, qz = ref.pack_codes(codes, bits), ref.pack_zeros(zeros, bits)
+    x = torch.randn(m, k).to(torch.bfloat16)
+    y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(),
+                          qz, bits, g)
+    y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
+                        qz.to(DEV), bits, g, None)
+    assert rel_err(y, y_ref) < 3e-2, f"rel={rel_err(y, y_ref)}"
+
+
+# ------------------------------------------------------------- rmsnorm
+@pytest.mark.parametrize("rows,hidden", [(1, 4096), (17, 256), (64, 8192)])
+def test_rmsnorm(rows, hidden):
+    x = torch.randn(rows, hidden).to(torch.bfloat16)
+    w = torch.randn(hidden).to(torch.bfloat16)
+    y = ext().rmsnorm(x.to(DEV), w.to(DEV), 1e-5, None)
+    y_ref = ref.rmsnorm(x, w, 1e-5)
+    assert rel_err(y, y_ref) < 2e-2
+
+
+def test_rmsnorm_residual():
+    x = torch.randn(9, 512).to(torch.bfloat16)
+    r = torch.randn(9, 512).to(torch.bfloat16)
+    w = torch.randn(512).to(torch.bfloat16)
+    y, nr = ext().rmsnorm_residual(x.to(DEV), r.to(DEV), w.to(DEV), 1e-5)
+    y_ref = ref.rmsnorm(x, w, 1e-5, residual=r)
+    nr_ref = ref.residual_add(x, r)
+    assert rel_err(y, y_ref) < 2e-2
+    assert rel_err(nr, nr_ref) < 1e-2
+
+
+# ----------------------------------------------------------------- rope
+@pytest.mark.parametrize("s,pos0", [(1, 77), (16, 0)])
+def test_rope(s, pos0):
+    b, hq, hkv, d = 2, 8, 2, 128
+    q = torch.randn(b, s, hq, d).to(torch.bfloat16)
+    k = torch.randn(b, s, hkv, d).to(torch.bfloat16)
+    pos = torch.arange(pos0, pos0 + s)
+    qg, kg = q.to(DEV), k.to(DEV)
+    inv = ref.rope_freqs(d, 500000.0).to(DEV)
+    ext().rope(qg, kg, pos.to(torch.int32).to(DEV), inv)
+    q_ref = ref.apply_rope(q.float(), pos)
+    k_ref = ref.apply_rope(k.float(), pos)
+    assert rel_err(qg, q_ref) < 2e-2
+    assert rel_err(kg, k_ref) < 2e-2
+
+
+# ---------------------------------------------------------- elementwise
+def test_silu_mul():
+    g = torch.randn(33, 1024).to(torch.bfloat16)
+    u = torch.randn(33, 1024).to(torch.bfloat16)
+    y = ext().silu_mul(g.to(DEV), u.to(DEV))
+    assert rel_err(y, ref.silu_mul(g, u)) < 2e-2
+
+
+def test_silu_mul_fused():
+    gu = torch.randn(8, 512).to(torch.bfloat16)
+    y = ext().silu_mul_fused(gu.to(DEV))
+    g, u = gu.split([256, 256], dim=-1)
+    assert rel_err(y, ref.silu_mul(g.contiguous(), u.contiguous())) < 2e-2
+
+
+def test_residual_add():
+    a = torch.randn(7, 300).to(torch.bfloat16)
+    b = torch.randn(7, 300).to(torch.bfloat16)
+    y = ext().residual_add(a.to(DEV), b.to(DEV))
+    assert rel_err(y, ref.residual_add(a, b)) < 1e-2
+
+
+def test_kv_append():
+    b, s, hkv, d, tmax = 2, 3, 4, 128, 64
+    kc = torch.zeros(b, hkv, tmax, d).to(torch.bfloat16).to(DEV)
+    vc = torch.zeros_like(kc)
+    kn = torch.randn(b, s, hkv, d).to(torch.bfloat16)
+    vn = torch.randn(b, s, hkv, d).to(torch.bfloat16)
+    ext().kv_append(kc, vc, kn.to(DEV), vn.to(DEV), 5)
+    assert torch.equal(kc[:, :, 5:8].cpu(), kn.transpose(1, 2))
+    assert torch.equal(vc[:, :, 5:8].cpu(), vn.transpose(1, 2))
+    assert kc[:, :, :5].abs().sum().item() == 0
+
+
+# ---------------------------------------------------------- attn decode
+@pytest.mark.parametrize("b,T,d", [(1, 17, 128), (2, 256, 128), (1, 700, 128),
+                                   (4, 64, 64), (1, 2048, 128)])
+def test_attn_decode(b, T, d):
+    hq, hkv = 8, 2
+    tmax = T + 13
+    torch.manual_seed(T)
+    q = torch.randn(b, hq, 1, d).to(torch.bfloat16)
+    kc = torch.randn(b, hkv, tmax, d).to(torch.bfloat16)
+    vc = torch.randn(b, hkv, tmax, d).to(torch.bfloat16)
+    scale = 1.0 / math.sqrt(d)
+    o = ext().attn_decode(q.to(DEV), kc.to(DEV), vc.to(DEV), T, scale)
+    o_ref = ref.attn_decode(q.float(), kc.float(), vc.float(), T, scale)
+    assert rel_err(o, o_ref) < 2e-2, f"rel={rel_err(o, o_ref)}"
