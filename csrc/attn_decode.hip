@@ -1,28 +1,30 @@
 // K8 — GQA decode attention: o = softmax(q·Kᵀ/√d)·V against the KV cache.
 // Memory-bound cache scan: flash-decoding split over T.
 //
-// Phase A: grid (b·hq, ceil(T/256)); each 256-thread workgroup owns a
-//   256-token chunk: one cache row per THREAD for q·K (scattered 16-B
-//   loads that still consume whole lines across the d-loop), then a
-//   per-wave PV pass (one V row per wave-instruction, coalesced 256 B)
-//   with p broadcast from LDS.  Partials (o fp32[d], m, l) go to scratch.
-// Phase B: one workgroup per (b, head) merges chunk partials with the
-//   standard flash rescale and writes bf16.
-// The 4 q-heads of a kv-head re-read the same K/V; chunk-major grid order
-// keeps those reads L2-resident.
+// Grid (b·hq, ceil(T/256)); each 256-thread workgroup owns a 256-token
+// chunk: one cache row per THREAD for q·K (scattered 16-B loads that
+// still consume whole lines across the d-loop), then a per-wave PV pass
+// (one V row per wave-instruction, coalesced 256 B, 8 rows in flight)
+// with p broadcast from LDS.  nchunks == 1 (the common short-T decode
+// case) writes the normalized output directly — no scratch round-trip,
+// no combine launch; otherwise partials {o fp32[d], m, l} go to scratch
+// and a combine kernel merges them with the standard flash rescale.
+// Guards are branch-free: cache-row indices clamp to the allocation and
+// p = 0 kills t >= T contributions.
 #include "common.h"
 
 namespace {
 
-constexpr int TPB = 256;       // threads (= tokens per chunk)
+constexpr int TPB = 256;
 constexpr int NWAVE = TPB / LFMQ_WAVE;
 
 template <int D>
 __global__ __launch_bounds__(TPB) void attn_decode_partial(
     const u32* __restrict__ q, const u32* __restrict__ kc,
-    const u32* __restrict__ vc, float* __restrict__ scratch, int b, int hq,
-    int hkv, int tmax, int T, float scale, int nchunks) {
-  const int bh = blockIdx.x;          // b*hq flat
+    const u32* __restrict__ vc, float* __restrict__ scratch,
+    u32* __restrict__ o_direct, int b, int hq, int hkv, int tmax, int T,
+    float scale, int nchunks) {
+  const int bh = blockIdx.x;
   const int chunk = blockIdx.y;
   const int bb = bh / hq;
   const int h = bh % hq;
@@ -40,9 +42,10 @@ __global__ __launch_bounds__(TPB) void attn_decode_partial(
 
   // ---- q·K: one cache row per thread --------------------------------
   const int t = chunk * TPB + threadIdx.x;
+  const int t_clamped = t < tmax ? t : tmax - 1;
   float s = -INFINITY;
-  if (t < T) {
-    const u32* krow = krow0 + (u64)t * dw;
+  {
+    const u32* krow = krow0 + (u64)t_clamped * dw;
     float acc = 0.f;
 #pragma unroll
     for (int i = 0; i < dw / 4; ++i) {
@@ -53,7 +56,7 @@ __global__ __launch_bounds__(TPB) void attn_decode_partial(
       acc = dot2_bf16(kv4[2], qv4[2], acc);
       acc = dot2_bf16(kv4[3], qv4[3], acc);
     }
-    s = acc * scale;
+    if (t < T) s = acc * scale;
   }
 
   // ---- chunk softmax --------------------------------------------------
@@ -82,26 +85,42 @@ __global__ __launch_bounds__(TPB) void attn_decode_partial(
     l_sh = l;
   }
 
-  // ---- PV: each wave sums its 64 tokens, one V row per instruction ---
+  // ---- PV: each wave sums its 64 tokens; 8 V rows in flight ----------
   float o0 = 0.f, o1 = 0.f;  // lane's dword (2 elems); lanes >= dw idle
   const int tbase = chunk * TPB + wave * LFMQ_WAVE;
-#pragma unroll 4
-  for (int tt = 0; tt < LFMQ_WAVE; ++tt) {
-    const int tv = tbase + tt;
-    if (tv >= T) break;
-    const float pv = p_sh[wave * LFMQ_WAVE + tt];
-    if (lane < dw) {
+  const int nt = min(LFMQ_WAVE, T - tbase);  // tokens this wave covers
+  if (lane < dw) {
+#pragma unroll 8
+    for (int tt = 0; tt < LFMQ_WAVE; ++tt) {
+      if (tt >= nt) break;
+      const int tv = tbase + tt;
+      const int tvc = tv < tmax ? tv : tmax - 1;
       float v0, v1;
-      bf16x2_to_f32(vrow0[(u64)tv * dw + lane], v0, v1);
+      bf16x2_to_f32(vrow0[(u64)tvc * dw + lane], v0, v1);
+      const float pv = p_sh[wave * LFMQ_WAVE + tt];
       o0 = fmaf(pv, v0, o0);
       o1 = fmaf(pv, v1, o1);
     }
-  }
-  if (lane < dw) {
     o_sh[wave][2 * lane] = o0;
     o_sh[wave][2 * lane + 1] = o1;
   }
   __syncthreads();
+
+  if (nchunks == 1) {
+    // direct write: o = Σ_w o_w / l
+    const float invl = 1.f / l_sh;
+    if (threadIdx.x < dw) {
+      const int j = threadIdx.x;
+      float a0 = 0.f, a1 = 0.f;
+#pragma unroll
+      for (int w = 0; w < NWAVE; ++w) {
+        a0 += o_sh[w][2 * j];
+        a1 += o_sh[w][2 * j + 1];
+      }
+      o_direct[(u64)bh * dw + j] = pack_bf16x2(a0 * invl, a1 * invl);
+    }
+    return;
+  }
 
   // ---- write partial: scratch[bh][chunk] = {o[D], m, l} ---------------
   float* out = scratch + ((u64)bh * nchunks + chunk) * (D + 2);
@@ -120,10 +139,10 @@ __global__ __launch_bounds__(TPB) void attn_decode_partial(
 template <int D>
 __global__ __launch_bounds__(128) void attn_decode_combine(
     const float* __restrict__ scratch, u32* __restrict__ o, int nchunks,
-    int used_chunks_stride, int T) {
+    int T) {
   const int bh = blockIdx.x;
   const int used = (T + TPB - 1) / TPB;
-  const float* base = scratch + (u64)bh * used_chunks_stride * (D + 2);
+  const float* base = scratch + (u64)bh * nchunks * (D + 2);
 
   float M = -INFINITY;
   for (int c = 0; c < used; ++c) M = fmaxf(M, base[c * (D + 2) + D]);
@@ -132,7 +151,7 @@ __global__ __launch_bounds__(128) void attn_decode_combine(
     L += __expf(base[c * (D + 2) + D] - M) * base[c * (D + 2) + D + 1];
   const float invL = 1.f / L;
 
-  const int j = threadIdx.x;  // two elems per thread via dword packing
+  const int j = threadIdx.x;
   if (2 * j < D) {
     float acc0 = 0.f, acc1 = 0.f;
     for (int c = 0; c < used; ++c) {
@@ -146,7 +165,7 @@ __global__ __launch_bounds__(128) void attn_decode_combine(
 
 }  // namespace
 
-// scratch must hold b*hq*ceil(T/256)*(d+2) floats
+// scratch must hold b*hq*ceil(T/256)*(d+2) floats (unused when T <= 256)
 hipError_t lfmq_attn_decode(const void* q, const void* kc, const void* vc,
                             void* o, void* scratch, int b, int hq, int hkv,
                             int tmax, int T, int d, float scale,
@@ -156,18 +175,20 @@ hipError_t lfmq_attn_decode(const void* q, const void* kc, const void* vc,
   const int nchunks = (T + TPB - 1) / TPB;
   dim3 gridA(b * hq, nchunks), blockA(TPB);
   dim3 gridB(b * hq), blockB(128);
-  if (d == 128) {
-    hipLaunchKernelGGL((attn_decode_partial<128>), gridA, blockA, 0, stream,
-                       (const u32*)q, (const u32*)kc, (const u32*)vc,
-                       (float*)scratch, b, hq, hkv, tmax, T, scale, nchunks);
-    hipLaunchKernelGGL((attn_decode_combine<128>), gridB, blockB, 0, stream,
-                       (const float*)scratch, (u32*)o, nchunks, nchunks, T);
-  } else {
-    hipLaunchKernelGGL((attn_decode_partial<64>), gridA, blockA, 0, stream,
-                       (const u32*)q, (const u32*)kc, (const u32*)vc,
-                       (float*)scratch, b, hq, hkv, tmax, T, scale, nchunks);
-    hipLaunchKernelGGL((attn_decode_combine<64>), gridB, blockB, 0, stream,
-                       (const float*)scratch, (u32*)o, nchunks, nchunks, T);
-  }
+#define LFMQ_ATTN(DD)                                                         \
+  do {                                                                        \
+    hipLaunchKernelGGL((attn_decode_partial<DD>), gridA, blockA, 0, stream,   \
+                       (const u32*)q, (const u32*)kc, (const u32*)vc,         \
+                       (float*)scratch, (u32*)o, b, hq, hkv, tmax, T, scale,  \
+                       nchunks);                                              \
+    if (nchunks > 1)                                                          \
+      hipLaunchKernelGGL((attn_decode_combine<DD>), gridB, blockB, 0, stream, \
+                         (const float*)scratch, (u32*)o, nchunks, T);         \
+  } while (0)
+  if (d == 128)
+    LFMQ_ATTN(128);
+  else
+    LFMQ_ATTN(64);
+#undef LFMQ_ATTN
   return hipGetLastError();
 }

@@ -67,11 +67,15 @@ __global__ __launch_bounds__(TPB, 1) void dq_gemm_kernel(
 
   const int ksteps = k / BK;
   for (int kt = 0; kt < ksteps; ++kt) {
-    // ---- stage A: x[arow_g][kt*64 + aslot*8 .. +8] -> ldsA[ar][..] ----
-    {
-      const u32 gbyte = (u32)(((u64)arow_g * k + kt * BK + aslot * 8) * 2);
+    // ---- stage A: x[arow_g][kt*64 + slot*8 .. +8] -> ldsA[ar][..] ----
+    // 128 rows x 8 slots of 16B = 1024 slot-tasks over 512 threads: each
+    // thread stages slots aslot and aslot+4 of its row.
+#pragma unroll
+    for (int s2 = 0; s2 < 2; ++s2) {
+      const int slot = aslot + 4 * s2;
+      const u32 gbyte = (u32)(((u64)arow_g * k + kt * BK + slot * 8) * 2);
       u32x4 v = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, gbyte, 0, 0);
-      *(u32x4*)(ldsA + swz((u32)(ar * BK * 2 + aslot * 16), ar)) = v;
+      *(u32x4*)(ldsA + swz((u32)(ar * BK * 2 + slot * 16), ar)) = v;
     }
     // ---- stage B: dequant 16 codes of row bcol_g -> ldsB[bc][..] ------
     {

@@ -3,21 +3,28 @@
 // The decode hot kernel: y[m,n] = x[m,k] · W[n,k]^T,
 // W int4/int8 group-quantized (layout: lfmq/ops/ref.py).  Memory-bound on
 // the packed weight stream — design targets the HBM3E roofline:
-//   - one wave per output column n; lanes split k (lane l owns 32
+//   - each wave owns COLS output columns; lanes split k (lane l owns 32
 //     consecutive k per 2048-wide k-iteration) => every wave-instruction
-//     reads 1 KiB of contiguous packed weight (64 lanes x dwordx4),
-//     perfectly coalesced, via hardware-bounds-checked buffer loads
-//     (OOB returns 0, so ragged k needs no branches).
-//   - weights are loaded non-temporal (aux=2): each line is read once by
-//     one CU (MI355X_MICROARCH.md "nt-weights").
+//     reads 1 KiB of contiguous packed weight, coalesced, via
+//     hardware-bounds-checked buffer loads (OOB returns 0: ragged k
+//     needs no branches).  COLS(=4) weight streams per wave keep 8+
+//     16-B loads in flight per k-iteration — measured 13 µs/call with a
+//     single stream (latency-bound), the multi-stream form is what
+//     reaches the HBM pace.
+//   - weights are loaded non-temporal (aux=2): each line is read once
+//     by one CU (MI355X_MICROARCH.md "nt-weights").
 //   - int4 dequant off the critical path: v_perm assembles bf16 pairs
-//     (128+c) directly from nibbles (csrc/common.h dequant4_pairs), dot
-//     products via v_dot2c_f32_bf16, and the (zero+128)·scale bias is
-//     folded per group:  s·Σ(c−z)x = s·(Σ(128+c)x − (128+z)·Σx).
-//     ~2.3 VALU per weight => VALU stays well under the HBM pace.
+//     (128+c) straight from nibbles (csrc/common.h), dot products via
+//     v_dot2c_f32_bf16, and the (zero+128)·scale bias is folded per
+//     group:  s·Σ(c−z)x = s·(Σ(128+c)x − (128+z)·Σx), with Σx computed
+//     ONCE per k-chunk and shared by all COLS columns.
 //   - activations x are tiny (m·k bf16) and L1/L2-resident; they are
-//     re-read per column from cache, not staged (GEMV row of the
-//     "glds vs register staging" table: LDS round trip is pure overhead).
+//     read once per wave k-iteration from cache, never staged (GEMV row
+//     of the guide's "glds vs register staging" table).
+// GATEUP variant (K9 fused into the K3a epilogue): the wave's columns
+// are {c, c+1} of gate and {inter+c, inter+c+1} of up from one fused
+// gate_up projection; the epilogue writes silu(gate)·up — the separate
+// silu·mul pass and its HBM round-trip disappear.
 // group_size must be a multiple of 32 (one lane-chunk never spans groups).
 #include "common.h"
 
@@ -26,139 +33,212 @@ namespace {
 constexpr int WAVES_PER_WG = 4;
 constexpr u32 ONE2_BF16 = 0x3F803F80u;  // bf16x2 (1.0, 1.0)
 
-template <int M, int BITS>
+template <int M, int BITS, int COLS, bool GATEUP>
 __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
     const u32* __restrict__ qweight, const u32* __restrict__ scales_u16,
     const u32* __restrict__ qzeros, const u32* __restrict__ x,
     u32* __restrict__ y, const u32* __restrict__ bias, int m, int n, int k,
     int g) {
+  static_assert(!GATEUP || COLS == 4, "gateup pairs need COLS=4");
   const int wave = threadIdx.x / LFMQ_WAVE;
   const int lane = threadIdx.x % LFMQ_WAVE;
-  const int col = blockIdx.x * WAVES_PER_WG + wave;
-  if (col >= n) return;
-
-  const int kw_row_bytes = k * BITS / 8;          // packed bytes per row
-  const int G = k / g;
-  const int zw = (G + (32 / BITS) - 1) / (32 / BITS);  // zeros words per row
-
-  auto w_rsrc = make_rsrc((const char*)qweight + (u64)col * kw_row_bytes,
-                          kw_row_bytes);
-  auto x_rsrc = make_rsrc(x, (u32)(m * k * 2));
-  const unsigned short* srow = (const unsigned short*)scales_u16 + (u64)col * G;
-  const u32* zrow = qzeros + (u64)col * zw;
-
-  float acc[M];
+  // column set: GATEUP -> {c, c+1, inter+c, inter+c+1} (inter = n/2);
+  // else COLS consecutive columns.
+  const int cbase = (blockIdx.x * WAVES_PER_WG + wave) * (GATEUP ? 2 : COLS);
+  int col[COLS];
 #pragma unroll
-  for (int mm = 0; mm < M; ++mm) acc[mm] = 0.f;
+  for (int c = 0; c < COLS; ++c)
+    col[c] = GATEUP ? ((c & 1) + cbase + (c >> 1) * (n / 2)) : cbase + c;
+  if (col[0] >= n) return;
+
+  const int kw_row_bytes = k * BITS / 8;
+  const int G = k / g;
+  const int zw = (G + (32 / BITS) - 1) / (32 / BITS);
+
+  __amdgpu_buffer_rsrc_t w_rsrc[COLS];
+#pragma unroll
+  for (int c = 0; c < COLS; ++c) {
+    const int cc = col[c] < n ? col[c] : n - 1;
+    w_rsrc[c] = make_rsrc((const char*)qweight + (u64)cc * kw_row_bytes,
+                          kw_row_bytes);
+  }
+  auto x_rsrc = make_rsrc(x, (u32)(m * k * 2));
+  const unsigned short* srow = (const unsigned short*)scales_u16;
+
+  float acc[COLS][M];
+#pragma unroll
+  for (int c = 0; c < COLS; ++c)
+#pragma unroll
+    for (int mm = 0; mm < M; ++mm) acc[c][mm] = 0.f;
 
   const int kiters = (k + 2047) / 2048;
   for (int it = 0; it < kiters; ++it) {
-    const int k0 = it * 2048 + lane * 32;  // this lane's first k
-    // group metadata (one group per lane-chunk: g % 32 == 0)
-    float s = 0.f, zb = 0.f;
-    if (k0 < k) {
-      const int gi = k0 / g;
-      s = bf16_bits_to_f32(srow[gi]);
-      const u32 zword = zrow[gi / (32 / BITS)];
+    const int k0 = it * 2048 + lane * 32;
+    // per-column group metadata (single group per 32-wide lane chunk)
+    float s[COLS], zb[COLS];
+    const int gi = k0 < k ? k0 / g : 0;
+#pragma unroll
+    for (int c = 0; c < COLS; ++c) {
+      const int cc = col[c] < n ? col[c] : n - 1;
+      s[c] = k0 < k ? bf16_bits_to_f32(srow[(u64)cc * G + gi]) : 0.f;
+      const u32 zword = qzeros[(u64)cc * zw + gi / (32 / BITS)];
       const u32 z = BITS == 4 ? unpack4(zword, gi & 7) : unpack8(zword, gi & 3);
-      zb = BITS == 4 ? (float)(z + 128u) : (float)z;
+      zb[c] = BITS == 4 ? (float)(z + 128u) : (float)z;
     }
 
     if (BITS == 4) {
       const u32 wbyte = (u32)(it * 1024 + lane * 16);
-      u32x4 w = __builtin_amdgcn_raw_buffer_load_b128(w_rsrc, wbyte, 0, /*nt*/ 2);
-      DequantPairs4 dq[4];
+      u32x4 w[COLS];
 #pragma unroll
-      for (int j = 0; j < 4; ++j) dq[j] = dequant4_pairs(w[j]);
+      for (int c = 0; c < COLS; ++c)
+        w[c] = __builtin_amdgcn_raw_buffer_load_b128(w_rsrc[c], wbyte, 0, 2);
 #pragma unroll
       for (int mm = 0; mm < M; ++mm) {
         if (mm >= m) break;
         const u32 xbyte = (u32)(mm * k * 2 + it * 4096 + lane * 64);
-        float cx = 0.f, xs = 0.f;
+        u32x4 xq0 = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte, 0, 0);
+        u32x4 xq1 = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte + 16, 0, 0);
+        u32x4 xq2 = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte + 32, 0, 0);
+        u32x4 xq3 = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte + 48, 0, 0);
+        float xs = 0.f;  // Σx over this lane chunk: shared by all columns
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-          u32x4 xq = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte + 16 * j, 0, 0);
-          cx = dot2_bf16(dq[j].p01, xq[0], cx);
-          cx = dot2_bf16(dq[j].p23, xq[1], cx);
-          cx = dot2_bf16(dq[j].p45, xq[2], cx);
-          cx = dot2_bf16(dq[j].p67, xq[3], cx);
+          const u32x4 xq = j == 0 ? xq0 : j == 1 ? xq1 : j == 2 ? xq2 : xq3;
           xs = dot2_bf16(ONE2_BF16, xq[0], xs);
           xs = dot2_bf16(ONE2_BF16, xq[1], xs);
           xs = dot2_bf16(ONE2_BF16, xq[2], xs);
           xs = dot2_bf16(ONE2_BF16, xq[3], xs);
         }
-        acc[mm] = fmaf(s, cx, acc[mm]);
-        acc[mm] = fmaf(-s * zb, xs, acc[mm]);
+#pragma unroll
+        for (int c = 0; c < COLS; ++c) {
+          float cx = 0.f;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            const u32x4 xq = j == 0 ? xq0 : j == 1 ? xq1 : j == 2 ? xq2 : xq3;
+            const DequantPairs4 dq = dequant4_pairs(w[c][j]);
+            cx = dot2_bf16(dq.p01, xq[0], cx);
+            cx = dot2_bf16(dq.p23, xq[1], cx);
+            cx = dot2_bf16(dq.p45, xq[2], cx);
+            cx = dot2_bf16(dq.p67, xq[3], cx);
+          }
+          acc[c][mm] = fmaf(s[c], cx, acc[c][mm]);
+          acc[c][mm] = fmaf(-s[c] * zb[c], xs, acc[c][mm]);
+        }
       }
-    } else {  // BITS == 8: byte codes via v_cvt_f32_ubyte
+    } else {  // BITS == 8
       const u32 wbyte = (u32)(it * 2048 + lane * 32);
-      u32x4 w0 = __builtin_amdgcn_raw_buffer_load_b128(w_rsrc, wbyte, 0, 2);
-      u32x4 w1 = __builtin_amdgcn_raw_buffer_load_b128(w_rsrc, wbyte + 16, 0, 2);
+      u32x4 w0[COLS], w1[COLS];
+#pragma unroll
+      for (int c = 0; c < COLS; ++c) {
+        w0[c] = __builtin_amdgcn_raw_buffer_load_b128(w_rsrc[c], wbyte, 0, 2);
+        w1[c] = __builtin_amdgcn_raw_buffer_load_b128(w_rsrc[c], wbyte + 16, 0, 2);
+      }
 #pragma unroll
       for (int mm = 0; mm < M; ++mm) {
         if (mm >= m) break;
         const u32 xbyte = (u32)(mm * k * 2 + it * 4096 + lane * 64);
-        float cx = 0.f, xs = 0.f;
+        float xv[32];
 #pragma unroll
-        for (int half = 0; half < 2; ++half) {
-          u32x4 wv = half ? w1 : w0;
-          u32x4 xa = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte + 32 * half, 0, 0);
-          u32x4 xb = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte + 32 * half + 16, 0, 0);
+        for (int h = 0; h < 4; ++h) {
+          u32x4 xq = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte + 16 * h, 0, 0);
 #pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            u32 xd0 = (j < 2) ? xa[2 * j] : xb[2 * (j - 2)];
-            u32 xd1 = (j < 2) ? xa[2 * j + 1] : xb[2 * (j - 2) + 1];
-            float x0, x1, x2, x3;
-            bf16x2_to_f32(xd0, x0, x1);
-            bf16x2_to_f32(xd1, x2, x3);
-            // (float)(byte) lowers to v_cvt_f32_ubyte0..3 on gfx950
-            cx = fmaf((float)(wv[j] & 0xFFu), x0, cx);
-            cx = fmaf((float)((wv[j] >> 8) & 0xFFu), x1, cx);
-            cx = fmaf((float)((wv[j] >> 16) & 0xFFu), x2, cx);
-            cx = fmaf((float)(wv[j] >> 24), x3, cx);
-            xs += x0 + x1 + x2 + x3;
-          }
+          for (int j = 0; j < 4; ++j)
+            bf16x2_to_f32(xq[j], xv[h * 8 + 2 * j], xv[h * 8 + 2 * j + 1]);
         }
-        acc[mm] = fmaf(s, cx, acc[mm]);
-        acc[mm] = fmaf(-s * zb, xs, acc[mm]);
+        float xs = 0.f;
+#pragma unroll
+        for (int i = 0; i < 32; ++i) xs += xv[i];
+#pragma unroll
+        for (int c = 0; c < COLS; ++c) {
+          float cx = 0.f;
+#pragma unroll
+          for (int half = 0; half < 2; ++half) {
+            const u32x4 wv = half ? w1[c] : w0[c];
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              const int e = half * 16 + j * 4;
+              cx = fmaf((float)(wv[j] & 0xFFu), xv[e], cx);
+              cx = fmaf((float)((wv[j] >> 8) & 0xFFu), xv[e + 1], cx);
+              cx = fmaf((float)((wv[j] >> 16) & 0xFFu), xv[e + 2], cx);
+              cx = fmaf((float)(wv[j] >> 24), xv[e + 3], cx);
+            }
+          }
+          acc[c][mm] = fmaf(s[c], cx, acc[c][mm]);
+          acc[c][mm] = fmaf(-s[c] * zb[c], xs, acc[c][mm]);
+        }
       }
     }
   }
 
-  // wave-reduce each row's partial and store
-  float b = 0.f;
-  if (bias != nullptr && lane == 0) {
-    const unsigned short* brow = (const unsigned short*)bias;
-    b = bf16_bits_to_f32(brow[col]);
-  }
+  // wave-reduce and store
 #pragma unroll
   for (int mm = 0; mm < M; ++mm) {
     if (mm >= m) break;
-    float v = wave_reduce_sum(acc[mm]);
+    float v[COLS];
+#pragma unroll
+    for (int c = 0; c < COLS; ++c) {
+      v[c] = wave_reduce_sum(acc[c][mm]);
+      if (bias != nullptr && col[c] < n)
+        v[c] += bf16_bits_to_f32(((const unsigned short*)bias)[col[c]]);
+    }
     if (lane == 0) {
       unsigned short* yrow = (unsigned short*)y;
-      yrow[(u64)mm * n + col] = (unsigned short)f32_to_bf16_bits_rne(v + b);
+      if (GATEUP) {
+        const int inter = n / 2;
+#pragma unroll
+        for (int p = 0; p < 2; ++p) {  // pairs (gate=v[p], up=v[p+2])
+          const int oc = cbase + p;
+          if (oc < inter) {
+            const float gt = v[p];
+            const float sil = gt / (1.f + __expf(-gt));
+            yrow[(u64)mm * inter + oc] =
+                (unsigned short)f32_to_bf16_bits_rne(sil * v[p + 2]);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int c = 0; c < COLS; ++c)
+          if (col[c] < n)
+            yrow[(u64)mm * n + col[c]] =
+                (unsigned short)f32_to_bf16_bits_rne(v[c]);
+      }
     }
   }
 }
 
-template <int BITS>
-hipError_t launch_bits(const void* qweight, const void* scales, const void* qzeros,
-                       const void* x, void* y, const void* bias, int m, int n,
-                       int k, int g, hipStream_t stream) {
-  dim3 grid(ceil_div_u32(n, WAVES_PER_WG));
+template <int BITS, int COLS, bool GATEUP, int M>
+void launch_one(dim3 grid, dim3 block, hipStream_t stream, const void* qweight,
+                const void* scales, const void* qzeros, const void* x, void* y,
+                const void* bias, int m, int n, int k, int g) {
+  hipLaunchKernelGGL((dq_gemv_kernel<M, BITS, COLS, GATEUP>), grid, block, 0,
+                     stream, (const u32*)qweight, (const u32*)scales,
+                     (const u32*)qzeros, (const u32*)x, (u32*)y,
+                     (const u32*)bias, m, n, k, g);
+}
+
+template <int BITS, bool GATEUP>
+hipError_t launch_bits(const void* qweight, const void* scales,
+                       const void* qzeros, const void* x, void* y,
+                       const void* bias, int m, int n, int k, int g,
+                       hipStream_t stream) {
   dim3 block(WAVES_PER_WG * LFMQ_WAVE);
-#define LFMQ_CASE(MM)                                                          \
-  hipLaunchKernelGGL((dq_gemv_kernel<MM, BITS>), grid, block, 0, stream,       \
-                     (const u32*)qweight, (const u32*)scales,                  \
-                     (const u32*)qzeros, (const u32*)x, (u32*)y,               \
-                     (const u32*)bias, m, n, k, g)
-  if (m <= 1) LFMQ_CASE(1);
-  else if (m <= 2) LFMQ_CASE(2);
-  else if (m <= 4) LFMQ_CASE(4);
-  else if (m <= 8) LFMQ_CASE(8);
-  else LFMQ_CASE(16);
-#undef LFMQ_CASE
+  if (GATEUP || m <= 4) {
+    // 4 columns per wave (pairs for gateup)
+    const int per_wg = WAVES_PER_WG * (GATEUP ? 2 : 4);
+    dim3 grid(ceil_div_u32(GATEUP ? n / 2 : n, per_wg));
+    if (m <= 1)
+      launch_one<BITS, 4, GATEUP, 1>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+    else if (m <= 2)
+      launch_one<BITS, 4, GATEUP, 2>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+    else
+      launch_one<BITS, 4, GATEUP, 4>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+  } else {
+    dim3 grid(ceil_div_u32(n, WAVES_PER_WG));
+    if (m <= 8)
+      launch_one<BITS, 1, false, 8>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+    else
+      launch_one<BITS, 1, false, 16>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+  }
   return hipGetLastError();
 }
 
@@ -170,8 +250,23 @@ hipError_t lfmq_dq_gemv(const void* qweight, const void* scales,
                         hipStream_t stream) {
   if (m > 16 || k % 8 != 0 || g % 32 != 0 || g > k) return hipErrorInvalidValue;
   if (bits == 4)
-    return launch_bits<4>(qweight, scales, qzeros, x, y, bias, m, n, k, g, stream);
+    return launch_bits<4, false>(qweight, scales, qzeros, x, y, bias, m, n, k, g, stream);
   if (bits == 8)
-    return launch_bits<8>(qweight, scales, qzeros, x, y, bias, m, n, k, g, stream);
+    return launch_bits<8, false>(qweight, scales, qzeros, x, y, bias, m, n, k, g, stream);
+  return hipErrorInvalidValue;
+}
+
+// fused gate_up GEMV + silu·mul epilogue: W is the fused gate_up
+// projection [n = 2*inter, k]; output is [m, inter] = silu(gate)*up.
+hipError_t lfmq_dq_gemv_gateup(const void* qweight, const void* scales,
+                               const void* qzeros, const void* x, void* y,
+                               int m, int n, int k, int g, int bits,
+                               hipStream_t stream) {
+  if (m > 4 || k % 8 != 0 || g % 32 != 0 || g > k || n % 2)
+    return hipErrorInvalidValue;
+  if (bits == 4)
+    return launch_bits<4, true>(qweight, scales, qzeros, x, y, nullptr, m, n, k, g, stream);
+  if (bits == 8)
+    return launch_bits<8, true>(qweight, scales, qzeros, x, y, nullptr, m, n, k, g, stream);
   return hipErrorInvalidValue;
 }

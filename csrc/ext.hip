@@ -27,6 +27,14 @@ hipError_t lfmq_attn_decode(const void*, const void*, const void*, void*,
                             void*, int, int, int, int, int, int, float,
                             hipStream_t);
 hipError_t lfmq_hessian_accum(const void*, void*, int, int, hipStream_t);
+hipError_t lfmq_dq_gemv_gateup(const void*, const void*, const void*,
+                               const void*, void*, int, int, int, int, int,
+                               hipStream_t);
+hipError_t lfmq_qkv_rope_append(const void*, void*, void*, void*, const void*,
+                                const void*, int, int, int, int, int, int,
+                                int, hipStream_t);
+hipError_t lfmq_argmax_rows(const void*, void*, void*, void*, int, int,
+                            hipStream_t);
 
 namespace {
 
@@ -208,6 +216,52 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   return o;
 }
 
+torch::Tensor dq_gemv_gateup(torch::Tensor x, torch::Tensor qweight,
+                             torch::Tensor scales, torch::Tensor qzeros,
+                             long bits, long g) {
+  check_t(x, torch::kBFloat16, "x");
+  check_t(qweight, torch::kInt32, "qweight");
+  const int m = (int)x.size(0), k = (int)x.size(1);
+  const int n = (int)qweight.size(0);
+  auto y = torch::empty({m, n / 2}, x.options());
+  CHECK_HIP(lfmq_dq_gemv_gateup(qweight.data_ptr(), scales.data_ptr(),
+                                qzeros.data_ptr(), x.data_ptr(), y.data_ptr(),
+                                m, n, k, (int)g, (int)bits, stream()));
+  return y;
+}
+
+torch::Tensor qkv_rope_append(torch::Tensor qkv, torch::Tensor kc,
+                              torch::Tensor vc, torch::Tensor positions,
+                              torch::Tensor inv_freq, long hq, long pos0) {
+  check_t(qkv, torch::kBFloat16, "qkv");
+  check_t(kc, torch::kBFloat16, "k_cache");
+  check_t(vc, torch::kBFloat16, "v_cache");
+  check_t(positions, torch::kInt32, "positions");
+  check_t(inv_freq, torch::kFloat32, "inv_freq");
+  const int b = (int)kc.size(0), hkv = (int)kc.size(1);
+  const int tmax = (int)kc.size(2), d = (int)kc.size(3);
+  const int s = (int)(qkv.numel() / ((hq + 2 * hkv) * d)) / b;
+  auto q_out = torch::empty({b, s, hq, (long)d}, qkv.options());
+  CHECK_HIP(lfmq_qkv_rope_append(qkv.data_ptr(), q_out.data_ptr(),
+                                 kc.data_ptr(), vc.data_ptr(),
+                                 positions.data_ptr(), inv_freq.data_ptr(), b,
+                                 s, (int)hq, hkv, d, tmax, (int)pos0,
+                                 stream()));
+  return q_out;
+}
+
+torch::Tensor greedy_sample(torch::Tensor logits) {
+  check_t(logits, torch::kBFloat16, "logits");
+  TORCH_CHECK(logits.dim() == 2, "logits must be [b, vocab]");
+  const int b = (int)logits.size(0), vocab = (int)logits.size(1);
+  auto out = torch::empty({b}, logits.options().dtype(torch::kInt64));
+  auto sf = torch::empty({b * 64}, logits.options().dtype(torch::kFloat32));
+  auto si = torch::empty({b * 64}, logits.options().dtype(torch::kInt32));
+  CHECK_HIP(lfmq_argmax_rows(logits.data_ptr(), out.data_ptr(), sf.data_ptr(),
+                             si.data_ptr(), b, vocab, stream()));
+  return out;
+}
+
 void hessian_accum(torch::Tensor h, torch::Tensor x) {
   check_t(h, torch::kFloat32, "H");
   check_t(x, torch::kBFloat16, "x");
@@ -231,4 +285,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kv_append", &kv_append, "KV cache append");
   m.def("attn_decode", &attn_decode, "GQA decode attention");
   m.def("hessian_accum", &hessian_accum, "H += 2 X^T X");
+  m.def("dq_gemv_gateup", &dq_gemv_gateup, "fused gate_up GEMV + silu*mul");
+  m.def("qkv_rope_append", &qkv_rope_append, "fused rope + KV append");
+  m.def("greedy_sample", &greedy_sample, "row argmax");
 }

@@ -97,12 +97,30 @@ class LlamaAttention(nn.Module):
         b, s, _ = x.shape
         hq, hkv, d = self.n_heads, self.n_kv_heads, self.head_dim
         qkv = self.qkv_proj(x)
+        pos0 = seq_len_after - s
+        if x.is_cuda:
+            # fused rope + KV append straight off the qkv buffer (one launch)
+            q = F.qkv_rope_append(qkv, cache_k, cache_v, positions, pos0, hq,
+                                  self.cfg.rope_theta)
+            if s == 1:
+                o = F.attn_decode(q.view(b, hq, 1, d), cache_k, cache_v,
+                                  seq_len_after, self.scale)
+                o = o.view(b, s, hq * d)
+            else:
+                qh = q.transpose(1, 2)
+                kc = cache_k[:, :, pos0:seq_len_after]
+                vc = cache_v[:, :, pos0:seq_len_after]
+                if pos0 == 0:
+                    o = F.attn_prefill(qh, kc, vc, causal=True, scale=self.scale)
+                else:
+                    o = self._chunk_attn(qh, cache_k, cache_v, pos0, seq_len_after)
+                o = o.transpose(1, 2).reshape(b, s, hq * d)
+            return self.o_proj(o)
         q, k, v = qkv.split([hq * d, hkv * d, hkv * d], dim=-1)
         q = q.reshape(b, s, hq, d)
         k = k.reshape(b, s, hkv, d)
         v = v.reshape(b, s, hkv, d)
         q, k = F.apply_rope(q, k, positions, self.cfg.rope_theta)
-        pos0 = seq_len_after - s
         F.kv_append(cache_k, cache_v, k, v, pos0)
         qh = q.transpose(1, 2)  # [b, hq, s, d]
         if s == 1:
@@ -140,9 +158,17 @@ class LlamaMLP(nn.Module):
         self.inter = cfg.intermediate_size
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        gu = self.gate_up_proj(x)
-        gate, up = gu.split([self.inter, self.inter], dim=-1)
-        return self.down_proj(F.silu_mul(gate, up))
+        from .qlinear import QLinear
+        gp = self.gate_up_proj
+        m = x.numel() // x.shape[-1]
+        if (x.is_cuda and m <= 4 and type(gp).forward is QLinear.forward
+                and isinstance(gp, QLinear)):
+            # fused gate_up GEMV + silu·mul epilogue (no separate pass)
+            h = F.dq_gemv_gateup(x, gp.qweight, gp.scales, gp.qzeros,
+                                 gp.bits, gp.group_size)
+            return self.down_proj(h)
+        gu = gp(x)
+        return self.down_proj(F.silu_mul_gateup(gu, self.inter))
 
 
 class LlamaDecoderLayer(nn.Module):
@@ -153,13 +179,24 @@ class LlamaDecoderLayer(nn.Module):
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_eps, dtype)
         self.mlp = LlamaMLP(cfg, dtype)
 
-    def forward(self, x, cache_k, cache_v, positions, seq_len_after):
-        h = self.input_layernorm(x)
+    def forward(self, x, cache_k, cache_v, positions, seq_len_after,
+                residual=None):
+        """Fused-residual form: returns (mlp_out, residual_after_attn) so
+        the NEXT norm folds the pending add (one rmsnorm_residual kernel
+        instead of add + rmsnorm).  residual=None => x is the raw input."""
+        if residual is None:
+            residual = x
+            h = self.input_layernorm(x)
+        else:
+            h, residual = F.rmsnorm_residual(x, residual,
+                                             self.input_layernorm.weight,
+                                             self.input_layernorm.eps)
         h = self.self_attn(h, cache_k, cache_v, positions, seq_len_after)
-        x = x + h
-        h = self.post_attention_layernorm(x)
-        x = x + self.mlp(h)
-        return x
+        h, residual = F.rmsnorm_residual(h, residual,
+                                         self.post_attention_layernorm.weight,
+                                         self.post_attention_layernorm.eps)
+        h = self.mlp(h)
+        return h, residual
 
 
 class LlamaForCausalLM(nn.Module):
@@ -201,9 +238,11 @@ class LlamaForCausalLM(nn.Module):
             positions = torch.arange(pos0, pos0 + s, device=input_ids.device)
         seq_after = pos0 + s
         x = self.embed_tokens(input_ids)
+        residual = None
         for i, layer in enumerate(self.layers):
-            x = layer(x, cache.k[i], cache.v[i], positions, seq_after)
-        x = self.norm(x)
+            x, residual = layer(x, cache.k[i], cache.v[i], positions,
+                                seq_after, residual)
+        x = self.norm(x, residual=residual)
         logits = self.lm_head(x)
         cache.advance(s)
         return logits
@@ -216,10 +255,10 @@ class LlamaForCausalLM(nn.Module):
             cache = self.make_cache(b, max_seq=s + max_new_tokens + 8)
         logits = self.forward(input_ids, cache)
         out = [input_ids]
-        tok = logits[:, -1:].argmax(dim=-1)
+        tok = F.greedy_sample(logits[:, -1]).unsqueeze(1)
         for _ in range(max_new_tokens - 1):
             out.append(tok)
             logits = self.forward(tok, cache)
-            tok = logits[:, -1:].argmax(dim=-1)
+            tok = F.greedy_sample(logits[:, -1]).unsqueeze(1)
         out.append(tok)
         return torch.cat(out, dim=1)

@@ -69,7 +69,16 @@ def dq_linear(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,
 
 def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     if gate.is_cuda:
-        return hip_ext().silu_mul(gate, up)
+        return hip_ext().silu_mul(gate.contiguous(), up.contiguous())
+    return ref.silu_mul(gate, up)
+
+
+def silu_mul_gateup(gate_up: torch.Tensor, inter: int) -> torch.Tensor:
+    """silu(gu[..., :inter]) * gu[..., inter:] without materializing the
+    split halves (one fused kernel on GPU)."""
+    if gate_up.is_cuda:
+        return hip_ext().silu_mul_fused(gate_up.contiguous())
+    gate, up = gate_up.split([inter, inter], dim=-1)
     return ref.silu_mul(gate, up)
 
 
@@ -97,6 +106,34 @@ def attn_prefill(q, k, v, causal: bool = True, scale: Optional[float] = None):
         p = torch.softmax(sc, dim=-1).to(q.dtype)
         return torch.matmul(p, vf)
     return ref.attn_prefill(q, k, v, causal, scale)
+
+
+def qkv_rope_append(qkv: torch.Tensor, k_cache: torch.Tensor,
+                    v_cache: torch.Tensor, positions: torch.Tensor,
+                    pos0: int, hq: int, theta: float = 500000.0) -> torch.Tensor:
+    """GPU-only fused decode epilogue of the qkv projection: rope(q),
+    rope(k)+append, append(v) in one launch.  qkv: [b, s, (hq+2hkv)*d];
+    returns q [b, s, hq, d] (== [b, hq, 1, d] layout at s == 1)."""
+    d = k_cache.shape[-1]
+    return hip_ext().qkv_rope_append(qkv.contiguous(), k_cache, v_cache,
+                                     positions.to(torch.int32),
+                                     _inv_freq(d, theta, qkv.device), hq, pos0)
+
+
+def dq_gemv_gateup(x: torch.Tensor, qweight: torch.Tensor,
+                   scales: torch.Tensor, qzeros: torch.Tensor, bits: int,
+                   group_size: int) -> torch.Tensor:
+    """GPU-only fused gate_up GEMV + silu·mul epilogue (m <= 4)."""
+    x2 = x.reshape(-1, x.shape[-1])
+    y = hip_ext().dq_gemv_gateup(x2, qweight, scales, qzeros, bits, group_size)
+    return y.reshape(*x.shape[:-1], y.shape[-1])
+
+
+def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
+    """argmax over the last dim of [b, vocab] logits -> int64 [b]."""
+    if logits.is_cuda:
+        return hip_ext().greedy_sample(logits.contiguous())
+    return logits.argmax(dim=-1)
 
 
 def kv_append(k_cache, v_cache, k_new, v_new, pos: int) -> None:

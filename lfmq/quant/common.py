@@ -51,7 +51,8 @@ def make_block_runner(model: nn.Module):
             ck = torch.zeros(b, hkv, s, d, device=x.device, dtype=x.dtype)
             cv = torch.zeros_like(ck)
             pos = torch.arange(s, device=x.device)
-            return block(x, ck, cv, pos, s)
+            h, res = block(x, ck, cv, pos, s)  # fused-residual form
+            return (h.float() + res.float()).to(h.dtype)
         return run
     if arch == "opt":
         nh = model.cfg.n_heads

@@ -22,6 +22,8 @@ SOURCES = [
     "csrc/dq_gemm.hip",
     "csrc/hessian.hip",
     "csrc/attn_decode.hip",
+    "csrc/qkv_rope_append.hip",
+    "csrc/sample.hip",
 ]
 
 setup(

@@ -198,3 +198,49 @@ def test_attn_decode(b, T, d):
     o = ext().attn_decode(q.to(DEV), kc.to(DEV), vc.to(DEV), T, scale)
     o_ref = ref.attn_decode(q.float(), kc.float(), vc.float(), T, scale)
     assert rel_err(o, o_ref) < 2e-2, f"rel={rel_err(o, o_ref)}"
+
+
+# ------------------------------------------------- fused decode kernels
+def test_dq_gemv_gateup():
+    torch.manual_seed(7)
+    inter, k, g = 256, 512, 128
+    w = torch.randn(2 * inter, k)
+    codes, scales, zeros = ref.quant_groupwise(w, 4, g, False)
+    qw, qz = ref.pack_codes(codes, 4), ref.pack_zeros(zeros, 4)
+    x = torch.randn(1, k).to(torch.bfloat16)
+    gu = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(), qz, 4, g)
+    y_ref = ref.silu_mul(gu[:, :inter].contiguous(), gu[:, inter:].contiguous())
+    y = ext().dq_gemv_gateup(x.to(DEV), qw.to(DEV),
+                             scales.to(torch.bfloat16).to(DEV), qz.to(DEV), 4, g)
+    assert y.shape == (1, inter)
+    assert rel_err(y, y_ref) < 2e-2
+
+
+def test_qkv_rope_append():
+    b, s, hq, hkv, d, tmax, pos0 = 2, 3, 8, 2, 128, 32, 5
+    torch.manual_seed(9)
+    qkv = torch.randn(b, s, (hq + 2 * hkv) * d).to(torch.bfloat16)
+    kc = torch.zeros(b, hkv, tmax, d).to(torch.bfloat16).to(DEV)
+    vc = torch.zeros_like(kc)
+    pos = torch.arange(pos0, pos0 + s)
+    inv = ref.rope_freqs(d, 500000.0).to(DEV)
+    q = ext().qkv_rope_append(qkv.to(DEV), kc, vc, pos.to(torch.int32).to(DEV),
+                              inv, hq, pos0)
+    # reference
+    qr, kr, vr = qkv.split([hq * d, hkv * d, hkv * d], dim=-1)
+    q_ref = ref.apply_rope(qr.reshape(b, s, hq, d).float(), pos)
+    k_ref = ref.apply_rope(kr.reshape(b, s, hkv, d).float(), pos)
+    v_ref = vr.reshape(b, s, hkv, d)
+    assert rel_err(q, q_ref) < 2e-2
+    assert rel_err(kc[:, :, pos0:pos0 + s], k_ref.transpose(1, 2)) < 2e-2
+    assert torch.equal(vc[:, :, pos0:pos0 + s].cpu(), v_ref.transpose(1, 2))
+    assert kc[:, :, :pos0].abs().sum().item() == 0
+
+
+def test_greedy_sample():
+    torch.manual_seed(11)
+    for b, vocab in [(1, 128256), (4, 512), (2, 50272)]:
+        logits = torch.randn(b, vocab).to(torch.bfloat16)
+        got = ext().greedy_sample(logits.to(DEV))
+        want = logits.float().argmax(dim=-1)
+        assert torch.equal(got.cpu(), want), (b, vocab)
