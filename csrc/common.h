@@ -92,6 +92,21 @@ __device__ __forceinline__ __amdgpu_buffer_rsrc_t make_rsrc(const void* p, u32 b
                                            /*swizzle fmt*/ 0x00020000u);
 }
 
+// Wave-uniform variant (guide T20): a descriptor built from a value the
+// compiler cannot PROVE uniform (e.g. a per-wave row pointer) gets every
+// buffer op wrapped in a ~10-instruction readfirstlane/saveexec waterfall
+// loop that serializes consecutive loads — measured 13 µs/call on the
+// decode GEMV.  readfirstlane the inputs once; descriptor becomes SGPRs.
+__device__ __forceinline__ __amdgpu_buffer_rsrc_t make_rsrc_uniform(
+    const void* p, u32 bytes) {
+  const u64 a = (u64)p;
+  const u32 lo = __builtin_amdgcn_readfirstlane((u32)a);
+  const u32 hi = __builtin_amdgcn_readfirstlane((u32)(a >> 32));
+  const u32 nb = __builtin_amdgcn_readfirstlane(bytes);
+  return __builtin_amdgcn_make_buffer_rsrc((void*)(((u64)hi << 32) | lo), 0,
+                                           nb, 0x00020000u);
+}
+
 // XCD-aware bijective block-id swizzle (guide §5: nwg need not divide 8)
 __device__ __forceinline__ u32 xcd_swizzle(u32 bid, u32 nwg) {
   const u32 nx = 8;

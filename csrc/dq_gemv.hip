@@ -2,29 +2,32 @@
 //
 // The decode hot kernel: y[m,n] = x[m,k] · W[n,k]^T,
 // W int4/int8 group-quantized (layout: lfmq/ops/ref.py).  Memory-bound on
-// the packed weight stream — design targets the HBM3E roofline:
-//   - each wave owns COLS output columns; lanes split k (lane l owns 32
-//     consecutive k per 2048-wide k-iteration) => every wave-instruction
-//     reads 1 KiB of contiguous packed weight, coalesced, via
-//     hardware-bounds-checked buffer loads (OOB returns 0: ragged k
-//     needs no branches).  COLS(=4) weight streams per wave keep 8+
-//     16-B loads in flight per k-iteration — measured 13 µs/call with a
-//     single stream (latency-bound), the multi-stream form is what
-//     reaches the HBM pace.
-//   - weights are loaded non-temporal (aux=2): each line is read once
-//     by one CU (MI355X_MICROARCH.md "nt-weights").
+// the packed weight stream; measured design history on MI355X:
+//   - per-wave buffer descriptors MUST be readfirstlane-uniform
+//     (csrc/common.h make_rsrc_uniform) or hipcc waterfalls every load
+//     (T20): 13 us/call -> 6.5.
+//   - a raw-read probe of this exact pattern (csrc/probe.hip) sustains
+//     6.3-6.7 TB/s, so the remaining gap was the per-iteration SCATTERED
+//     2-byte scale + zero loads (8 TA-heavy lane-scattered loads per
+//     k-iteration).  They are hoisted: each column's whole scale/zero
+//     row is preloaded ONCE, coalesced, into per-lane registers (lane g
+//     holds group g) and fetched in-loop by a single cross-lane shuffle.
+//   - each wave owns COLS(=4) columns; lanes split k (lane l owns 32
+//     consecutive k per 2048-wide k-iteration): every wave-instruction
+//     reads 1 KiB of contiguous packed weight via hardware-bounds-
+//     checked buffer loads (OOB -> 0: ragged k needs no branches),
+//     non-temporal (each line is read once by one CU).
 //   - int4 dequant off the critical path: v_perm assembles bf16 pairs
-//     (128+c) straight from nibbles (csrc/common.h), dot products via
-//     v_dot2c_f32_bf16, and the (zero+128)·scale bias is folded per
-//     group:  s·Σ(c−z)x = s·(Σ(128+c)x − (128+z)·Σx), with Σx computed
-//     ONCE per k-chunk and shared by all COLS columns.
-//   - activations x are tiny (m·k bf16) and L1/L2-resident; they are
-//     read once per wave k-iteration from cache, never staged (GEMV row
-//     of the guide's "glds vs register staging" table).
+//     (128+c) straight from nibbles, dots via v_dot2c_f32_bf16, and the
+//     (zero+128)·scale bias is folded per group:
+//       s·Σ(c−z)x = s·(Σ(128+c)x − (128+z)·Σx),
+//     Σx computed once per k-chunk and shared by all COLS columns.
+//   - activations x are tiny and L1/L2-resident: read per wave
+//     k-iteration from cache, never staged (GEMV row of the guide's
+//     "glds vs register staging" table).
 // GATEUP variant (K9 fused into the K3a epilogue): the wave's columns
-// are {c, c+1} of gate and {inter+c, inter+c+1} of up from one fused
-// gate_up projection; the epilogue writes silu(gate)·up — the separate
-// silu·mul pass and its HBM round-trip disappear.
+// are {c, c+1} of gate and {inter+c, inter+c+1} of up of one fused
+// gate_up projection; the epilogue writes silu(gate)·up directly.
 // group_size must be a multiple of 32 (one lane-chunk never spans groups).
 #include "common.h"
 
@@ -42,8 +45,6 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
   static_assert(!GATEUP || COLS == 4, "gateup pairs need COLS=4");
   const int wave = threadIdx.x / LFMQ_WAVE;
   const int lane = threadIdx.x % LFMQ_WAVE;
-  // column set: GATEUP -> {c, c+1, inter+c, inter+c+1} (inter = n/2);
-  // else COLS consecutive columns.
   const int cbase = (blockIdx.x * WAVES_PER_WG + wave) * (GATEUP ? 2 : COLS);
   int col[COLS];
 #pragma unroll
@@ -53,17 +54,51 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
 
   const int kw_row_bytes = k * BITS / 8;
   const int G = k / g;
-  const int zw = (G + (32 / BITS) - 1) / (32 / BITS);
+  const int per_z = 32 / BITS;
+  const int zw = (G + per_z - 1) / per_z;
 
   __amdgpu_buffer_rsrc_t w_rsrc[COLS];
 #pragma unroll
   for (int c = 0; c < COLS; ++c) {
     const int cc = col[c] < n ? col[c] : n - 1;
-    w_rsrc[c] = make_rsrc((const char*)qweight + (u64)cc * kw_row_bytes,
-                          kw_row_bytes);
+    w_rsrc[c] = make_rsrc_uniform((const char*)qweight + (u64)cc * kw_row_bytes,
+                                  kw_row_bytes);
   }
   auto x_rsrc = make_rsrc(x, (u32)(m * k * 2));
-  const unsigned short* srow = (const unsigned short*)scales_u16;
+  auto s_rsrc = make_rsrc(scales_u16, (u32)((u64)n * G * 2));
+  auto z_rsrc = make_rsrc(qzeros, (u32)((u64)n * zw * 4));
+
+  // ---- hoisted scale/zero preload: lane gg holds (s, s·zbias) of
+  // group gg (and gg+64 in the hi pair when G > 64) --------------------
+  float s_lo[COLS], sz_lo[COLS], s_hi[COLS], sz_hi[COLS];
+#pragma unroll
+  for (int c = 0; c < COLS; ++c) {
+    const int cc = col[c] < n ? col[c] : n - 1;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      if (half == 1 && G <= 64) {
+        s_hi[c] = 0.f;
+        sz_hi[c] = 0.f;
+        continue;
+      }
+      const int gg = lane + 64 * half;
+      const u64 sidx = (u64)cc * G + gg;
+      const u32 sraw = __builtin_amdgcn_raw_buffer_load_b32(
+          s_rsrc, (u32)(sidx * 2) & ~3u, 0, 0);
+      const float s = bf16_bits_to_f32((sidx & 1) ? (sraw >> 16) : (sraw & 0xFFFF));
+      const u32 zword = __builtin_amdgcn_raw_buffer_load_b32(
+          z_rsrc, (u32)(((u64)cc * zw + gg / per_z) * 4), 0, 0);
+      const u32 z = BITS == 4 ? unpack4(zword, gg % 8) : unpack8(zword, gg % 4);
+      const float zb = BITS == 4 ? (float)(z + 128u) : (float)z;
+      if (half == 0) {
+        s_lo[c] = gg < G ? s : 0.f;
+        sz_lo[c] = gg < G ? s * zb : 0.f;
+      } else {
+        s_hi[c] = gg < G ? s : 0.f;
+        sz_hi[c] = gg < G ? s * zb : 0.f;
+      }
+    }
+  }
 
   float acc[COLS][M];
 #pragma unroll
@@ -74,17 +109,9 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
   const int kiters = (k + 2047) / 2048;
   for (int it = 0; it < kiters; ++it) {
     const int k0 = it * 2048 + lane * 32;
-    // per-column group metadata (single group per 32-wide lane chunk)
-    float s[COLS], zb[COLS];
-    const int gi = k0 < k ? k0 / g : 0;
-#pragma unroll
-    for (int c = 0; c < COLS; ++c) {
-      const int cc = col[c] < n ? col[c] : n - 1;
-      s[c] = k0 < k ? bf16_bits_to_f32(srow[(u64)cc * G + gi]) : 0.f;
-      const u32 zword = qzeros[(u64)cc * zw + gi / (32 / BITS)];
-      const u32 z = BITS == 4 ? unpack4(zword, gi & 7) : unpack8(zword, gi & 3);
-      zb[c] = BITS == 4 ? (float)(z + 128u) : (float)z;
-    }
+    const int gi = k0 < k ? k0 / g : 0;  // OOB lanes read s = 0 anyway
+    const bool ghi = gi >= 64;
+    const int gsel = ghi ? gi - 64 : gi;
 
     if (BITS == 4) {
       const u32 wbyte = (u32)(it * 1024 + lane * 16);
@@ -111,6 +138,17 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
         }
 #pragma unroll
         for (int c = 0; c < COLS; ++c) {
+          // NB: __shfl evaluates its ARGUMENT in the source lane, so the
+          // lo/hi array pick must use THIS lane's ghi after shuffling both.
+          const float s_l = __shfl(s_lo[c], gsel, 64);
+          const float s_h = G > 64 ? __shfl(s_hi[c], gsel, 64) : 0.f;
+          const float sz_l = __shfl(sz_lo[c], gsel, 64);
+          const float sz_h = G > 64 ? __shfl(sz_hi[c], gsel, 64) : 0.f;
+          // inactive lanes (k0 >= k) must contribute 0: their x loads can
+          // land in the NEXT row of x when m > 1 (whole-tensor bounds).
+          const float live = k0 < k ? 1.f : 0.f;
+          const float s = (ghi ? s_h : s_l) * live;
+          const float sz = (ghi ? sz_h : sz_l) * live;
           float cx = 0.f;
 #pragma unroll
           for (int j = 0; j < 4; ++j) {
@@ -121,8 +159,8 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
             cx = dot2_bf16(dq.p45, xq[2], cx);
             cx = dot2_bf16(dq.p67, xq[3], cx);
           }
-          acc[c][mm] = fmaf(s[c], cx, acc[c][mm]);
-          acc[c][mm] = fmaf(-s[c] * zb[c], xs, acc[c][mm]);
+          acc[c][mm] = fmaf(s, cx, acc[c][mm]);
+          acc[c][mm] = fmaf(-sz, xs, acc[c][mm]);
         }
       }
     } else {  // BITS == 8
@@ -150,6 +188,13 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
         for (int i = 0; i < 32; ++i) xs += xv[i];
 #pragma unroll
         for (int c = 0; c < COLS; ++c) {
+          const float s_l = __shfl(s_lo[c], gsel, 64);
+          const float s_h = G > 64 ? __shfl(s_hi[c], gsel, 64) : 0.f;
+          const float sz_l = __shfl(sz_lo[c], gsel, 64);
+          const float sz_h = G > 64 ? __shfl(sz_hi[c], gsel, 64) : 0.f;
+          const float live = k0 < k ? 1.f : 0.f;
+          const float s = (ghi ? s_h : s_l) * live;
+          const float sz = (ghi ? sz_h : sz_l) * live;
           float cx = 0.f;
 #pragma unroll
           for (int half = 0; half < 2; ++half) {
@@ -163,14 +208,14 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
               cx = fmaf((float)(wv[j] >> 24), xv[e + 3], cx);
             }
           }
-          acc[c][mm] = fmaf(s[c], cx, acc[c][mm]);
-          acc[c][mm] = fmaf(-s[c] * zb[c], xs, acc[c][mm]);
+          acc[c][mm] = fmaf(s, cx, acc[c][mm]);
+          acc[c][mm] = fmaf(-sz, xs, acc[c][mm]);
         }
       }
     }
   }
 
-  // wave-reduce and store
+  // ---- wave-reduce and store -----------------------------------------
 #pragma unroll
   for (int mm = 0; mm < M; ++mm) {
     if (mm >= m) break;
@@ -223,7 +268,6 @@ hipError_t launch_bits(const void* qweight, const void* scales,
                        hipStream_t stream) {
   dim3 block(WAVES_PER_WG * LFMQ_WAVE);
   if (GATEUP || m <= 4) {
-    // 4 columns per wave (pairs for gateup)
     const int per_wg = WAVES_PER_WG * (GATEUP ? 2 : 4);
     dim3 grid(ceil_div_u32(GATEUP ? n / 2 : n, per_wg));
     if (m <= 1)
@@ -233,11 +277,12 @@ hipError_t launch_bits(const void* qweight, const void* scales,
     else
       launch_one<BITS, 4, GATEUP, 4>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
   } else {
-    dim3 grid(ceil_div_u32(n, WAVES_PER_WG));
+    // larger decode batches: 2 columns per wave (register budget)
+    dim3 grid(ceil_div_u32(n, WAVES_PER_WG * 2));
     if (m <= 8)
-      launch_one<BITS, 1, false, 8>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+      launch_one<BITS, 2, false, 8>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
     else
-      launch_one<BITS, 1, false, 16>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+      launch_one<BITS, 2, false, 16>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
   }
   return hipGetLastError();
 }

@@ -35,6 +35,8 @@ hipError_t lfmq_qkv_rope_append(const void*, void*, void*, void*, const void*,
                                 int, hipStream_t);
 hipError_t lfmq_argmax_rows(const void*, void*, void*, void*, int, int,
                             hipStream_t);
+hipError_t lfmq_read_probe(const void*, void*, long, long, int, int,
+                           hipStream_t);
 
 namespace {
 
@@ -262,6 +264,12 @@ torch::Tensor greedy_sample(torch::Tensor logits) {
   return out;
 }
 
+void read_probe(torch::Tensor src, torch::Tensor sink, long row_bytes,
+                long n_rows, long nt, long streams) {
+  CHECK_HIP(lfmq_read_probe(src.data_ptr(), sink.data_ptr(), row_bytes,
+                            n_rows, (int)nt, (int)streams, stream()));
+}
+
 void hessian_accum(torch::Tensor h, torch::Tensor x) {
   check_t(h, torch::kFloat32, "H");
   check_t(x, torch::kBFloat16, "x");
@@ -288,4 +296,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dq_gemv_gateup", &dq_gemv_gateup, "fused gate_up GEMV + silu*mul");
   m.def("qkv_rope_append", &qkv_rope_append, "fused rope + KV append");
   m.def("greedy_sample", &greedy_sample, "row argmax");
+  m.def("read_probe", &read_probe, "bw probe");
 }

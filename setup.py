@@ -24,6 +24,7 @@ SOURCES = [
     "csrc/attn_decode.hip",
     "csrc/qkv_rope_append.hip",
     "csrc/sample.hip",
+    "csrc/probe.hip",
 ]
 
 setup(
