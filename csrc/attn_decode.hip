@@ -23,9 +23,11 @@ __global__ __launch_bounds__(TPB) void attn_decode_partial(
     const u32* __restrict__ q, const u32* __restrict__ kc,
     const u32* __restrict__ vc, float* __restrict__ scratch,
     u32* __restrict__ o_direct, int b, int hq, int hkv, int tmax, int T,
-    float scale, int nchunks) {
+    const i32* __restrict__ T_dev, float scale, int nchunks) {
+  if (T_dev != nullptr) T = *T_dev;  // hipGraph mode: T lives on device
   const int bh = blockIdx.x;
   const int chunk = blockIdx.y;
+  if (chunk * TPB >= T) return;  // fixed max-chunk grid under graphs
   const int bb = bh / hq;
   const int h = bh % hq;
   const int hk = h / (hq / hkv);
@@ -139,7 +141,8 @@ __global__ __launch_bounds__(TPB) void attn_decode_partial(
 template <int D>
 __global__ __launch_bounds__(128) void attn_decode_combine(
     const float* __restrict__ scratch, u32* __restrict__ o, int nchunks,
-    int T) {
+    int T, const i32* __restrict__ T_dev) {
+  if (T_dev != nullptr) T = *T_dev;
   const int bh = blockIdx.x;
   const int used = (T + TPB - 1) / TPB;
   const float* base = scratch + (u64)bh * nchunks * (D + 2);
@@ -165,11 +168,13 @@ __global__ __launch_bounds__(128) void attn_decode_combine(
 
 }  // namespace
 
-// scratch must hold b*hq*ceil(T/256)*(d+2) floats (unused when T <= 256)
+// scratch must hold b*hq*nchunks*(d+2) floats (unused when nchunks == 1).
+// T_dev != null => graph mode: T read on device, grid sized for max_T
+// (pass max_T as T; nchunks = ceil(max_T/256)).
 hipError_t lfmq_attn_decode(const void* q, const void* kc, const void* vc,
                             void* o, void* scratch, int b, int hq, int hkv,
-                            int tmax, int T, int d, float scale,
-                            hipStream_t stream) {
+                            int tmax, int T, const void* T_dev, int d,
+                            float scale, hipStream_t stream) {
   if (d != 128 && d != 64) return hipErrorInvalidValue;
   if (hq % hkv) return hipErrorInvalidValue;
   const int nchunks = (T + TPB - 1) / TPB;
@@ -179,11 +184,12 @@ hipError_t lfmq_attn_decode(const void* q, const void* kc, const void* vc,
   do {                                                                        \
     hipLaunchKernelGGL((attn_decode_partial<DD>), gridA, blockA, 0, stream,   \
                        (const u32*)q, (const u32*)kc, (const u32*)vc,         \
-                       (float*)scratch, (u32*)o, b, hq, hkv, tmax, T, scale,  \
-                       nchunks);                                              \
+                       (float*)scratch, (u32*)o, b, hq, hkv, tmax, T,         \
+                       (const i32*)T_dev, scale, nchunks);                    \
     if (nchunks > 1)                                                          \
       hipLaunchKernelGGL((attn_decode_combine<DD>), gridB, blockB, 0, stream, \
-                         (const float*)scratch, (u32*)o, nchunks, T);         \
+                         (const float*)scratch, (u32*)o, nchunks, T,          \
+                         (const i32*)T_dev);                                  \
   } while (0)
   if (d == 128)
     LFMQ_ATTN(128);

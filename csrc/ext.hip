@@ -24,15 +24,17 @@ hipError_t lfmq_residual_add(const void*, const void*, void*, long,
 hipError_t lfmq_kv_append(void*, void*, const void*, const void*, int, int,
                           int, int, int, int, hipStream_t);
 hipError_t lfmq_attn_decode(const void*, const void*, const void*, void*,
-                            void*, int, int, int, int, int, int, float,
-                            hipStream_t);
+                            void*, int, int, int, int, int, const void*, int,
+                            float, hipStream_t);
 hipError_t lfmq_hessian_accum(const void*, void*, int, int, hipStream_t);
 hipError_t lfmq_dq_gemv_gateup(const void*, const void*, const void*,
                                const void*, void*, int, int, int, int, int,
                                hipStream_t);
 hipError_t lfmq_qkv_rope_append(const void*, void*, void*, void*, const void*,
                                 const void*, int, int, int, int, int, int,
-                                int, hipStream_t);
+                                int, const void*, hipStream_t);
+hipError_t lfmq_advance_record(void*, void*, const void*, void*, int, int,
+                               hipStream_t);
 hipError_t lfmq_argmax_rows(const void*, void*, void*, void*, int, int,
                             hipStream_t);
 hipError_t lfmq_read_probe(const void*, void*, long, long, int, int,
@@ -202,19 +204,27 @@ void kv_append(torch::Tensor kc, torch::Tensor vc, torch::Tensor kn,
 }
 
 torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
-                          long seq_len, double scale) {
+                          long seq_len, double scale,
+                          c10::optional<torch::Tensor> t_dev) {
   check_t(q, torch::kBFloat16, "q");
   check_t(kc, torch::kBFloat16, "k_cache");
   check_t(vc, torch::kBFloat16, "v_cache");
   const int b = (int)q.size(0), hq = (int)q.size(1), d = (int)q.size(3);
   const int hkv = (int)kc.size(1), tmax = (int)kc.size(2);
+  // graph mode (t_dev set): seq_len is MAX T; grid fixed, T read on device
   const int nchunks = ((int)seq_len + 255) / 256;
   auto scratch = torch::empty({(long)b * hq * nchunks * (d + 2)},
                               q.options().dtype(torch::kFloat32));
   auto o = torch::empty_like(q);
+  const void* tptr = nullptr;
+  if (t_dev.has_value()) {
+    check_t(*t_dev, torch::kInt32, "t_dev");
+    tptr = t_dev->data_ptr();
+  }
   CHECK_HIP(lfmq_attn_decode(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
                              o.data_ptr(), scratch.data_ptr(), b, hq, hkv,
-                             tmax, (int)seq_len, d, (float)scale, stream()));
+                             tmax, (int)seq_len, tptr, d, (float)scale,
+                             stream()));
   return o;
 }
 
@@ -234,7 +244,8 @@ torch::Tensor dq_gemv_gateup(torch::Tensor x, torch::Tensor qweight,
 
 torch::Tensor qkv_rope_append(torch::Tensor qkv, torch::Tensor kc,
                               torch::Tensor vc, torch::Tensor positions,
-                              torch::Tensor inv_freq, long hq, long pos0) {
+                              torch::Tensor inv_freq, long hq, long pos0,
+                              c10::optional<torch::Tensor> pos_dev) {
   check_t(qkv, torch::kBFloat16, "qkv");
   check_t(kc, torch::kBFloat16, "k_cache");
   check_t(vc, torch::kBFloat16, "v_cache");
@@ -244,10 +255,15 @@ torch::Tensor qkv_rope_append(torch::Tensor qkv, torch::Tensor kc,
   const int tmax = (int)kc.size(2), d = (int)kc.size(3);
   const int s = (int)(qkv.numel() / ((hq + 2 * hkv) * d)) / b;
   auto q_out = torch::empty({b, s, hq, (long)d}, qkv.options());
+  const void* pdev = nullptr;
+  if (pos_dev.has_value()) {
+    check_t(*pos_dev, torch::kInt32, "pos_dev");
+    pdev = pos_dev->data_ptr();
+  }
   CHECK_HIP(lfmq_qkv_rope_append(qkv.data_ptr(), q_out.data_ptr(),
                                  kc.data_ptr(), vc.data_ptr(),
                                  positions.data_ptr(), inv_freq.data_ptr(), b,
-                                 s, (int)hq, hkv, d, tmax, (int)pos0,
+                                 s, (int)hq, hkv, d, tmax, (int)pos0, pdev,
                                  stream()));
   return q_out;
 }
@@ -268,6 +284,16 @@ void read_probe(torch::Tensor src, torch::Tensor sink, long row_bytes,
                 long n_rows, long nt, long streams) {
   CHECK_HIP(lfmq_read_probe(src.data_ptr(), sink.data_ptr(), row_bytes,
                             n_rows, (int)nt, (int)streams, stream()));
+}
+
+void advance_record(torch::Tensor state, torch::Tensor tok_buf,
+                    torch::Tensor sampled, torch::Tensor ring) {
+  check_t(state, torch::kInt32, "state");
+  const int b = (int)tok_buf.size(0);
+  const int ring_len = (int)ring.size(-1);
+  CHECK_HIP(lfmq_advance_record(state.data_ptr(), tok_buf.data_ptr(),
+                                sampled.data_ptr(), ring.data_ptr(), b,
+                                ring_len, stream()));
 }
 
 void hessian_accum(torch::Tensor h, torch::Tensor x) {
@@ -297,4 +323,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("qkv_rope_append", &qkv_rope_append, "fused rope + KV append");
   m.def("greedy_sample", &greedy_sample, "row argmax");
   m.def("read_probe", &read_probe, "bw probe");
+  m.def("advance_record", &advance_record, "graph decode-state advance");
 }

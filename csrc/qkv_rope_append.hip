@@ -14,7 +14,9 @@ __global__ void qkv_rope_append_kernel(
     const u32* __restrict__ qkv, u32* __restrict__ q_out,
     u32* __restrict__ kc, u32* __restrict__ vc,
     const i32* __restrict__ positions, const float* __restrict__ inv_freq,
-    int b, int s, int hq, int hkv, int d, int tmax, int pos0) {
+    int b, int s, int hq, int hkv, int d, int tmax, int pos0,
+    const i32* __restrict__ pos_dev) {
+  if (pos_dev != nullptr) pos0 = *pos_dev;  // hipGraph mode
   const int dp = d / 4;          // dword-pairs per rotated half
   const int dw = d / 2;          // dwords per head
   const int qkv_dw = (hq + 2 * hkv) * dw;
@@ -76,7 +78,7 @@ hipError_t lfmq_qkv_rope_append(const void* qkv, void* q_out, void* kc,
                                 void* vc, const void* positions,
                                 const void* inv_freq, int b, int s, int hq,
                                 int hkv, int d, int tmax, int pos0,
-                                hipStream_t stream) {
+                                const void* pos_dev, hipStream_t stream) {
   if (d % 4) return hipErrorInvalidValue;
   const u64 total = (u64)b * s * ((hq + hkv) * (d / 4) + hkv * (d / 2));
   u32 blocks = (u32)((total + 255) / 256);
@@ -85,6 +87,6 @@ hipError_t lfmq_qkv_rope_append(const void* qkv, void* q_out, void* kc,
   hipLaunchKernelGGL(qkv_rope_append_kernel, dim3(blocks), dim3(256), 0,
                      stream, (const u32*)qkv, (u32*)q_out, (u32*)kc, (u32*)vc,
                      (const i32*)positions, (const float*)inv_freq, b, s, hq,
-                     hkv, d, tmax, pos0);
+                     hkv, d, tmax, pos0, (const i32*)pos_dev);
   return hipGetLastError();
 }

@@ -83,3 +83,38 @@ hipError_t lfmq_argmax_rows(const void* logits, void* out, void* scratch_f,
                      (long*)out);
   return hipGetLastError();
 }
+
+namespace {
+
+// hipGraph decode-step bookkeeping: record the sampled token into the
+// output ring, feed it back as the next step's input, and advance the
+// device-side decode state {pos0, T_after, step} — all inside the graph,
+// so replay needs no host work at all.
+__global__ __launch_bounds__(64) void advance_record_kernel(
+    i32* __restrict__ state, long* __restrict__ tok_buf,
+    const long* __restrict__ sampled, long* __restrict__ ring, int b,
+    int ring_len) {
+  if (threadIdx.x == 0) {
+    const int step = state[2];
+    for (int bb = 0; bb < b; ++bb) {
+      // record the token this step CONSUMED (t_r), then feed the newly
+      // sampled t_{r+1} back as the next replay's input
+      if (step < ring_len) ring[(u64)bb * ring_len + step] = tok_buf[bb];
+      tok_buf[bb] = sampled[bb];
+    }
+    state[0] += 1;
+    state[1] += 1;
+    state[2] += 1;
+  }
+}
+
+}  // namespace
+
+hipError_t lfmq_advance_record(void* state, void* tok_buf, const void* sampled,
+                               void* ring, int b, int ring_len,
+                               hipStream_t stream) {
+  hipLaunchKernelGGL(advance_record_kernel, dim3(1), dim3(64), 0, stream,
+                     (i32*)state, (long*)tok_buf, (const long*)sampled,
+                     (long*)ring, b, ring_len);
+  return hipGetLastError();
+}

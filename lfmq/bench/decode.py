@@ -55,24 +55,42 @@ def build_quantized_llama(model_name: str, scheme: str, bits: int, g: int,
 
 @torch.no_grad()
 def bench_decode(model, cfg, batch: int, prompt: int, steps: int, warmup: int,
-                 device) -> dict:
+                 device, use_graph: bool = True) -> dict:
     ids = torch.randint(0, cfg.vocab_size, (batch, prompt), device=device)
-    cache = model.make_cache(batch, max_seq=prompt + warmup + steps + 8,
-                             device=device)
-    logits = model(ids, cache)          # prefill (untimed)
-    tok = logits[:, -1:].argmax(dim=-1)
-    for _ in range(warmup):
-        logits = model(tok, cache)
-        tok = logits[:, -1:].argmax(dim=-1)
-
     use_cuda = torch.cuda.is_available() and str(device) != "cpu"
+    graphed = False
+    if use_cuda and use_graph:
+        try:
+            from ..models.graph import GraphDecoder
+            dec = GraphDecoder(model, batch,
+                               max_seq=prompt + warmup + steps + 8,
+                               max_new_tokens=warmup + steps + 4)
+            dec.prefill(ids)
+            dec.decode(max(warmup, 1))   # includes one-time capture
+            graphed = True
+        except Exception as e:  # pragma: no cover - graph capture fallback
+            import sys
+            print(f"[bench] hipGraph capture failed ({e}); eager decode",
+                  file=sys.stderr)
+    if not graphed:
+        cache = model.make_cache(batch, max_seq=prompt + warmup + steps + 8,
+                                 device=device)
+        logits = model(ids, cache)          # prefill (untimed)
+        tok = logits[:, -1:].argmax(dim=-1)
+        for _ in range(warmup):
+            logits = model(tok, cache)
+            tok = logits[:, -1:].argmax(dim=-1)
+
     lfmq_dist.barrier()
     if use_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(steps):
-        logits = model(tok, cache)
-        tok = logits[:, -1:].argmax(dim=-1)
+    if graphed:
+        dec.decode(steps)
+    else:
+        for _ in range(steps):
+            logits = model(tok, cache)
+            tok = logits[:, -1:].argmax(dim=-1)
     if use_cuda:
         torch.cuda.synchronize()
     t1 = time.perf_counter()
@@ -93,4 +111,5 @@ def bench_decode(model, cfg, batch: int, prompt: int, steps: int, warmup: int,
         "warmup": warmup,
         "batch": batch,
         "prompt": prompt,
+        "hip_graph": graphed,
     }

@@ -133,6 +133,16 @@ class LlamaAttention(nn.Module):
         o = o.transpose(1, 2).reshape(b, s, hq * d)
         return self.o_proj(o)
 
+    def forward_decode_graph(self, x, cache_k, cache_v, state, max_seq):
+        """hipGraph-capturable decode step (s == 1, GPU only)."""
+        b, s, _ = x.shape
+        qkv = self.qkv_proj(x)
+        q = F.qkv_rope_append_graph(qkv, cache_k, cache_v, state,
+                                    self.n_heads, self.cfg.rope_theta)
+        o = F.attn_decode_graph(q.view(b, self.n_heads, 1, self.head_dim),
+                                cache_k, cache_v, state, max_seq, self.scale)
+        return self.o_proj(o.view(b, s, self.n_heads * self.head_dim))
+
     def _chunk_attn(self, q, cache_k, cache_v, pos0, seq_len_after):
         b, hq, s, d = q.shape
         hkv = cache_k.shape[1]
@@ -198,6 +208,23 @@ class LlamaDecoderLayer(nn.Module):
         h = self.mlp(h)
         return h, residual
 
+    def forward_decode_step(self, x, cache_k, cache_v, state, max_seq,
+                            residual=None):
+        if residual is None:
+            residual = x
+            h = self.input_layernorm(x)
+        else:
+            h, residual = F.rmsnorm_residual(x, residual,
+                                             self.input_layernorm.weight,
+                                             self.input_layernorm.eps)
+        h = self.self_attn.forward_decode_graph(h, cache_k, cache_v, state,
+                                                max_seq)
+        h, residual = F.rmsnorm_residual(h, residual,
+                                         self.post_attention_layernorm.weight,
+                                         self.post_attention_layernorm.eps)
+        h = self.mlp(h)
+        return h, residual
+
 
 class LlamaForCausalLM(nn.Module):
     def __init__(self, cfg: LlamaConfig, dtype=torch.bfloat16):
@@ -246,6 +273,16 @@ class LlamaForCausalLM(nn.Module):
         logits = self.lm_head(x)
         cache.advance(s)
         return logits
+
+    def forward_decode_step(self, input_ids, cache, state, max_seq):
+        """One hipGraph-capturable decode step: [b,1] ids -> [b,1,vocab]."""
+        x = self.embed_tokens(input_ids)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            x, residual = layer.forward_decode_step(
+                x, cache.k[i], cache.v[i], state, max_seq, residual)
+        x = self.norm(x, residual=residual)
+        return self.lm_head(x)
 
     @torch.no_grad()
     def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,

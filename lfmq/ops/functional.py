@@ -85,7 +85,8 @@ def silu_mul_gateup(gate_up: torch.Tensor, inter: int) -> torch.Tensor:
 def attn_decode(q, k_cache, v_cache, seq_len: int, scale: Optional[float] = None):
     if q.is_cuda:
         return hip_ext().attn_decode(
-            q, k_cache, v_cache, seq_len, scale or 1.0 / math.sqrt(q.shape[-1]))
+            q, k_cache, v_cache, seq_len,
+            scale or 1.0 / math.sqrt(q.shape[-1]), None)
     return ref.attn_decode(q, k_cache, v_cache, seq_len, scale)
 
 
@@ -117,7 +118,24 @@ def qkv_rope_append(qkv: torch.Tensor, k_cache: torch.Tensor,
     d = k_cache.shape[-1]
     return hip_ext().qkv_rope_append(qkv.contiguous(), k_cache, v_cache,
                                      positions.to(torch.int32),
-                                     _inv_freq(d, theta, qkv.device), hq, pos0)
+                                     _inv_freq(d, theta, qkv.device), hq,
+                                     pos0, None)
+
+
+def qkv_rope_append_graph(qkv: torch.Tensor, k_cache: torch.Tensor,
+                          v_cache: torch.Tensor, state: torch.Tensor,
+                          hq: int, theta: float = 500000.0) -> torch.Tensor:
+    """hipGraph decode variant: position read from state[0] on device."""
+    d = k_cache.shape[-1]
+    pos_dev = state[0:1]
+    return hip_ext().qkv_rope_append(qkv.contiguous(), k_cache, v_cache,
+                                     pos_dev, _inv_freq(d, theta, qkv.device),
+                                     hq, 0, pos_dev)
+
+
+def attn_decode_graph(q, k_cache, v_cache, state, max_t: int, scale: float):
+    """hipGraph decode variant: T read from state[1]; grid sized for max_t."""
+    return hip_ext().attn_decode(q, k_cache, v_cache, max_t, scale, state[1:2])
 
 
 def dq_gemv_gateup(x: torch.Tensor, qweight: torch.Tensor,

@@ -71,3 +71,23 @@ def test_decode_matches_prefill_gpu():
     ld = torch.cat(outs, dim=1)
     rel = (lp - ld).abs().max() / lp.abs().max()
     assert rel < 0.05, f"decode/prefill mismatch rel={float(rel)}"
+
+
+def test_graph_decoder_matches_eager():
+    """hipGraph-captured decode must reproduce eager greedy decode."""
+    from lfmq.models.graph import GraphDecoder
+    cfg = LlamaConfig(vocab_size=2048, hidden_size=512, intermediate_size=1024,
+                      n_layers=2, n_heads=8, n_kv_heads=2, head_dim=64,
+                      max_seq=256)
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=64)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=13), qcfg).to(DEV)
+    ids = torch.randint(0, cfg.vocab_size, (1, 12), device=DEV)
+    eager = qm.generate(ids, max_new_tokens=10)[:, 12:]
+    dec = GraphDecoder(qm, batch=1, max_seq=64, max_new_tokens=16)
+    dec.prefill(ids)
+    got = dec.decode(10)
+    assert torch.equal(got.cpu(), eager.cpu()), (got, eager)
+    # continue decoding — state must carry across calls
+    more = dec.decode(3)
+    eager2 = qm.generate(ids, max_new_tokens=13)[:, 12 + 10:]
+    assert torch.equal(more.cpu(), eager2.cpu())
