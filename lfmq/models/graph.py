@@ -33,6 +33,7 @@ class GraphDecoder:
         self.tok = torch.zeros(batch, dtype=torch.int64, device=self.device)
         self.ring = torch.zeros(batch, max_new_tokens, dtype=torch.int64,
                                 device=self.device)
+        self.step_count = 0
         self.graph: Optional[torch.cuda.CUDAGraph] = None
 
     @torch.no_grad()
@@ -67,6 +68,7 @@ class GraphDecoder:
         self.tok.copy_(first)
         self.state.copy_(torch.tensor([L, L + 1, 0], dtype=torch.int32))
         self.ring.zero_()
+        self.step_count = 0
 
     @torch.no_grad()
     def decode(self, n_tokens: int) -> torch.Tensor:
@@ -80,9 +82,9 @@ class GraphDecoder:
             self.state.copy_(saved[0])
             self.tok.copy_(saved[1])
             self.cache.seq_len = saved[2]
-        # token 0 (from prefill) goes through advance_record on first replay
+        start = self.step_count
         for _ in range(n_tokens):
             self.graph.replay()
+        self.step_count += n_tokens
         self.cache.seq_len += n_tokens
-        out = self.ring[:, :n_tokens].clone()
-        return out
+        return self.ring[:, start:start + n_tokens].clone()

@@ -195,7 +195,7 @@ def test_attn_decode(b, T, d):
     kc = torch.randn(b, hkv, tmax, d).to(torch.bfloat16)
     vc = torch.randn(b, hkv, tmax, d).to(torch.bfloat16)
     scale = 1.0 / math.sqrt(d)
-    o = ext().attn_decode(q.to(DEV), kc.to(DEV), vc.to(DEV), T, scale)
+    o = ext().attn_decode(q.to(DEV), kc.to(DEV), vc.to(DEV), T, scale, None)
     o_ref = ref.attn_decode(q.float(), kc.float(), vc.float(), T, scale)
     assert rel_err(o, o_ref) < 2e-2, f"rel={rel_err(o, o_ref)}"
 
@@ -225,7 +225,7 @@ def test_qkv_rope_append():
     pos = torch.arange(pos0, pos0 + s)
     inv = ref.rope_freqs(d, 500000.0).to(DEV)
     q = ext().qkv_rope_append(qkv.to(DEV), kc, vc, pos.to(torch.int32).to(DEV),
-                              inv, hq, pos0)
+                              inv, hq, pos0, None)
     # reference
     qr, kr, vr = qkv.split([hq * d, hkv * d, hkv * d], dim=-1)
     q_ref = ref.apply_rope(qr.reshape(b, s, hq, d).float(), pos)
