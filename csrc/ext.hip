@@ -37,7 +37,7 @@ hipError_t lfmq_advance_record(void*, void*, const void*, void*, int, int,
                                hipStream_t);
 hipError_t lfmq_argmax_rows(const void*, void*, void*, void*, int, int,
                             hipStream_t);
-hipError_t lfmq_read_probe(const void*, void*, long, long, int, int,
+hipError_t lfmq_read_probe(const void*, void*, long, long, int, int, int,
                            hipStream_t);
 
 namespace {
@@ -281,9 +281,10 @@ torch::Tensor greedy_sample(torch::Tensor logits) {
 }
 
 void read_probe(torch::Tensor src, torch::Tensor sink, long row_bytes,
-                long n_rows, long nt, long streams) {
+                long n_rows, long nt, long streams, long mode) {
   CHECK_HIP(lfmq_read_probe(src.data_ptr(), sink.data_ptr(), row_bytes,
-                            n_rows, (int)nt, (int)streams, stream()));
+                            n_rows, (int)nt, (int)streams, (int)mode,
+                            stream()));
 }
 
 void advance_record(torch::Tensor state, torch::Tensor tok_buf,

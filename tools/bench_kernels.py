@@ -66,7 +66,7 @@ def main():
         q = torch.randn(1, 32, 1, 128, device=DEV).to(torch.bfloat16)
         kc = torch.randn(1, 8, tmax, 128, device=DEV).to(torch.bfloat16)
         vc = torch.randn_like(kc)
-        t = timeit(lambda: ext.attn_decode(q, kc, vc, T, 0.0883))
+        t = timeit(lambda: ext.attn_decode(q, kc, vc, T, 0.0883, None))
         bytes_ = 2 * 8 * T * 128 * 2
         print(f"attn T={T:5d}: {t*1e6:8.2f} us  ({bytes_/t/1e12:5.2f} TB/s cache read)")
 
@@ -84,7 +84,7 @@ def main():
     vc = torch.zeros_like(kc)
     pos = torch.tensor([5], dtype=torch.int32, device=DEV)
     inv = ref.rope_freqs(128).to(DEV)
-    t = timeit(lambda: ext.qkv_rope_append(qkv, kc, vc, pos, inv, 32, 5))
+    t = timeit(lambda: ext.qkv_rope_append(qkv, kc, vc, pos, inv, 32, 5, None))
     print(f"qkv_rope_append decode: {t*1e6:6.2f} us")
 
 
@@ -96,13 +96,13 @@ def bench_read():
         row_bytes = row_kb * 1024
         n_rows = gb * 2**30 // row_bytes
         src = torch.randint(-2**31, 2**31-1, (gb * 2**28,), dtype=torch.int32, device=DEV)
-        for streams in (1, 4, 8):
-            for nt in (0, 2):
-                def fn():
-                    lfmq_hip.read_probe(src, sink, row_bytes, n_rows, nt, streams)
-                t = timeit(fn, iters=10, warmup=3)
-                print(f"read row={row_kb}KB streams={streams} nt={nt}: "
-                      f"{gb/t/1e3:6.2f} TB/s")
+        for streams, nt, mode in [(4, 2, 0), (4, 2, 1), (4, 2, 2),
+                                  (4, 0, 2), (8, 2, 0), (1, 2, 0)]:
+            def fn():
+                lfmq_hip.read_probe(src, sink, row_bytes, n_rows, nt, streams, mode)
+            t = timeit(fn, iters=10, warmup=3)
+            print(f"read row={row_kb}KB streams={streams} nt={nt} mode={mode}: "
+                  f"{gb/t/1e3:6.2f} TB/s")
 
 
 if __name__ == "__main__":
