@@ -36,12 +36,12 @@ namespace {
 constexpr int WAVES_PER_WG = 4;
 constexpr u32 ONE2_BF16 = 0x3F803F80u;  // bf16x2 (1.0, 1.0)
 
-template <int M, int BITS, int COLS, bool GATEUP>
+template <int M, int BITS, int COLS, bool GATEUP, int KI, bool RMS>
 __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
     const u32* __restrict__ qweight, const u32* __restrict__ scales_u16,
     const u32* __restrict__ qzeros, const u32* __restrict__ x,
-    u32* __restrict__ y, const u32* __restrict__ bias, int m, int n, int k,
-    int g) {
+    u32* __restrict__ y, const u32* __restrict__ bias,
+    const u32* __restrict__ resid, int m, int n, int k, int g, float eps) {
   static_assert(!GATEUP || COLS == 4, "gateup pairs need COLS=4");
   const int wave = threadIdx.x / LFMQ_WAVE;
   const int lane = threadIdx.x % LFMQ_WAVE;
@@ -106,7 +106,90 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
 #pragma unroll
     for (int mm = 0; mm < M; ++mm) acc[c][mm] = 0.f;
 
+  float ss_acc[M];  // Σx² per row (RMS mode)
+#pragma unroll
+  for (int mm = 0; mm < M; ++mm) ss_acc[mm] = 0.f;
+
   const int kiters = (k + 2047) / 2048;
+
+  if (BITS == 4 && KI > 0) {
+    // Fully-unrolled k-loop: per-iteration (cx, xs) partials accumulate in
+    // registers; scales are applied ONLY in the epilogue, so the hoisted
+    // scale/zero preload never blocks the weight stream (probe.hip mode 2
+    // shows the full dequant+dot loop sustains >6 TB/s once unserialized).
+    float cxa[COLS][M][KI], xsa[M][KI];
+#pragma unroll
+    for (int it = 0; it < KI; ++it) {
+      const u32 wbyte = (u32)(it * 1024 + lane * 16);
+      u32x4 w[COLS];
+#pragma unroll
+      for (int c = 0; c < COLS; ++c)
+        w[c] = __builtin_amdgcn_raw_buffer_load_b128(w_rsrc[c], wbyte, 0, 2);
+#pragma unroll
+      for (int mm = 0; mm < M; ++mm) {
+        if (mm >= m) break;
+        const u32 xbyte = (u32)(mm * k * 2 + it * 4096 + lane * 64);
+        u32x4 xq0 = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte, 0, 0);
+        u32x4 xq1 = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte + 16, 0, 0);
+        u32x4 xq2 = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte + 32, 0, 0);
+        u32x4 xq3 = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, xbyte + 48, 0, 0);
+        float xs = 0.f;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const u32x4 xq = j == 0 ? xq0 : j == 1 ? xq1 : j == 2 ? xq2 : xq3;
+          xs = dot2_bf16(ONE2_BF16, xq[0], xs);
+          xs = dot2_bf16(ONE2_BF16, xq[1], xs);
+          xs = dot2_bf16(ONE2_BF16, xq[2], xs);
+          xs = dot2_bf16(ONE2_BF16, xq[3], xs);
+          if (RMS) {
+            ss_acc[mm] = dot2_bf16(xq[0], xq[0], ss_acc[mm]);
+            ss_acc[mm] = dot2_bf16(xq[1], xq[1], ss_acc[mm]);
+            ss_acc[mm] = dot2_bf16(xq[2], xq[2], ss_acc[mm]);
+            ss_acc[mm] = dot2_bf16(xq[3], xq[3], ss_acc[mm]);
+          }
+        }
+        xsa[mm][it] = xs;
+#pragma unroll
+        for (int c = 0; c < COLS; ++c) {
+          float cx = 0.f;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            const u32x4 xq = j == 0 ? xq0 : j == 1 ? xq1 : j == 2 ? xq2 : xq3;
+            const DequantPairs4 dq = dequant4_pairs(w[c][j]);
+            cx = dot2_bf16(dq.p01, xq[0], cx);
+            cx = dot2_bf16(dq.p23, xq[1], cx);
+            cx = dot2_bf16(dq.p45, xq[2], cx);
+            cx = dot2_bf16(dq.p67, xq[3], cx);
+          }
+          cxa[c][mm][it] = cx;
+        }
+      }
+    }
+    // scale epilogue (the preloaded s/sz registers are ready by now)
+#pragma unroll
+    for (int it = 0; it < KI; ++it) {
+      const int k0 = it * 2048 + lane * 32;
+      const int gi = k0 < k ? k0 / g : 0;
+      const bool ghi = gi >= 64;
+      const int gsel = ghi ? gi - 64 : gi;
+      const float live = k0 < k ? 1.f : 0.f;
+#pragma unroll
+      for (int c = 0; c < COLS; ++c) {
+        const float s_l = __shfl(s_lo[c], gsel, 64);
+        const float s_h = G > 64 ? __shfl(s_hi[c], gsel, 64) : 0.f;
+        const float sz_l = __shfl(sz_lo[c], gsel, 64);
+        const float sz_h = G > 64 ? __shfl(sz_hi[c], gsel, 64) : 0.f;
+        const float s = (ghi ? s_h : s_l) * live;
+        const float sz = (ghi ? sz_h : sz_l) * live;
+#pragma unroll
+        for (int mm = 0; mm < M; ++mm) {
+          if (mm >= m) break;
+          acc[c][mm] = fmaf(s, cxa[c][mm][it], acc[c][mm]);
+          acc[c][mm] = fmaf(-sz, xsa[mm][it], acc[c][mm]);
+        }
+      }
+    }
+  } else
   for (int it = 0; it < kiters; ++it) {
     const int k0 = it * 2048 + lane * 32;
     const int gi = k0 < k ? k0 / g : 0;  // OOB lanes read s = 0 anyway
@@ -135,6 +218,12 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
           xs = dot2_bf16(ONE2_BF16, xq[1], xs);
           xs = dot2_bf16(ONE2_BF16, xq[2], xs);
           xs = dot2_bf16(ONE2_BF16, xq[3], xs);
+          if (RMS) {
+            ss_acc[mm] = dot2_bf16(xq[0], xq[0], ss_acc[mm]);
+            ss_acc[mm] = dot2_bf16(xq[1], xq[1], ss_acc[mm]);
+            ss_acc[mm] = dot2_bf16(xq[2], xq[2], ss_acc[mm]);
+            ss_acc[mm] = dot2_bf16(xq[3], xq[3], ss_acc[mm]);
+          }
         }
 #pragma unroll
         for (int c = 0; c < COLS; ++c) {
@@ -186,6 +275,10 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
         float xs = 0.f;
 #pragma unroll
         for (int i = 0; i < 32; ++i) xs += xv[i];
+        if (RMS) {
+#pragma unroll
+          for (int i = 0; i < 32; ++i) ss_acc[mm] = fmaf(xv[i], xv[i], ss_acc[mm]);
+        }
 #pragma unroll
         for (int c = 0; c < COLS; ++c) {
           const float s_l = __shfl(s_lo[c], gsel, 64);
@@ -219,12 +312,24 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
 #pragma unroll
   for (int mm = 0; mm < M; ++mm) {
     if (mm >= m) break;
+    float inv = 1.f;
+    if (RMS) {
+      // fused RMSNorm: the wave's lanes saw the whole x row, so the
+      // row's Σx² wave-reduces locally; the norm WEIGHT is pre-folded
+      // into the quantized weights (lfmq.quant.fold_rms_norms), leaving
+      // a scalar rsqrt on the accumulator: Σ wq·(x·inv) = inv·Σ wq·x.
+      const float ss = wave_reduce_sum(ss_acc[mm]);
+      inv = rsqrtf(ss / (float)k + eps);
+    }
     float v[COLS];
 #pragma unroll
     for (int c = 0; c < COLS; ++c) {
-      v[c] = wave_reduce_sum(acc[c][mm]);
+      v[c] = wave_reduce_sum(acc[c][mm]) * inv;
       if (bias != nullptr && col[c] < n)
         v[c] += bf16_bits_to_f32(((const unsigned short*)bias)[col[c]]);
+      if (!GATEUP && resid != nullptr && col[c] < n)
+        v[c] += bf16_bits_to_f32(
+            ((const unsigned short*)resid)[(u64)mm * n + col[c]]);
     }
     if (lane == 0) {
       unsigned short* yrow = (unsigned short*)y;
@@ -251,53 +356,79 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
   }
 }
 
-template <int BITS, int COLS, bool GATEUP, int M>
+template <int BITS, int COLS, bool GATEUP, int M, bool RMS>
 void launch_one(dim3 grid, dim3 block, hipStream_t stream, const void* qweight,
                 const void* scales, const void* qzeros, const void* x, void* y,
-                const void* bias, int m, int n, int k, int g) {
-  hipLaunchKernelGGL((dq_gemv_kernel<M, BITS, COLS, GATEUP>), grid, block, 0,
-                     stream, (const u32*)qweight, (const u32*)scales,
-                     (const u32*)qzeros, (const u32*)x, (u32*)y,
-                     (const u32*)bias, m, n, k, g);
+                const void* bias, const void* resid, int m, int n, int k,
+                int g, float eps) {
+  const int kiters = (k + 2047) / 2048;
+#define LFMQ_KI(KIV)                                                          \
+  hipLaunchKernelGGL((dq_gemv_kernel<M, BITS, COLS, GATEUP, KIV, RMS>), grid, \
+                     block, 0, stream, (const u32*)qweight,                   \
+                     (const u32*)scales, (const u32*)qzeros, (const u32*)x,   \
+                     (u32*)y, (const u32*)bias, (const u32*)resid, m, n, k,   \
+                     g, eps)
+  if (BITS == 4 && COLS == 4) {
+    if (kiters == 1) LFMQ_KI(1);
+    else if (kiters == 2) LFMQ_KI(2);
+    else if (kiters == 7) LFMQ_KI(7);
+    else LFMQ_KI(0);
+  } else {
+    LFMQ_KI(0);
+  }
+#undef LFMQ_KI
 }
 
-template <int BITS, bool GATEUP>
+template <int BITS, bool GATEUP, bool RMS>
 hipError_t launch_bits(const void* qweight, const void* scales,
                        const void* qzeros, const void* x, void* y,
-                       const void* bias, int m, int n, int k, int g,
-                       hipStream_t stream) {
+                       const void* bias, const void* resid, int m, int n,
+                       int k, int g, float eps, hipStream_t stream) {
   dim3 block(WAVES_PER_WG * LFMQ_WAVE);
   if (GATEUP || m <= 4) {
     const int per_wg = WAVES_PER_WG * (GATEUP ? 2 : 4);
     dim3 grid(ceil_div_u32(GATEUP ? n / 2 : n, per_wg));
     if (m <= 1)
-      launch_one<BITS, 4, GATEUP, 1>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+      launch_one<BITS, 4, GATEUP, 1, RMS>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
     else if (m <= 2)
-      launch_one<BITS, 4, GATEUP, 2>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+      launch_one<BITS, 4, GATEUP, 2, RMS>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
     else
-      launch_one<BITS, 4, GATEUP, 4>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+      launch_one<BITS, 4, GATEUP, 4, RMS>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
   } else {
     // larger decode batches: 2 columns per wave (register budget)
     dim3 grid(ceil_div_u32(n, WAVES_PER_WG * 2));
     if (m <= 8)
-      launch_one<BITS, 2, false, 8>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+      launch_one<BITS, 2, false, 8, false>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
     else
-      launch_one<BITS, 2, false, 16>(grid, block, stream, qweight, scales, qzeros, x, y, bias, m, n, k, g);
+      launch_one<BITS, 2, false, 16, false>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
   }
   return hipGetLastError();
 }
 
 }  // namespace
 
+// rms_eps >= 0 enables the fused-RMSNorm input path (norm weight must be
+// pre-folded into the quantized weights); resid != null fuses a residual
+// add into the epilogue (invalid under TP row-parallel world > 1 — the
+// caller adds after the all-reduce instead).
 hipError_t lfmq_dq_gemv(const void* qweight, const void* scales,
                         const void* qzeros, const void* x, void* y,
-                        const void* bias, int m, int n, int k, int g, int bits,
+                        const void* bias, const void* resid, int m, int n,
+                        int k, int g, int bits, float rms_eps,
                         hipStream_t stream) {
   if (m > 16 || k % 8 != 0 || g % 32 != 0 || g > k) return hipErrorInvalidValue;
-  if (bits == 4)
-    return launch_bits<4, false>(qweight, scales, qzeros, x, y, bias, m, n, k, g, stream);
-  if (bits == 8)
-    return launch_bits<8, false>(qweight, scales, qzeros, x, y, bias, m, n, k, g, stream);
+  const bool rms = rms_eps >= 0.f;
+  if (rms && m > 4) return hipErrorInvalidValue;
+  if (bits == 4) {
+    if (rms)
+      return launch_bits<4, false, true>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, rms_eps, stream);
+    return launch_bits<4, false, false>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, 0.f, stream);
+  }
+  if (bits == 8) {
+    if (rms)
+      return launch_bits<8, false, true>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, rms_eps, stream);
+    return launch_bits<8, false, false>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, 0.f, stream);
+  }
   return hipErrorInvalidValue;
 }
 
@@ -306,12 +437,19 @@ hipError_t lfmq_dq_gemv(const void* qweight, const void* scales,
 hipError_t lfmq_dq_gemv_gateup(const void* qweight, const void* scales,
                                const void* qzeros, const void* x, void* y,
                                int m, int n, int k, int g, int bits,
-                               hipStream_t stream) {
+                               float rms_eps, hipStream_t stream) {
   if (m > 4 || k % 8 != 0 || g % 32 != 0 || g > k || n % 2)
     return hipErrorInvalidValue;
-  if (bits == 4)
-    return launch_bits<4, true>(qweight, scales, qzeros, x, y, nullptr, m, n, k, g, stream);
-  if (bits == 8)
-    return launch_bits<8, true>(qweight, scales, qzeros, x, y, nullptr, m, n, k, g, stream);
+  const bool rms = rms_eps >= 0.f;
+  if (bits == 4) {
+    if (rms)
+      return launch_bits<4, true, true>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, rms_eps, stream);
+    return launch_bits<4, true, false>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, 0.f, stream);
+  }
+  if (bits == 8) {
+    if (rms)
+      return launch_bits<8, true, true>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, rms_eps, stream);
+    return launch_bits<8, true, false>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, 0.f, stream);
+  }
   return hipErrorInvalidValue;
 }

@@ -52,6 +52,9 @@ def build_model(model_cfg: Union[dict, LlamaConfig, OPTConfig], dtype=None,
 def quantize(model: nn.Module, cfg: QuantConfig, calib=None,
              device: Optional[torch.device] = None) -> nn.Module:
     """Quantize every target nn.Linear in-place; returns the model."""
+    if cfg.fold_rms and getattr(model.cfg, "arch", None) == "llama":
+        from .quant.fold import fold_rms_norms
+        fold_rms_norms(model)
     if cfg.scheme == "rtn":
         from .quant.rtn import RTNQuantizer
         return RTNQuantizer(cfg).run(model)
@@ -110,4 +113,8 @@ def load(path: str, device: str = "cpu", tp_rank: int = 0, tp_world: int = 1,
     model = _build_quantized_skeleton(meta, dtype=dtype)
     state = load_packed(path, device="cpu")
     model.load_state_dict(state, strict=True)
-    return model.to(device)
+    model = model.to(device)
+    if getattr(model.cfg, "arch", None) == "llama":
+        from .quant.fold import refresh_norm_fold_flags
+        refresh_norm_fold_flags(model)
+    return model

@@ -50,6 +50,10 @@ class QuantConfig:
     awq_grid: int = 20              # α grid resolution over [0, 1]
     awq_clip: bool = False          # optional weight-clip search
 
+    # fold RMSNorm weights into the next projection pre-quantization
+    # (enables the fused-norm decode GEMV; lfmq/quant/fold.py)
+    fold_rms: bool = True
+
     # calibration
     calib_seq_len: int = 512
     calib_n_seqs: int = 16

@@ -136,12 +136,24 @@ class LlamaAttention(nn.Module):
     def forward_decode_graph(self, x, cache_k, cache_v, state, max_seq):
         """hipGraph-capturable decode step (s == 1, GPU only)."""
         b, s, _ = x.shape
-        qkv = self.qkv_proj(x)
+        o = self.decode_qkv_attn(x, cache_k, cache_v, state, max_seq)
+        return self.o_proj(o)
+
+    def decode_qkv_attn(self, x, cache_k, cache_v, state, max_seq,
+                        rms_eps: float = -1.0):
+        """qkv -> rope+append -> attention; returns [b, s, hq*d] (pre
+        o_proj).  rms_eps >= 0 fuses the input RMSNorm into the qkv GEMV
+        (norm weight pre-folded)."""
+        b, s, _ = x.shape
+        if rms_eps >= 0:
+            qkv = self.qkv_proj.forward_rms(x, rms_eps)
+        else:
+            qkv = self.qkv_proj(x)
         q = F.qkv_rope_append_graph(qkv, cache_k, cache_v, state,
                                     self.n_heads, self.cfg.rope_theta)
         o = F.attn_decode_graph(q.view(b, self.n_heads, 1, self.head_dim),
                                 cache_k, cache_v, state, max_seq, self.scale)
-        return self.o_proj(o.view(b, s, self.n_heads * self.head_dim))
+        return o.view(b, s, self.n_heads * self.head_dim)
 
     def _chunk_attn(self, q, cache_k, cache_v, pos0, seq_len_after):
         b, hq, s, d = q.shape
@@ -225,6 +237,23 @@ class LlamaDecoderLayer(nn.Module):
         h = self.mlp(h)
         return h, residual
 
+    def forward_decode_fused(self, x, cache_k, cache_v, state, max_seq):
+        """Folded-norm decode step: both RMSNorms ride inside the GEMVs
+        and both residual adds ride in the o/down epilogues — 6 kernels
+        per layer, no separate norm/add launches.  Requires
+        _in_norm_folded and _post_norm_folded (lfmq.quant.fold)."""
+        eps = self.input_layernorm.eps
+        o = self.self_attn.decode_qkv_attn(x, cache_k, cache_v, state,
+                                           max_seq, rms_eps=eps)
+        x = self.self_attn.o_proj.forward_resid(o, x)
+        mlp = self.mlp
+        h = F.dq_gemv_gateup(x, mlp.gate_up_proj.qweight,
+                             mlp.gate_up_proj.scales, mlp.gate_up_proj.qzeros,
+                             mlp.gate_up_proj.bits, mlp.gate_up_proj.group_size,
+                             rms_eps=self.post_attention_layernorm.eps)
+        x = mlp.down_proj.forward_resid(h, x)
+        return x
+
 
 class LlamaForCausalLM(nn.Module):
     def __init__(self, cfg: LlamaConfig, dtype=torch.bfloat16):
@@ -274,14 +303,32 @@ class LlamaForCausalLM(nn.Module):
         cache.advance(s)
         return logits
 
+    def _decode_fused_ready(self) -> bool:
+        from .qlinear import QLinear
+        for layer in self.layers:
+            if not (getattr(layer, "_in_norm_folded", False)
+                    and getattr(layer, "_post_norm_folded", False)):
+                return False
+            if not isinstance(layer.self_attn.qkv_proj, QLinear):
+                return False
+        return True
+
     def forward_decode_step(self, input_ids, cache, state, max_seq):
         """One hipGraph-capturable decode step: [b,1] ids -> [b,1,vocab]."""
         x = self.embed_tokens(input_ids)
-        residual = None
-        for i, layer in enumerate(self.layers):
-            x, residual = layer.forward_decode_step(
-                x, cache.k[i], cache.v[i], state, max_seq, residual)
-        x = self.norm(x, residual=residual)
+        if getattr(self, "_fused_decode", None) is None:
+            self._fused_decode = self._decode_fused_ready()
+        if self._fused_decode:
+            for i, layer in enumerate(self.layers):
+                x = layer.forward_decode_fused(x, cache.k[i], cache.v[i],
+                                               state, max_seq)
+            x = self.norm(x)
+        else:
+            residual = None
+            for i, layer in enumerate(self.layers):
+                x, residual = layer.forward_decode_step(
+                    x, cache.k[i], cache.v[i], state, max_seq, residual)
+            x = self.norm(x, residual=residual)
         return self.lm_head(x)
 
     @torch.no_grad()

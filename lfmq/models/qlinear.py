@@ -79,6 +79,17 @@ class QLinear(nn.Module):
         return F.dq_linear(x, self.qweight, self.scales, self.qzeros,
                            self.bits, self.group_size, self.bias)
 
+    def forward_rms(self, x: torch.Tensor, eps: float) -> torch.Tensor:
+        """Decode fast path: rmsnorm(x, 1) fused into the GEMV."""
+        return F.dq_linear(x, self.qweight, self.scales, self.qzeros,
+                           self.bits, self.group_size, self.bias,
+                           rms_eps=eps)
+
+    def forward_resid(self, x: torch.Tensor, resid: torch.Tensor) -> torch.Tensor:
+        """Decode fast path: y = xW^T + resid fused in the epilogue."""
+        return F.dq_linear(x, self.qweight, self.scales, self.qzeros,
+                           self.bits, self.group_size, self.bias, resid=resid)
+
 
 class Int8PerTensorLinear(nn.Module):
     """Per-tensor symmetric int8 linear — the OPT-125m CPU decode path
@@ -136,3 +147,11 @@ class RowParallelQLinear(QLinear):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         y = super().forward(x)
         return lfmq_dist.all_reduce_sum(y)
+
+    def forward_resid(self, x: torch.Tensor, resid: torch.Tensor) -> torch.Tensor:
+        # the residual may be added only ONCE, after the reduce
+        if lfmq_dist.get_world_size() > 1:
+            y = self.forward(x)
+            from ..ops import hip_ext
+            return hip_ext().residual_add(y.contiguous(), resid.contiguous())
+        return super().forward_resid(x, resid)

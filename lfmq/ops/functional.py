@@ -58,12 +58,18 @@ def apply_rope(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
 
 def dq_linear(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,
               qzeros: torch.Tensor, bits: int, group_size: int,
-              bias: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """Fused dequant-GEMM  y = x @ dequant(W)^T  (+bias)."""
+              bias: Optional[torch.Tensor] = None,
+              resid: Optional[torch.Tensor] = None,
+              rms_eps: float = -1.0) -> torch.Tensor:
+    """Fused dequant-GEMM y = x @ dequant(W)^T (+bias) (+residual)
+    (rms_eps >= 0: fused-RMSNorm input, decode m<=4 GPU path only)."""
     if x.is_cuda:
         x2 = x.reshape(-1, x.shape[-1])
-        y = hip_ext().dq_linear(x2, qweight, scales, qzeros, bits, group_size, bias)
+        r2 = resid.reshape(-1, resid.shape[-1]).contiguous() if resid is not None else None
+        y = hip_ext().dq_linear(x2, qweight, scales, qzeros, bits, group_size,
+                                bias, r2, rms_eps)
         return y.reshape(*x.shape[:-1], y.shape[-1])
+    assert resid is None and rms_eps < 0, "fused rms/resid is GPU-only"
     return ref.dq_linear(x, qweight, scales, qzeros, bits, group_size, bias).to(x.dtype)
 
 
@@ -140,10 +146,12 @@ def attn_decode_graph(q, k_cache, v_cache, state, max_t: int, scale: float):
 
 def dq_gemv_gateup(x: torch.Tensor, qweight: torch.Tensor,
                    scales: torch.Tensor, qzeros: torch.Tensor, bits: int,
-                   group_size: int) -> torch.Tensor:
-    """GPU-only fused gate_up GEMV + silu·mul epilogue (m <= 4)."""
+                   group_size: int, rms_eps: float = -1.0) -> torch.Tensor:
+    """GPU-only fused gate_up GEMV + silu·mul epilogue (m <= 4);
+    rms_eps >= 0 additionally fuses the RMSNorm of the input."""
     x2 = x.reshape(-1, x.shape[-1])
-    y = hip_ext().dq_gemv_gateup(x2, qweight, scales, qzeros, bits, group_size)
+    y = hip_ext().dq_gemv_gateup(x2, qweight, scales, qzeros, bits,
+                                 group_size, rms_eps)
     return y.reshape(*x.shape[:-1], y.shape[-1])
 
 

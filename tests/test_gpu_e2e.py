@@ -74,12 +74,14 @@ def test_decode_matches_prefill_gpu():
 
 
 def test_graph_decoder_matches_eager():
-    """hipGraph-captured decode must reproduce eager greedy decode."""
+    """hipGraph-captured decode must reproduce eager greedy decode
+    (fold_rms=False so both sides run the identical unfused kernels)."""
     from lfmq.models.graph import GraphDecoder
     cfg = LlamaConfig(vocab_size=2048, hidden_size=512, intermediate_size=1024,
                       n_layers=2, n_heads=8, n_kv_heads=2, head_dim=64,
                       max_seq=256)
-    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=64)
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=64,
+                            fold_rms=False)
     qm = lfmq.quantize(lfmq.build_model(cfg, seed=13), qcfg).to(DEV)
     ids = torch.randint(0, cfg.vocab_size, (1, 12), device=DEV)
     eager = qm.generate(ids, max_new_tokens=10)[:, 12:]
@@ -91,3 +93,29 @@ def test_graph_decoder_matches_eager():
     more = dec.decode(3)
     eager2 = qm.generate(ids, max_new_tokens=13)[:, 12 + 10:]
     assert torch.equal(more.cpu(), eager2.cpu())
+
+
+def test_fused_fold_decode_matches_unfused():
+    """Folded-norm fused decode (6-kernel layer) vs the unfused path on
+    the same underlying weights: logits must agree closely."""
+    from lfmq.models.graph import GraphDecoder
+    cfg = LlamaConfig(vocab_size=2048, hidden_size=512, intermediate_size=1024,
+                      n_layers=2, n_heads=8, n_kv_heads=2, head_dim=64,
+                      max_seq=256)
+    torch.manual_seed(31)
+    ids = torch.randint(0, cfg.vocab_size, (1, 12), device=DEV)
+
+    qcfg_f = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=64, fold_rms=True)
+    mf = lfmq.quantize(lfmq.build_model(cfg, seed=7), qcfg_f).to(DEV)
+    assert mf._decode_fused_ready()
+    df = GraphDecoder(mf, 1, max_seq=64, max_new_tokens=16)
+    df.prefill(ids)
+    toks_f = df.decode(8)
+
+    qcfg_u = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=64, fold_rms=False)
+    mu = lfmq.quantize(lfmq.build_model(cfg, seed=7), qcfg_u).to(DEV)
+    du = GraphDecoder(mu, 1, max_seq=64, max_new_tokens=16)
+    du.prefill(ids)
+    toks_u = du.decode(8)
+    agree = (toks_f == toks_u).float().mean().item()
+    assert agree >= 0.75, f"fused vs unfused token agreement {agree}"
