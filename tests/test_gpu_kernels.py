@@ -77,7 +77,7 @@ def test_dq_gemv(bits, m, n, k, g):
     y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(),
                           qz, bits, g)
     y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
-                        qz.to(DEV), bits, g, None)
+                        qz.to(DEV), bits, g, None, None, -1.0)
     assert y.shape == (m, n)
     assert rel_err(y, y_ref) < 2e-2, f"rel={rel_err(y, y_ref)}"
 
@@ -92,7 +92,7 @@ def test_dq_gemv_bias():
     y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(), qz,
                           4, g, bias.float())
     y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
-                        qz.to(DEV), 4, g, bias.to(DEV))
+                        qz.to(DEV), 4, g, bias.to(DEV), None, -1.0)
     assert rel_err(y, y_ref) < 2e-2
 
 
@@ -109,7 +109,7 @@ def test_dq_gemm(bits, m, n, k, g):
     y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(),
                           qz, bits, g)
     y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
-                        qz.to(DEV), bits, g, None)
+                        qz.to(DEV), bits, g, None, None, -1.0)
     assert rel_err(y, y_ref) < 3e-2, f"rel={rel_err(y, y_ref)}"
 
 
@@ -211,7 +211,8 @@ def test_dq_gemv_gateup():
     gu = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(), qz, 4, g)
     y_ref = ref.silu_mul(gu[:, :inter].contiguous(), gu[:, inter:].contiguous())
     y = ext().dq_gemv_gateup(x.to(DEV), qw.to(DEV),
-                             scales.to(torch.bfloat16).to(DEV), qz.to(DEV), 4, g)
+                             scales.to(torch.bfloat16).to(DEV), qz.to(DEV),
+                             4, g, -1.0)
     assert y.shape == (1, inter)
     assert rel_err(y, y_ref) < 2e-2
 
@@ -244,3 +245,34 @@ def test_greedy_sample():
         got = ext().greedy_sample(logits.to(DEV))
         want = logits.float().argmax(dim=-1)
         assert torch.equal(got.cpu(), want), (b, vocab)
+
+
+
+def test_dq_gemv_rms_fused():
+    """GEMV with fused RMSNorm input vs rmsnorm+gemv reference."""
+    torch.manual_seed(17)
+    n, k, g, eps = 64, 4096, 128, 1e-5
+    w = torch.randn(n, k)
+    codes, scales, zeros = ref.quant_groupwise(w, 4, g, False)
+    qw, qz = ref.pack_codes(codes, 4), ref.pack_zeros(zeros, 4)
+    x = torch.randn(1, k).to(torch.bfloat16)
+    xn = ref.rmsnorm(x, torch.ones(k).to(torch.bfloat16), eps)
+    y_ref = ref.dq_linear(xn.float(), qw, scales.to(torch.bfloat16).float(), qz, 4, g)
+    y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
+                        qz.to(DEV), 4, g, None, None, eps)
+    assert rel_err(y, y_ref) < 2e-2, f"rel={rel_err(y, y_ref)}"
+
+
+def test_dq_gemv_resid_fused():
+    torch.manual_seed(19)
+    n, k, g = 128, 2048, 128
+    w = torch.randn(n, k)
+    codes, scales, zeros = ref.quant_groupwise(w, 4, g, False)
+    qw, qz = ref.pack_codes(codes, 4), ref.pack_zeros(zeros, 4)
+    x = torch.randn(2, k).to(torch.bfloat16)
+    r = torch.randn(2, n).to(torch.bfloat16)
+    y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(),
+                          qz, 4, g) + r.float()
+    y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
+                        qz.to(DEV), 4, g, None, r.to(DEV), -1.0)
+    assert rel_err(y, y_ref) < 2e-2
