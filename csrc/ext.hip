@@ -35,6 +35,10 @@ hipError_t lfmq_qkv_rope_append(const void*, void*, void*, void*, const void*,
                                 int, const void*, hipStream_t);
 hipError_t lfmq_advance_record(void*, void*, const void*, void*, int, int,
                                hipStream_t);
+hipError_t lfmq_grouped_dq_gemv(const void*, const void*, const void*,
+                                const void*, void*, const void*, const void*,
+                                int, int, int, int, int, int, bool,
+                                hipStream_t);
 hipError_t lfmq_argmax_rows(const void*, void*, void*, void*, int, int,
                             hipStream_t);
 hipError_t lfmq_read_probe(const void*, void*, long, long, int, int, int,
@@ -297,6 +301,42 @@ void read_probe(torch::Tensor src, torch::Tensor sink, long row_bytes,
                             stream()));
 }
 
+torch::Tensor grouped_dq_gateup(torch::Tensor x, torch::Tensor src_tok,
+                                torch::Tensor offsets, torch::Tensor qweight,
+                                torch::Tensor scales, torch::Tensor qzeros,
+                                long bits, long g) {
+  check_t(x, torch::kBFloat16, "x");
+  check_t(qweight, torch::kInt32, "qweight");
+  check_t(src_tok, torch::kInt32, "src_tok");
+  check_t(offsets, torch::kInt32, "offsets");
+  const int E = (int)qweight.size(0), n = (int)qweight.size(1) == 0 ? 0 : (int)qweight.size(1);
+  const int k = (int)x.size(1);
+  const long slots = src_tok.numel();
+  auto y = torch::empty({slots, (long)(n / 2)}, x.options());
+  CHECK_HIP(lfmq_grouped_dq_gemv(qweight.data_ptr(), scales.data_ptr(),
+                                 qzeros.data_ptr(), x.data_ptr(), y.data_ptr(),
+                                 src_tok.data_ptr(), offsets.data_ptr(), E, n,
+                                 k, (int)g, (int)bits, (int)x.size(0), true,
+                                 stream()));
+  return y;
+}
+
+torch::Tensor grouped_dq_linear(torch::Tensor x, torch::Tensor offsets,
+                                torch::Tensor qweight, torch::Tensor scales,
+                                torch::Tensor qzeros, long bits, long g) {
+  check_t(x, torch::kBFloat16, "x");
+  check_t(qweight, torch::kInt32, "qweight");
+  check_t(offsets, torch::kInt32, "offsets");
+  const int E = (int)qweight.size(0), n = (int)qweight.size(1);
+  const int k = (int)x.size(1);
+  auto y = torch::empty({x.size(0), (long)n}, x.options());
+  CHECK_HIP(lfmq_grouped_dq_gemv(qweight.data_ptr(), scales.data_ptr(),
+                                 qzeros.data_ptr(), x.data_ptr(), y.data_ptr(),
+                                 nullptr, offsets.data_ptr(), E, n, k, (int)g,
+                                 (int)bits, (int)x.size(0), false, stream()));
+  return y;
+}
+
 void advance_record(torch::Tensor state, torch::Tensor tok_buf,
                     torch::Tensor sampled, torch::Tensor ring) {
   check_t(state, torch::kInt32, "state");
@@ -335,4 +375,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("greedy_sample", &greedy_sample, "row argmax");
   m.def("read_probe", &read_probe, "bw probe");
   m.def("advance_record", &advance_record, "graph decode-state advance");
+  m.def("grouped_dq_gateup", &grouped_dq_gateup, "MoE grouped gate_up+silu");
+  m.def("grouped_dq_linear", &grouped_dq_linear, "MoE grouped linear");
 }

@@ -87,12 +87,11 @@ def _scale_plan(model) -> List[dict]:
              "linears": ["fc2"]},
         ]
     if arch == "mixtral":
-        plan = [{"prev": ("norm", "input_layernorm", None),
+        return [{"prev": ("norm", "input_layernorm", None),
                  "linears": ["self_attn.qkv_proj"]},
                 {"prev": ("norm", "post_attention_layernorm", None),
-                 "linears": [f"moe.experts.{e}.w1" for e in range(model.cfg.n_experts)]
-                            + [f"moe.experts.{e}.w3" for e in range(model.cfg.n_experts)]}]
-        return plan
+                 "linears": [f"moe.experts.{e}.w13"
+                             for e in range(model.cfg.n_experts)]}]
     raise ValueError(f"unknown arch {arch!r}")
 
 

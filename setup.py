@@ -25,6 +25,7 @@ SOURCES = [
     "csrc/qkv_rope_append.hip",
     "csrc/sample.hip",
     "csrc/probe.hip",
+    "csrc/grouped_dq_gemm.hip",
 ]
 
 setup(

@@ -119,3 +119,21 @@ def test_fused_fold_decode_matches_unfused():
     toks_u = du.decode(8)
     agree = (toks_f == toks_u).float().mean().item()
     assert agree >= 0.75, f"fused vs unfused token agreement {agree}"
+
+
+def test_mixtral_gpu_decode():
+    """Mixtral on GPU: grouped-kernel decode vs CPU reference logits."""
+    from lfmq.models.mixtral import MixtralConfig
+    cfg = MixtralConfig(vocab_size=512, hidden_size=256, intermediate_size=512,
+                        n_layers=2, n_heads=4, n_kv_heads=2, head_dim=64,
+                        n_experts=4, top_k=2, max_seq=128)
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=64)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=19), qcfg)
+    ids = torch.randint(0, cfg.vocab_size, (1, 8))
+    l_cpu = qm(ids, qm.make_cache(1, 32)).float()
+    qg = qm.to(DEV).consolidate_moe()
+    l_gpu = qg(ids.to(DEV), qg.make_cache(1, 32)).float()
+    rel = (l_gpu.cpu() - l_cpu).abs().max() / l_cpu.abs().max()
+    assert rel < 0.05, float(rel)
+    out = qg.generate(ids.to(DEV), max_new_tokens=6)
+    assert out.shape == (1, 14)

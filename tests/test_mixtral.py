@@ -1,0 +1,77 @@
+"""Mixtral MoE: CPU e2e + routing math + EP all-to-all (gloo)."""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+import lfmq
+from lfmq.models.mixtral import MixtralConfig
+from lfmq.quant.calib import CalibStream
+
+
+def test_mixtral_cpu_e2e_rtn(tmp_path):
+    cfg = MixtralConfig.tiny()
+    fp = lfmq.build_model(cfg, seed=8)
+    ids = torch.randint(0, cfg.vocab_size, (1, 10))
+    lf = fp(ids, fp.make_cache(1, 32)).float()
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=8), qcfg)
+    p = str(tmp_path / "mx")
+    lfmq.save(qm, p, qcfg)
+    m2 = lfmq.load(p)
+    lq = m2(ids, m2.make_cache(1, 32)).float()
+    pf = torch.log_softmax(lf, -1)
+    pq = torch.log_softmax(lq, -1)
+    kl = (pf.exp() * (pf - pq)).sum(-1).mean()
+    assert kl < 0.5, float(kl)
+    out = m2.generate(ids, max_new_tokens=4)
+    assert out.shape == (1, 14)
+
+
+def test_mixtral_gptq_quantize():
+    cfg = MixtralConfig.tiny()
+    qcfg = lfmq.QuantConfig(scheme="gptq", bits=4, group_size=32,
+                            calib_n_seqs=2, calib_seq_len=32)
+    calib = CalibStream(cfg.vocab_size, 32, 2, 7)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=4), qcfg, calib=calib)
+    ids = torch.randint(0, cfg.vocab_size, (1, 8))
+    logits = qm(ids, qm.make_cache(1, 16))
+    assert torch.isfinite(logits.float()).all()
+
+
+def test_router_topk_weights():
+    cfg = MixtralConfig.tiny()
+    m = lfmq.build_model(cfg, seed=1)
+    moe = m.layers[0].moe
+    x = torch.randn(5, cfg.hidden_size).to(torch.bfloat16)
+    w, ids = moe.route(x)
+    assert w.shape == (5, cfg.top_k) and ids.shape == (5, cfg.top_k)
+    assert torch.allclose(w.sum(-1), torch.ones(5), atol=1e-5)
+    assert ids.max() < cfg.n_experts
+
+
+def _ep_worker(rank, world, port, tmpdir):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        cfg = MixtralConfig.tiny()
+        qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+        qm = lfmq.quantize(lfmq.build_model(cfg, seed=17), qcfg)
+        ids = torch.arange(10).remainder(cfg.vocab_size).reshape(1, 10)
+        ref_logits = qm(ids, qm.make_cache(1, 32))  # unsharded reference
+        from lfmq.dist.ep import apply_expert_parallel
+        apply_expert_parallel(qm, rank, world)
+        ep_logits = qm(ids, qm.make_cache(1, 32))
+        rel = (ep_logits.float() - ref_logits.float()).abs().max() / \
+            ref_logits.float().abs().max()
+        assert rel < 0.03, f"EP mismatch rel={float(rel)}"
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ep_gloo_matches_dense(tmp_path):
+    mp.spawn(_ep_worker, args=(2, 29712, str(tmp_path)), nprocs=2, join=True)
