@@ -126,6 +126,9 @@ class MixtralMoE(nn.Module):
         x2 = x.reshape(-1, h)
         weights, ids = self.route(x2)
         t = x2.shape[0]
+        if (x.is_cuda and self._grouped is None
+                and isinstance(self.experts[0].w13, QLinear)):
+            self.consolidate()  # lazy one-time stacking for the grouped kernel
         if x.is_cuda and self._grouped is not None and t * self.top_k <= 64:
             y2 = self._forward_grouped(x2, weights, ids)
         else:
