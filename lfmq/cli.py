@@ -1,0 +1,150 @@
+"""lfmq CLI: quantize / eval / bench / serve.
+
+    python -m lfmq.cli quantize --model llama3-8b --scheme gptq --out ckpt/
+    python -m lfmq.cli eval --ckpt ckpt/
+    python -m lfmq.cli bench --ckpt ckpt/ --steps 128
+    python -m lfmq.cli serve --ckpt ckpt/ --port 8000
+"""
+from __future__ import annotations
+
+import json
+import time
+from typing import Optional
+
+import torch
+import typer
+
+app = typer.Typer(add_completion=False, pretty_exceptions_enable=False)
+
+MODEL_PRESETS = {
+    "llama3-8b": ("llama", "llama3_8b"),
+    "llama3-70b": ("llama", "llama3_70b"),
+    "mixtral-8x7b": ("mixtral", "mixtral_8x7b"),
+    "opt-125m": ("opt", "opt_125m"),
+    "tiny-llama": ("llama", "tiny"),
+    "tiny-mixtral": ("mixtral", "tiny"),
+    "tiny-opt": ("opt", "tiny"),
+}
+
+
+def _build_preset(name: str, seed: int, layers: int = 0):
+    import lfmq
+    from lfmq.models.llama import LlamaConfig
+    from lfmq.models.mixtral import MixtralConfig
+    from lfmq.models.opt import OPTConfig
+    arch, ctor = MODEL_PRESETS[name]
+    cfg_cls = {"llama": LlamaConfig, "mixtral": MixtralConfig,
+               "opt": OPTConfig}[arch]
+    cfg = getattr(cfg_cls, ctor)()
+    if layers:
+        cfg.n_layers = layers
+    return lfmq.build_model(cfg, seed=seed), cfg
+
+
+def _device() -> str:
+    return "cuda:0" if torch.cuda.is_available() else "cpu"
+
+
+@app.command()
+def quantize(model: str = typer.Option("llama3-8b"),
+             scheme: str = typer.Option("gptq"),
+             bits: int = typer.Option(4),
+             group_size: int = typer.Option(128),
+             out: str = typer.Option(...),
+             seed: int = typer.Option(0),
+             layers: int = typer.Option(0, help="override layer count"),
+             calib_seqs: int = typer.Option(16),
+             calib_len: int = typer.Option(512),
+             resume: bool = typer.Option(False, help="resume a partial run"),
+             act_order: bool = typer.Option(False)):
+    """Random-init a model (no network => synthetic weights), calibrate,
+    quantize and write the packed checkpoint."""
+    import lfmq
+    from lfmq.obs import phase_timer
+
+    dev = _device()
+    qcfg = lfmq.QuantConfig(scheme=scheme, bits=bits, group_size=group_size,
+                            calib_n_seqs=calib_seqs, calib_seq_len=calib_len,
+                            act_order=act_order)
+    with phase_timer("build model"):
+        m, cfg = _build_preset(model, seed, layers)
+        m = m.to(dev)
+    with phase_timer(f"{scheme} quantize"):
+        if scheme == "gptq" and resume:
+            from lfmq.quant.gptq import GPTQQuantizer
+            m = GPTQQuantizer(qcfg).run(m, device=torch.device(dev),
+                                        resume_dir=out)
+        else:
+            m = lfmq.quantize(m, qcfg, device=torch.device(dev))
+    with phase_timer("save packed checkpoint"):
+        lfmq.save(m, out, qcfg)
+    typer.echo(f"saved packed checkpoint to {out}")
+
+
+@app.command()
+def eval(ckpt: str = typer.Option(...),
+         seq_len: int = typer.Option(2048),
+         n_seqs: int = typer.Option(4),
+         seed: int = typer.Option(777)):
+    """Delta-perplexity vs the bf16 model (BASELINE.md protocol):
+    teacher-forced NLL on an identical fixed-seed synthetic stream,
+    plus logit KL and per-layer weight MSE (random-init caveat)."""
+    import lfmq
+    from lfmq.bench.ppl import bench_ppl, weight_mse
+    from lfmq.checkpoint import load_meta
+
+    dev = _device()
+    meta = load_meta(ckpt)
+    qm = lfmq.load(ckpt, device=dev)
+    fp = lfmq.build_model(meta["model"], seed=0)
+    fp.init_random(0)
+    fp = fp.to(dev)
+    r = bench_ppl(fp, qm, qm.cfg.vocab_size, seq_len=seq_len, n_seqs=n_seqs,
+                  seed=seed, device=dev)
+    r["weight_mse_mean"] = (
+        sum(weight_mse(fp.cpu(), qm.cpu()).values()) /
+        max(len(weight_mse(fp, qm)), 1))
+    typer.echo(json.dumps(r, indent=2))
+
+
+@app.command()
+def bench(ckpt: Optional[str] = typer.Option(None),
+          model: str = typer.Option("llama3-8b"),
+          scheme: str = typer.Option("rtn"),
+          batch: int = typer.Option(1),
+          prompt: int = typer.Option(128),
+          steps: int = typer.Option(512),
+          warmup: int = typer.Option(16),
+          no_graph: bool = typer.Option(False)):
+    """Decode tok/s (the BASELINE.json headline metric)."""
+    import lfmq
+    from lfmq.bench.decode import bench_decode, build_quantized_llama
+
+    dev = _device()
+    if ckpt:
+        qm = lfmq.load(ckpt, device=dev)
+        cfg = qm.cfg
+    else:
+        qm, cfg, _ = build_quantized_llama(model, scheme, 4, 128, dev)
+    r = bench_decode(qm, cfg, batch, prompt, steps, warmup, dev,
+                     use_graph=not no_graph)
+    typer.echo(json.dumps(r, indent=2))
+
+
+@app.command()
+def serve(ckpt: str = typer.Option(...),
+          host: str = typer.Option("127.0.0.1"),
+          port: int = typer.Option(8000)):
+    """Minimal generation server (fastapi/uvicorn) with prometheus
+    counters."""
+    import lfmq
+    from lfmq.obs.serve import make_app
+
+    dev = _device()
+    qm = lfmq.load(ckpt, device=dev)
+    import uvicorn
+    uvicorn.run(make_app(qm), host=host, port=port, log_level="info")
+
+
+if __name__ == "__main__":
+    app()

@@ -126,7 +126,13 @@ class GPTQQuantizer:
 
     @torch.no_grad()
     def run(self, model: nn.Module, calib: Optional[CalibStream] = None,
-            device: Optional[torch.device] = None) -> nn.Module:
+            device: Optional[torch.device] = None,
+            resume_dir: Optional[str] = None) -> nn.Module:
+        """resume_dir: per-block completion checkpoints —
+        a 70B calibration is long; `quantize --resume` skips blocks whose
+        quantized state was already written (atomic tmp+rename)."""
+        import os
+        import tempfile
         cfg = self.cfg
         device = device or next(model.parameters()).device
         if calib is None:
@@ -138,9 +144,25 @@ class GPTQQuantizer:
         run_block = make_block_runner(model)
         xs = _capture_block0_inputs(model, calib, device)
 
+        if resume_dir:
+            os.makedirs(os.path.join(resume_dir, "blocks"), exist_ok=True)
+
         target_names = {n for n, _ in iter_target_linears(model, cfg)}
-        for bname, block in decoder_blocks(model):
+        for bi, (bname, block) in enumerate(decoder_blocks(model)):
             block.to(device)
+            if resume_dir:
+                bpath = os.path.join(resume_dir, "blocks", f"{bi:04d}.pt")
+                if os.path.exists(bpath):
+                    saved = torch.load(bpath, weights_only=False)
+                    for lname, meta in saved["manifest"].items():
+                        lin = block.get_submodule(lname)
+                        q = QLinear(meta["in"], meta["out"], meta["bits"],
+                                    meta["g"], bias=meta["bias"]).to(device)
+                        replace_module(block, lname, q)
+                    block.load_state_dict(saved["state"])
+                    xs = [run_block(block, x.to(device)).to(xs[0].device)
+                          for x in xs]
+                    continue
             linears = {n: m for n, m in block.named_modules()
                        if isinstance(m, nn.Linear) and f"{bname}.{n}" in
                        {t for t in target_names if t.startswith(bname + ".")}}
@@ -170,6 +192,22 @@ class GPTQQuantizer:
                 q.to(device)
                 replace_module(block, lname, q)
                 del hess[lname]
+
+            if resume_dir:
+                man = {}
+                for lname, mod in block.named_modules():
+                    if isinstance(mod, QLinear):
+                        man[lname] = {"in": mod.in_features,
+                                      "out": mod.out_features,
+                                      "bits": mod.bits, "g": mod.group_size,
+                                      "bias": mod.bias is not None}
+                fd, tmp = tempfile.mkstemp(dir=os.path.join(resume_dir, "blocks"))
+                os.close(fd)
+                torch.save({"state": {k: v.cpu() for k, v in
+                                      block.state_dict().items()},
+                            "manifest": man}, tmp)
+                os.replace(tmp, os.path.join(resume_dir, "blocks",
+                                             f"{bi:04d}.pt"))
 
             # next block's inputs come from the QUANTIZED block
             xs = [run_block(block, x.to(device)).to(xs[0].device) for x in xs]

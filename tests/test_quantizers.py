@@ -154,3 +154,29 @@ def test_qlinear_w8():
     x = torch.randn(4, 64)
     rel = (q(x).float() - lin(x).float()).abs().mean() / lin(x).float().abs().mean()
     assert rel < 0.05
+
+
+def test_gptq_resume(tmp_path):
+    """Per-block resume: a second run with resume_dir skips finished
+    blocks and yields identical weights."""
+    import lfmq
+    from lfmq.models.llama import LlamaConfig
+    from lfmq.quant.gptq import GPTQQuantizer
+    from lfmq.quant.calib import CalibStream
+    cfg = LlamaConfig.tiny()
+    qcfg = lfmq.QuantConfig(scheme="gptq", bits=4, group_size=64,
+                            calib_n_seqs=2, calib_seq_len=32)
+    rd = str(tmp_path / "resume")
+    calib = CalibStream(cfg.vocab_size, 32, 2, qcfg.calib_seed)
+    m1 = lfmq.build_model(cfg, seed=6)
+    m1 = GPTQQuantizer(qcfg).run(m1, calib=calib, resume_dir=rd)
+    # second run from scratch model with the SAME resume dir: all blocks
+    # restored, result identical
+    calib2 = CalibStream(cfg.vocab_size, 32, 2, qcfg.calib_seed)
+    m2 = lfmq.build_model(cfg, seed=6)
+    m2 = GPTQQuantizer(qcfg).run(m2, calib=calib2, resume_dir=rd)
+    for (k1, v1), (k2, v2) in zip(m1.state_dict().items(),
+                                  m2.state_dict().items()):
+        assert k1 == k2
+        if "layers." in k1:
+            assert torch.equal(v1, v2), k1
