@@ -42,14 +42,21 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
     const u32* __restrict__ qzeros, const u32* __restrict__ x,
     u32* __restrict__ y, const u32* __restrict__ bias,
     const u32* __restrict__ resid, int m, int n, int k, int g, float eps) {
-  static_assert(!GATEUP || COLS == 4, "gateup pairs need COLS=4");
+  static_assert(!GATEUP || COLS == 4 || COLS == 2, "gateup needs COLS 2/4");
   const int wave = threadIdx.x / LFMQ_WAVE;
   const int lane = threadIdx.x % LFMQ_WAVE;
-  const int cbase = (blockIdx.x * WAVES_PER_WG + wave) * (GATEUP ? 2 : COLS);
+  const int cbase = (blockIdx.x * WAVES_PER_WG + wave) *
+                    (GATEUP ? COLS / 2 : COLS);
   int col[COLS];
 #pragma unroll
-  for (int c = 0; c < COLS; ++c)
-    col[c] = GATEUP ? ((c & 1) + cbase + (c >> 1) * (n / 2)) : cbase + c;
+  for (int c = 0; c < COLS; ++c) {
+    if (GATEUP && COLS == 2)
+      col[c] = cbase + c * (n / 2);          // {gate c, up c}
+    else if (GATEUP)
+      col[c] = (c & 1) + cbase + (c >> 1) * (n / 2);  // {g c, g c+1, u c, u c+1}
+    else
+      col[c] = cbase + c;
+  }
   if (col[0] >= n) return;
 
   const int kw_row_bytes = k * BITS / 8;
@@ -336,13 +343,13 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
       if (GATEUP) {
         const int inter = n / 2;
 #pragma unroll
-        for (int p = 0; p < 2; ++p) {  // pairs (gate=v[p], up=v[p+2])
+        for (int p = 0; p < COLS / 2; ++p) {  // pair p: gate v[p], up v[p + COLS/2]
           const int oc = cbase + p;
           if (oc < inter) {
             const float gt = v[p];
             const float sil = gt / (1.f + __expf(-gt));
             yrow[(u64)mm * inter + oc] =
-                (unsigned short)f32_to_bf16_bits_rne(sil * v[p + 2]);
+                (unsigned short)f32_to_bf16_bits_rne(sil * v[p + COLS / 2]);
           }
         }
       } else {
@@ -368,7 +375,7 @@ void launch_one(dim3 grid, dim3 block, hipStream_t stream, const void* qweight,
                      (const u32*)scales, (const u32*)qzeros, (const u32*)x,   \
                      (u32*)y, (const u32*)bias, (const u32*)resid, m, n, k,   \
                      g, eps)
-  if (BITS == 4 && COLS == 4) {
+  if (BITS == 4 && (COLS == 4 || COLS == 2 || COLS == 1)) {
     if (kiters == 1) LFMQ_KI(1);
     else if (kiters == 2) LFMQ_KI(2);
     else if (kiters == 7) LFMQ_KI(7);
@@ -383,17 +390,30 @@ template <int BITS, bool GATEUP, bool RMS>
 hipError_t launch_bits(const void* qweight, const void* scales,
                        const void* qzeros, const void* x, void* y,
                        const void* bias, const void* resid, int m, int n,
-                       int k, int g, float eps, hipStream_t stream) {
+                       int k, int g, float eps, int cols_hint,
+                       hipStream_t stream) {
   dim3 block(WAVES_PER_WG * LFMQ_WAVE);
   if (GATEUP || m <= 4) {
-    const int per_wg = WAVES_PER_WG * (GATEUP ? 2 : 4);
+    // columns per wave: fewer columns -> more workgroups; small-n GEMVs
+    // are occupancy/latency-bound (probe.hip: the inner loop itself
+    // sustains >6 TB/s at large grids).  cols_hint 0 = auto.
+    int cols = cols_hint;
+    if (cols == 0) cols = (GATEUP || n >= 16384) ? 4 : 2;
+    const int per_wg = WAVES_PER_WG * (GATEUP ? cols / 2 : cols);
     dim3 grid(ceil_div_u32(GATEUP ? n / 2 : n, per_wg));
-    if (m <= 1)
-      launch_one<BITS, 4, GATEUP, 1, RMS>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
-    else if (m <= 2)
-      launch_one<BITS, 4, GATEUP, 2, RMS>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
-    else
-      launch_one<BITS, 4, GATEUP, 4, RMS>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
+#define LFMQ_M(CC)                                                            \
+    do {                                                                      \
+      if (m <= 1)                                                             \
+        launch_one<BITS, CC, GATEUP, 1, RMS>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps); \
+      else if (m <= 2)                                                        \
+        launch_one<BITS, CC, GATEUP, 2, RMS>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps); \
+      else                                                                    \
+        launch_one<BITS, CC, GATEUP, 4, RMS>(grid, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps); \
+    } while (0)
+    if (cols == 4) LFMQ_M(4);
+    else if (cols == 1) LFMQ_M(GATEUP ? 2 : 1);
+    else LFMQ_M(2);
+#undef LFMQ_M
   } else {
     // larger decode batches: 2 columns per wave (register budget)
     dim3 grid(ceil_div_u32(n, WAVES_PER_WG * 2));
@@ -414,20 +434,20 @@ hipError_t launch_bits(const void* qweight, const void* scales,
 hipError_t lfmq_dq_gemv(const void* qweight, const void* scales,
                         const void* qzeros, const void* x, void* y,
                         const void* bias, const void* resid, int m, int n,
-                        int k, int g, int bits, float rms_eps,
+                        int k, int g, int bits, float rms_eps, int cols_hint,
                         hipStream_t stream) {
   if (m > 16 || k % 8 != 0 || g % 32 != 0 || g > k) return hipErrorInvalidValue;
   const bool rms = rms_eps >= 0.f;
   if (rms && m > 4) return hipErrorInvalidValue;
   if (bits == 4) {
     if (rms)
-      return launch_bits<4, false, true>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, rms_eps, stream);
-    return launch_bits<4, false, false>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, 0.f, stream);
+      return launch_bits<4, false, true>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, rms_eps, cols_hint, stream);
+    return launch_bits<4, false, false>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, 0.f, cols_hint, stream);
   }
   if (bits == 8) {
     if (rms)
-      return launch_bits<8, false, true>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, rms_eps, stream);
-    return launch_bits<8, false, false>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, 0.f, stream);
+      return launch_bits<8, false, true>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, rms_eps, cols_hint, stream);
+    return launch_bits<8, false, false>(qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, 0.f, cols_hint, stream);
   }
   return hipErrorInvalidValue;
 }
@@ -437,19 +457,20 @@ hipError_t lfmq_dq_gemv(const void* qweight, const void* scales,
 hipError_t lfmq_dq_gemv_gateup(const void* qweight, const void* scales,
                                const void* qzeros, const void* x, void* y,
                                int m, int n, int k, int g, int bits,
-                               float rms_eps, hipStream_t stream) {
+                               float rms_eps, int cols_hint,
+                               hipStream_t stream) {
   if (m > 4 || k % 8 != 0 || g % 32 != 0 || g > k || n % 2)
     return hipErrorInvalidValue;
   const bool rms = rms_eps >= 0.f;
   if (bits == 4) {
     if (rms)
-      return launch_bits<4, true, true>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, rms_eps, stream);
-    return launch_bits<4, true, false>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, 0.f, stream);
+      return launch_bits<4, true, true>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, rms_eps, cols_hint, stream);
+    return launch_bits<4, true, false>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, 0.f, cols_hint, stream);
   }
   if (bits == 8) {
     if (rms)
-      return launch_bits<8, true, true>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, rms_eps, stream);
-    return launch_bits<8, true, false>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, 0.f, stream);
+      return launch_bits<8, true, true>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, rms_eps, cols_hint, stream);
+    return launch_bits<8, true, false>(qweight, scales, qzeros, x, y, nullptr, nullptr, m, n, k, g, 0.f, cols_hint, stream);
   }
   return hipErrorInvalidValue;
 }

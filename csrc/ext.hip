@@ -9,7 +9,7 @@ hipError_t lfmq_pack(const void*, void*, long, int, int, hipStream_t);
 hipError_t lfmq_unpack(const void*, void*, long, int, int, hipStream_t);
 hipError_t lfmq_dq_gemv(const void*, const void*, const void*, const void*,
                         void*, const void*, const void*, int, int, int, int,
-                        int, float, hipStream_t);
+                        int, float, int, hipStream_t);
 hipError_t lfmq_dq_gemm(const void*, const void*, const void*, const void*,
                         void*, const void*, int, int, int, int, int,
                         hipStream_t);
@@ -29,7 +29,7 @@ hipError_t lfmq_attn_decode(const void*, const void*, const void*, void*,
 hipError_t lfmq_hessian_accum(const void*, void*, int, int, hipStream_t);
 hipError_t lfmq_dq_gemv_gateup(const void*, const void*, const void*,
                                const void*, void*, int, int, int, int, int,
-                               float, hipStream_t);
+                               float, int, hipStream_t);
 hipError_t lfmq_qkv_rope_append(const void*, void*, void*, void*, const void*,
                                 const void*, int, int, int, int, int, int,
                                 int, const void*, hipStream_t);
@@ -86,7 +86,8 @@ torch::Tensor unpack(torch::Tensor packed, long bits, long k) {
 torch::Tensor dq_linear(torch::Tensor x, torch::Tensor qweight,
                         torch::Tensor scales, torch::Tensor qzeros, long bits,
                         long g, c10::optional<torch::Tensor> bias,
-                        c10::optional<torch::Tensor> resid, double rms_eps) {
+                        c10::optional<torch::Tensor> resid, double rms_eps,
+                        long cols_hint) {
   check_t(x, torch::kBFloat16, "x");
   check_t(qweight, torch::kInt32, "qweight");
   check_t(scales, torch::kBFloat16, "scales");
@@ -110,7 +111,7 @@ torch::Tensor dq_linear(torch::Tensor x, torch::Tensor qweight,
     CHECK_HIP(lfmq_dq_gemv(qweight.data_ptr(), scales.data_ptr(),
                            qzeros.data_ptr(), x.data_ptr(), y.data_ptr(), bptr,
                            rptr, m, n, k, (int)g, (int)bits, (float)rms_eps,
-                           stream()));
+                           (int)cols_hint, stream()));
   } else {
     TORCH_CHECK(rptr == nullptr && rms_eps < 0,
                 "fused rms/resid is a decode (m<=16) path");
@@ -243,7 +244,8 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
 
 torch::Tensor dq_gemv_gateup(torch::Tensor x, torch::Tensor qweight,
                              torch::Tensor scales, torch::Tensor qzeros,
-                             long bits, long g, double rms_eps) {
+                             long bits, long g, double rms_eps,
+                             long cols_hint) {
   check_t(x, torch::kBFloat16, "x");
   check_t(qweight, torch::kInt32, "qweight");
   const int m = (int)x.size(0), k = (int)x.size(1);
@@ -252,7 +254,7 @@ torch::Tensor dq_gemv_gateup(torch::Tensor x, torch::Tensor qweight,
   CHECK_HIP(lfmq_dq_gemv_gateup(qweight.data_ptr(), scales.data_ptr(),
                                 qzeros.data_ptr(), x.data_ptr(), y.data_ptr(),
                                 m, n, k, (int)g, (int)bits, (float)rms_eps,
-                                stream()));
+                                (int)cols_hint, stream()));
   return y;
 }
 

@@ -67,7 +67,7 @@ def dq_linear(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,
         x2 = x.reshape(-1, x.shape[-1])
         r2 = resid.reshape(-1, resid.shape[-1]).contiguous() if resid is not None else None
         y = hip_ext().dq_linear(x2, qweight, scales, qzeros, bits, group_size,
-                                bias, r2, rms_eps)
+                                bias, r2, rms_eps, 0)
         return y.reshape(*x.shape[:-1], y.shape[-1])
     assert resid is None and rms_eps < 0, "fused rms/resid is GPU-only"
     return ref.dq_linear(x, qweight, scales, qzeros, bits, group_size, bias).to(x.dtype)
@@ -151,7 +151,7 @@ def dq_gemv_gateup(x: torch.Tensor, qweight: torch.Tensor,
     rms_eps >= 0 additionally fuses the RMSNorm of the input."""
     x2 = x.reshape(-1, x.shape[-1])
     y = hip_ext().dq_gemv_gateup(x2, qweight, scales, qzeros, bits,
-                                 group_size, rms_eps)
+                                 group_size, rms_eps, 0)
     return y.reshape(*x.shape[:-1], y.shape[-1])
 
 

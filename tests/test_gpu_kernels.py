@@ -77,7 +77,7 @@ def test_dq_gemv(bits, m, n, k, g):
     y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(),
                           qz, bits, g)
     y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
-                        qz.to(DEV), bits, g, None, None, -1.0)
+                        qz.to(DEV), bits, g, None, None, -1.0, 0)
     assert y.shape == (m, n)
     assert rel_err(y, y_ref) < 2e-2, f"rel={rel_err(y, y_ref)}"
 
@@ -92,7 +92,7 @@ def test_dq_gemv_bias():
     y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(), qz,
                           4, g, bias.float())
     y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
-                        qz.to(DEV), 4, g, bias.to(DEV), None, -1.0)
+                        qz.to(DEV), 4, g, bias.to(DEV), None, -1.0, 0)
     assert rel_err(y, y_ref) < 2e-2
 
 
@@ -109,7 +109,7 @@ def test_dq_gemm(bits, m, n, k, g):
     y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(),
                           qz, bits, g)
     y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
-                        qz.to(DEV), bits, g, None, None, -1.0)
+                        qz.to(DEV), bits, g, None, None, -1.0, 0)
     assert rel_err(y, y_ref) < 3e-2, f"rel={rel_err(y, y_ref)}"
 
 
@@ -212,7 +212,7 @@ def test_dq_gemv_gateup():
     y_ref = ref.silu_mul(gu[:, :inter].contiguous(), gu[:, inter:].contiguous())
     y = ext().dq_gemv_gateup(x.to(DEV), qw.to(DEV),
                              scales.to(torch.bfloat16).to(DEV), qz.to(DEV),
-                             4, g, -1.0)
+                             4, g, -1.0, 0)
     assert y.shape == (1, inter)
     assert rel_err(y, y_ref) < 2e-2
 
@@ -259,7 +259,7 @@ def test_dq_gemv_rms_fused():
     xn = ref.rmsnorm(x, torch.ones(k).to(torch.bfloat16), eps)
     y_ref = ref.dq_linear(xn.float(), qw, scales.to(torch.bfloat16).float(), qz, 4, g)
     y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
-                        qz.to(DEV), 4, g, None, None, eps)
+                        qz.to(DEV), 4, g, None, None, eps, 0)
     assert rel_err(y, y_ref) < 2e-2, f"rel={rel_err(y, y_ref)}"
 
 
@@ -274,7 +274,7 @@ def test_dq_gemv_resid_fused():
     y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(),
                           qz, 4, g) + r.float()
     y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
-                        qz.to(DEV), 4, g, None, r.to(DEV), -1.0)
+                        qz.to(DEV), 4, g, None, r.to(DEV), -1.0, 0)
     assert rel_err(y, y_ref) < 2e-2
 
 

@@ -36,14 +36,17 @@ def bench_gemv(m, n, k, g, bits=4, gateup=False):
     qz = torch.randint(-2**31, 2**31 - 1, (n, (G + per - 1) // per), dtype=torch.int32, device=DEV)
     x = torch.randn(m, k, device=DEV).to(torch.bfloat16)
     ext = hip_ext()
-    if gateup:
-        fn = lambda: ext.dq_gemv_gateup(x, qw, sc, qz, bits, g)
-    else:
-        fn = lambda: ext.dq_linear(x, qw, sc, qz, bits, g, None)
-    t = timeit(fn)
-    wbytes = n * k * bits / 8 + n * G * 2 + qz.numel() * 4
-    print(f"gemv{'-gu' if gateup else '   '} m={m:3d} n={n:6d} k={k:6d} g={g} "
-          f"w{bits}: {t*1e6:8.2f} us  {wbytes/t/1e12:6.2f} TB/s")
+    for cols in (1, 2, 4):
+        if gateup:
+            if cols == 1:
+                continue
+            fn = lambda: ext.dq_gemv_gateup(x, qw, sc, qz, bits, g, -1.0, cols)
+        else:
+            fn = lambda: ext.dq_linear(x, qw, sc, qz, bits, g, None, None, -1.0, cols)
+        t = timeit(fn)
+        wbytes = n * k * bits / 8 + n * G * 2 + qz.numel() * 4
+        print(f"gemv{'-gu' if gateup else '   '} m={m:3d} n={n:6d} k={k:6d} g={g} "
+              f"w{bits} cols={cols}: {t*1e6:8.2f} us  {wbytes/t/1e12:6.2f} TB/s")
 
 
 def main():
