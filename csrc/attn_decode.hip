@@ -138,6 +138,183 @@ __global__ __launch_bounds__(TPB) void attn_decode_partial(
   }
 }
 
+// Fused decode attention (s == 1): consumes the RAW qkv projection
+// buffer — ropes q and k in a cooperative prologue (pos = T-1), appends
+// the new k/v row to the cache (the designated q-head-0 workgroup of
+// each kv head), computes q·k_new by block reduction (so no chunk ever
+// depends on another workgroup's cache write), then runs the same
+// chunked flash-decoding as attn_decode_partial.  Replaces the separate
+// qkv_rope_append launch on the decode path.
+template <int D>
+__global__ __launch_bounds__(TPB) void attn_decode_fused(
+    const u32* __restrict__ qkv, u32* __restrict__ kc, u32* __restrict__ vc,
+    float* __restrict__ scratch, u32* __restrict__ o_direct,
+    const float* __restrict__ inv_freq, int b, int hq, int hkv, int tmax,
+    int T, const i32* __restrict__ T_dev, float scale, int nchunks) {
+  if (T_dev != nullptr) T = *T_dev;
+  const int bh = blockIdx.x;
+  const int chunk = blockIdx.y;
+  if (chunk * TPB >= T) return;
+  const int bb = bh / hq;
+  const int h = bh % hq;
+  const int rep = hq / hkv;
+  const int hk = h / rep;
+  const int dw = D / 2;
+  const int pos = T - 1;
+
+  u32* krow_cache = kc + ((u64)bb * hkv + hk) * tmax * dw;
+  u32* vrow_cache = vc + ((u64)bb * hkv + hk) * tmax * dw;
+  const u32* qkv_row = qkv + (u64)bb * (hq + 2 * hkv) * dw;
+
+  __shared__ float p_sh[TPB];
+  __shared__ float red[NWAVE];
+  __shared__ float o_sh[NWAVE][D];
+  __shared__ float m_sh, l_sh, slast_sh;
+  __shared__ unsigned short q_sh[D], k_sh[D];
+  __shared__ u32 v_sh[D / 2];
+
+  // ---- prologue: rope q (head h) and k (head hk), stage v ------------
+  {
+    const unsigned short* qk16 = (const unsigned short*)qkv_row;
+    float prod = 0.f;
+    if (threadIdx.x < D) {
+      const int e = threadIdx.x;
+      const int i = e & (D / 2 - 1);      // pair index
+      const int hi = e >= D / 2;
+      const float ang = (float)pos * inv_freq[i];
+      float sv, cv;
+      __sincosf(ang, &sv, &cv);
+      const float q1 = bf16_bits_to_f32(qk16[h * D + i]);
+      const float q2 = bf16_bits_to_f32(qk16[h * D + i + D / 2]);
+      const float k1 = bf16_bits_to_f32(qk16[(hq + hk) * D + i]);
+      const float k2 = bf16_bits_to_f32(qk16[(hq + hk) * D + i + D / 2]);
+      const float qr = hi ? (q1 * sv + q2 * cv) : (q1 * cv - q2 * sv);
+      const float kr = hi ? (k1 * sv + k2 * cv) : (k1 * cv - k2 * sv);
+      q_sh[e] = (unsigned short)f32_to_bf16_bits_rne(qr);
+      k_sh[e] = (unsigned short)f32_to_bf16_bits_rne(kr);
+      prod = qr * kr;                      // q·k_new partial (fp32)
+    }
+    if (threadIdx.x < dw)                  // v passthrough (dwords)
+      v_sh[threadIdx.x] = qkv_row[((u64)hq + hkv + hk) * dw + threadIdx.x];
+    // block-reduce q·k_new
+    float wsum = wave_reduce_sum(prod);
+    const int wave0 = threadIdx.x / LFMQ_WAVE;
+    if (threadIdx.x % LFMQ_WAVE == 0) red[wave0] = wsum;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float t = 0.f;
+#pragma unroll
+      for (int i = 0; i < NWAVE; ++i) t += red[i];
+      slast_sh = t;
+    }
+    // designated writer appends the new k/v row to the cache
+    if (h % rep == 0 && chunk == pos / TPB) {
+      if (threadIdx.x < dw) {
+        const unsigned short* ks = k_sh;
+        krow_cache[(u64)pos * dw + threadIdx.x] =
+            (u32)ks[2 * threadIdx.x] | ((u32)ks[2 * threadIdx.x + 1] << 16);
+        vrow_cache[(u64)pos * dw + threadIdx.x] = v_sh[threadIdx.x];
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- q·K over this chunk (q from LDS; t == pos uses the reduction) --
+  const int t = chunk * TPB + threadIdx.x;
+  const int t_clamped = t < tmax ? t : tmax - 1;
+  float s = -INFINITY;
+  {
+    const u32* krow = (const u32*)krow_cache + (u64)t_clamped * dw;
+    const u32* q32 = (const u32*)q_sh;
+    float acc = 0.f;
+#pragma unroll
+    for (int i = 0; i < dw / 4; ++i) {
+      const u32x4 kv4 = *(const u32x4*)(krow + 4 * i);
+      acc = dot2_bf16(kv4[0], q32[4 * i + 0], acc);
+      acc = dot2_bf16(kv4[1], q32[4 * i + 1], acc);
+      acc = dot2_bf16(kv4[2], q32[4 * i + 2], acc);
+      acc = dot2_bf16(kv4[3], q32[4 * i + 3], acc);
+    }
+    if (t == pos) acc = slast_sh;          // never trust the stale cache row
+    if (t < T) s = acc * scale;
+  }
+
+  // ---- chunk softmax (identical to attn_decode_partial) ---------------
+  const int wave = threadIdx.x / LFMQ_WAVE;
+  const int lane = threadIdx.x % LFMQ_WAVE;
+  float wm = wave_reduce_max(s);
+  if (lane == 0) red[wave] = wm;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float m = red[0];
+#pragma unroll
+    for (int i = 1; i < NWAVE; ++i) m = fmaxf(m, red[i]);
+    m_sh = m;
+  }
+  __syncthreads();
+  const float m = m_sh;
+  const float p = (t < T) ? __expf(s - m) : 0.f;
+  p_sh[threadIdx.x] = p;
+  float wl = wave_reduce_sum(p);
+  if (lane == 0) red[wave] = wl;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float l = 0.f;
+#pragma unroll
+    for (int i = 0; i < NWAVE; ++i) l += red[i];
+    l_sh = l;
+  }
+
+  // ---- PV (v of t == pos comes from LDS) ------------------------------
+  float o0 = 0.f, o1 = 0.f;
+  const int tbase = chunk * TPB + wave * LFMQ_WAVE;
+  const int nt = min(LFMQ_WAVE, T - tbase);
+  if (lane < dw) {
+#pragma unroll 8
+    for (int tt = 0; tt < LFMQ_WAVE; ++tt) {
+      if (tt >= nt) break;
+      const int tv = tbase + tt;
+      const int tvc = tv < tmax ? tv : tmax - 1;
+      const u32 vd = (tv == pos) ? v_sh[lane]
+                                 : vrow_cache[(u64)tvc * dw + lane];
+      float v0, v1;
+      bf16x2_to_f32(vd, v0, v1);
+      const float pv = p_sh[wave * LFMQ_WAVE + tt];
+      o0 = fmaf(pv, v0, o0);
+      o1 = fmaf(pv, v1, o1);
+    }
+    o_sh[wave][2 * lane] = o0;
+    o_sh[wave][2 * lane + 1] = o1;
+  }
+  __syncthreads();
+
+  if (nchunks == 1) {
+    const float invl = 1.f / l_sh;
+    if (threadIdx.x < dw) {
+      const int j = threadIdx.x;
+      float a0 = 0.f, a1 = 0.f;
+#pragma unroll
+      for (int w = 0; w < NWAVE; ++w) {
+        a0 += o_sh[w][2 * j];
+        a1 += o_sh[w][2 * j + 1];
+      }
+      o_direct[(u64)bh * dw + j] = pack_bf16x2(a0 * invl, a1 * invl);
+    }
+    return;
+  }
+  float* out = scratch + ((u64)bh * nchunks + chunk) * (D + 2);
+  for (int j = threadIdx.x; j < D; j += TPB) {
+    float v = 0.f;
+#pragma unroll
+    for (int w = 0; w < NWAVE; ++w) v += o_sh[w][j];
+    out[j] = v;
+  }
+  if (threadIdx.x == 0) {
+    out[D] = m;
+    out[D + 1] = l_sh;
+  }
+}
+
 template <int D>
 __global__ __launch_bounds__(128) void attn_decode_combine(
     const float* __restrict__ scratch, u32* __restrict__ o, int nchunks,
@@ -196,5 +373,37 @@ hipError_t lfmq_attn_decode(const void* q, const void* kc, const void* vc,
   else
     LFMQ_ATTN(64);
 #undef LFMQ_ATTN
+  return hipGetLastError();
+}
+
+// Fused rope+append+attention decode entry (s == 1).  qkv is the raw
+// [b, (hq+2hkv)*d] projection output; writes the new k/v cache row
+// itself.  scratch sized as lfmq_attn_decode.
+hipError_t lfmq_attn_decode_fused(const void* qkv, void* kc, void* vc,
+                                  void* o, void* scratch,
+                                  const void* inv_freq, int b, int hq,
+                                  int hkv, int tmax, int T, const void* T_dev,
+                                  int d, float scale, hipStream_t stream) {
+  if (d != 128 && d != 64) return hipErrorInvalidValue;
+  if (hq % hkv) return hipErrorInvalidValue;
+  const int nchunks = (T + TPB - 1) / TPB;
+  dim3 gridA(b * hq, nchunks), blockA(TPB);
+  dim3 gridB(b * hq), blockB(128);
+#define LFMQ_ATTNF(DD)                                                        \
+  do {                                                                        \
+    hipLaunchKernelGGL((attn_decode_fused<DD>), gridA, blockA, 0, stream,     \
+                       (const u32*)qkv, (u32*)kc, (u32*)vc, (float*)scratch,  \
+                       (u32*)o, (const float*)inv_freq, b, hq, hkv, tmax, T,  \
+                       (const i32*)T_dev, scale, nchunks);                    \
+    if (nchunks > 1)                                                          \
+      hipLaunchKernelGGL((attn_decode_combine<DD>), gridB, blockB, 0, stream, \
+                         (const float*)scratch, (u32*)o, nchunks, T,          \
+                         (const i32*)T_dev);                                  \
+  } while (0)
+  if (d == 128)
+    LFMQ_ATTNF(128);
+  else
+    LFMQ_ATTNF(64);
+#undef LFMQ_ATTNF
   return hipGetLastError();
 }

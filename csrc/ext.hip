@@ -26,6 +26,9 @@ hipError_t lfmq_kv_append(void*, void*, const void*, const void*, int, int,
 hipError_t lfmq_attn_decode(const void*, const void*, const void*, void*,
                             void*, int, int, int, int, int, const void*, int,
                             float, hipStream_t);
+hipError_t lfmq_attn_decode_fused(const void*, void*, void*, void*, void*,
+                                  const void*, int, int, int, int, int,
+                                  const void*, int, float, hipStream_t);
 hipError_t lfmq_hessian_accum(const void*, void*, int, int, hipStream_t);
 hipError_t lfmq_dq_gemv_gateup(const void*, const void*, const void*,
                                const void*, void*, int, int, int, int, int,
@@ -349,6 +352,30 @@ void advance_record(torch::Tensor state, torch::Tensor tok_buf,
                                 ring_len, stream()));
 }
 
+torch::Tensor attn_decode_fused(torch::Tensor qkv, torch::Tensor kc,
+                                torch::Tensor vc, long seq_len, double scale,
+                                torch::Tensor inv_freq, long hq,
+                                c10::optional<torch::Tensor> t_dev) {
+  check_t(qkv, torch::kBFloat16, "qkv");
+  check_t(kc, torch::kBFloat16, "k_cache");
+  check_t(vc, torch::kBFloat16, "v_cache");
+  check_t(inv_freq, torch::kFloat32, "inv_freq");
+  const int b = (int)kc.size(0), hkv = (int)kc.size(1);
+  const int tmax = (int)kc.size(2), d = (int)kc.size(3);
+  const int nchunks = ((int)seq_len + 255) / 256;
+  auto scratch = torch::empty({(long)b * hq * nchunks * (d + 2)},
+                              qkv.options().dtype(torch::kFloat32));
+  auto o = torch::empty({(long)b, 1L, hq * (long)d}, qkv.options());
+  const void* tptr = nullptr;
+  if (t_dev.has_value()) tptr = t_dev->data_ptr();
+  CHECK_HIP(lfmq_attn_decode_fused(qkv.data_ptr(), kc.data_ptr(),
+                                   vc.data_ptr(), o.data_ptr(),
+                                   scratch.data_ptr(), inv_freq.data_ptr(), b,
+                                   (int)hq, hkv, tmax, (int)seq_len, tptr, d,
+                                   (float)scale, stream()));
+  return o;
+}
+
 void hessian_accum(torch::Tensor h, torch::Tensor x) {
   check_t(h, torch::kFloat32, "H");
   check_t(x, torch::kBFloat16, "x");
@@ -377,6 +404,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("greedy_sample", &greedy_sample, "row argmax");
   m.def("read_probe", &read_probe, "bw probe");
   m.def("advance_record", &advance_record, "graph decode-state advance");
+  m.def("attn_decode_fused", &attn_decode_fused,
+        "fused rope+KV-append+attention decode");
   m.def("grouped_dq_gateup", &grouped_dq_gateup, "MoE grouped gate_up+silu");
   m.def("grouped_dq_linear", &grouped_dq_linear, "MoE grouped linear");
 }

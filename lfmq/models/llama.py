@@ -141,18 +141,17 @@ class LlamaAttention(nn.Module):
 
     def decode_qkv_attn(self, x, cache_k, cache_v, state, max_seq,
                         rms_eps: float = -1.0):
-        """qkv -> rope+append -> attention; returns [b, s, hq*d] (pre
-        o_proj).  rms_eps >= 0 fuses the input RMSNorm into the qkv GEMV
-        (norm weight pre-folded)."""
+        """qkv GEMV -> ONE fused rope+KV-append+attention launch; returns
+        [b, s, hq*d] (pre o_proj).  rms_eps >= 0 additionally fuses the
+        input RMSNorm into the qkv GEMV (norm weight pre-folded)."""
         b, s, _ = x.shape
         if rms_eps >= 0:
             qkv = self.qkv_proj.forward_rms(x, rms_eps)
         else:
             qkv = self.qkv_proj(x)
-        q = F.qkv_rope_append_graph(qkv, cache_k, cache_v, state,
-                                    self.n_heads, self.cfg.rope_theta)
-        o = F.attn_decode_graph(q.view(b, self.n_heads, 1, self.head_dim),
-                                cache_k, cache_v, state, max_seq, self.scale)
+        o = F.attn_decode_fused(qkv.view(b, -1), cache_k, cache_v, state,
+                                max_seq, self.scale, self.n_heads,
+                                self.cfg.rope_theta)
         return o.view(b, s, self.n_heads * self.head_dim)
 
     def _chunk_attn(self, q, cache_k, cache_v, pos0, seq_len_after):

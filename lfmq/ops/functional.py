@@ -144,6 +144,21 @@ def attn_decode_graph(q, k_cache, v_cache, state, max_t: int, scale: float):
     return hip_ext().attn_decode(q, k_cache, v_cache, max_t, scale, state[1:2])
 
 
+def attn_decode_fused(qkv, k_cache, v_cache, state_or_T, max_t: int,
+                      scale: float, hq: int, theta: float = 500000.0):
+    """One launch: rope(q,k) off the raw qkv buffer, KV-cache append,
+    chunked decode attention.  state_or_T: device state tensor (graph
+    mode, T = state[1]) or a host int T."""
+    d = k_cache.shape[-1]
+    inv = _inv_freq(d, theta, qkv.device)
+    if torch.is_tensor(state_or_T):
+        return hip_ext().attn_decode_fused(qkv.contiguous(), k_cache, v_cache,
+                                           max_t, scale, inv, hq,
+                                           state_or_T[1:2])
+    return hip_ext().attn_decode_fused(qkv.contiguous(), k_cache, v_cache,
+                                       int(state_or_T), scale, inv, hq, None)
+
+
 def dq_gemv_gateup(x: torch.Tensor, qweight: torch.Tensor,
                    scales: torch.Tensor, qzeros: torch.Tensor, bits: int,
                    group_size: int, rms_eps: float = -1.0) -> torch.Tensor:

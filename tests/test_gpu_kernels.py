@@ -325,3 +325,57 @@ def test_grouped_dq_kernels():
         y_ref[s_] = ref.dq_linear(hh[s_].float().unsqueeze(0), qw2[e],
                                   sc2[e].float(), qz2[e], 4, 32)
     assert rel_err(y, y_ref) < 2e-2
+
+
+def test_attn_decode_fused_kernel():
+    """Fused rope+append+attention vs the composed reference path."""
+    import lfmq_hip
+    b, hq, hkv, d, tmax, T = 2, 8, 2, 128, 96, 41
+    theta = 500000.0
+    torch.manual_seed(29)
+    qkv = torch.randn(b, (hq + 2 * hkv) * d).to(torch.bfloat16)
+    kc = torch.randn(b, hkv, tmax, d).to(torch.bfloat16)
+    vc = torch.randn(b, hkv, tmax, d).to(torch.bfloat16)
+    pos = torch.tensor([T - 1])
+    # reference: rope q/k, append, attn over T
+    qr, kr, vr = qkv.split([hq * d, hkv * d, hkv * d], dim=-1)
+    q_ref = ref.apply_rope(qr.reshape(b, 1, hq, d).float(), pos, theta)
+    k_ref = ref.apply_rope(kr.reshape(b, 1, hkv, d).float(), pos, theta)
+    kc_ref, vc_ref = kc.float().clone(), vc.float().clone()
+    kc_ref[:, :, T - 1] = k_ref[:, 0]
+    vc_ref[:, :, T - 1] = vr.reshape(b, 1, hkv, d)[:, 0].float()
+    import math as _m
+    o_ref = ref.attn_decode(q_ref.transpose(1, 2), kc_ref, vc_ref, T,
+                            1.0 / _m.sqrt(d))
+    # fused kernel
+    kcg, vcg = kc.to(DEV), vc.to(DEV)
+    inv = ref.rope_freqs(d, theta).to(DEV)
+    o = lfmq_hip.attn_decode_fused(qkv.to(DEV), kcg, vcg, T,
+                                   1.0 / _m.sqrt(d), inv, hq, None)
+    assert rel_err(o.reshape(b, hq, 1, d), o_ref) < 2e-2
+    # cache rows written
+    assert rel_err(kcg[:, :, T - 1], k_ref[:, 0].transpose(0, 1).transpose(0, 1)) < 2e-2
+    assert rel_err(vcg[:, :, T - 1],
+                   vr.reshape(b, hkv, d)) < 1e-6
+
+
+def test_attn_decode_fused_multichunk():
+    import lfmq_hip, math as _m
+    b, hq, hkv, d, tmax, T = 1, 8, 2, 128, 600, 500
+    torch.manual_seed(37)
+    qkv = torch.randn(b, (hq + 2 * hkv) * d).to(torch.bfloat16)
+    kc = torch.randn(b, hkv, tmax, d).to(torch.bfloat16)
+    vc = torch.randn(b, hkv, tmax, d).to(torch.bfloat16)
+    pos = torch.tensor([T - 1])
+    qr, kr, vr = qkv.split([hq * d, hkv * d, hkv * d], dim=-1)
+    q_ref = ref.apply_rope(qr.reshape(b, 1, hq, d).float(), pos)
+    k_ref = ref.apply_rope(kr.reshape(b, 1, hkv, d).float(), pos)
+    kc_ref, vc_ref = kc.float().clone(), vc.float().clone()
+    kc_ref[:, :, T - 1] = k_ref[:, 0]
+    vc_ref[:, :, T - 1] = vr.reshape(b, 1, hkv, d)[:, 0].float()
+    o_ref = ref.attn_decode(q_ref.transpose(1, 2), kc_ref, vc_ref, T,
+                            1.0 / _m.sqrt(d))
+    inv = ref.rope_freqs(d, 500000.0).to(DEV)
+    o = lfmq_hip.attn_decode_fused(qkv.to(DEV), kc.to(DEV), vc.to(DEV), T,
+                                   1.0 / _m.sqrt(d), inv, hq, None)
+    assert rel_err(o.reshape(b, hq, 1, d), o_ref) < 2e-2
