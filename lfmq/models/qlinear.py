@@ -71,11 +71,29 @@ class QLinear(nn.Module):
     @torch.no_grad()
     def dequant(self) -> torch.Tensor:
         """fp32 [n, k] — tests and AWQ search only."""
+        g_idx = getattr(self, "g_idx", None)
         return ref.dequant_weight(self.qweight.cpu(), self.scales.cpu(),
                                   self.qzeros.cpu(), self.bits, self.group_size,
-                                  self.in_features)
+                                  self.in_features,
+                                  g_idx.cpu() if g_idx is not None else None)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if getattr(self, "g_idx", None) is not None:
+            # act-order (g_idx) layout: per-column group indices are not
+            # supported by the fused GPU kernels in this round — run the
+            # exact dequant+matmul path (loud, not a silent wrong answer).
+            if x.is_cuda:
+                raise NotImplementedError(
+                    "act_order (g_idx) checkpoints decode on the CPU path "
+                    "only in this round; re-quantize with act_order=False "
+                    "for the fused GPU kernels")
+            w = ref.dequant_weight(self.qweight, self.scales, self.qzeros,
+                                   self.bits, self.group_size,
+                                   self.in_features, self.g_idx)
+            y = x.float() @ w.t()
+            if self.bias is not None:
+                y = y + self.bias.float()
+            return y.to(x.dtype)
         return F.dq_linear(x, self.qweight, self.scales, self.qzeros,
                            self.bits, self.group_size, self.bias)
 

@@ -139,12 +139,18 @@ def unpack_zeros(packed: torch.Tensor, bits: int, G: int) -> torch.Tensor:
 def dequant_weight(
     qweight: torch.Tensor, scales: torch.Tensor, qzeros: torch.Tensor,
     bits: int, group_size: int, k: int,
+    g_idx: "Optional[torch.Tensor]" = None,
 ) -> torch.Tensor:
-    """Packed tensors -> fp32 weight [n, k] (W[n,k]: out x in, torch Linear layout)."""
+    """Packed tensors -> fp32 weight [n, k] (W[n,k]: out x in, torch Linear
+    layout).  g_idx (act-order GPTQ): per-column group index."""
     n = qweight.shape[0]
     G = scales.shape[1]
     codes = unpack_codes(qweight, bits, k)
     zeros = unpack_zeros(qzeros, bits, G)
+    if g_idx is not None:
+        gi = g_idx.long()
+        return ((codes.float() - zeros.float()[:, gi])
+                * scales.float()[:, gi])
     g = k if group_size == -1 else group_size
     return dequant_groupwise(codes, scales.float(), zeros, g)
 
