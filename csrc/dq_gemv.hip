@@ -36,7 +36,8 @@ namespace {
 constexpr int WAVES_PER_WG = 4;
 constexpr u32 ONE2_BF16 = 0x3F803F80u;  // bf16x2 (1.0, 1.0)
 
-template <int M, int BITS, int COLS, bool GATEUP, int KI, bool RMS>
+template <int M, int BITS, int COLS, bool GATEUP, int KI, bool RMS,
+          bool KSPLIT = false>
 __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
     const u32* __restrict__ qweight, const u32* __restrict__ scales_u16,
     const u32* __restrict__ qzeros, const u32* __restrict__ x,
@@ -45,7 +46,11 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
   static_assert(!GATEUP || COLS == 4 || COLS == 2, "gateup needs COLS 2/4");
   const int wave = threadIdx.x / LFMQ_WAVE;
   const int lane = threadIdx.x % LFMQ_WAVE;
-  const int cbase = (blockIdx.x * WAVES_PER_WG + wave) *
+  // KSPLIT: all 4 waves share the SAME columns and split the k-iteration
+  // range (the decode GEMV is latency-bound on the per-wave serial
+  // k-chain — SQ_WAIT_ANY 62% measured); partials combine through LDS.
+  const int cbase = (KSPLIT ? (int)blockIdx.x
+                            : ((int)blockIdx.x * WAVES_PER_WG + wave)) *
                     (GATEUP ? COLS / 2 : COLS);
   int col[COLS];
 #pragma unroll
@@ -118,8 +123,14 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
   for (int mm = 0; mm < M; ++mm) ss_acc[mm] = 0.f;
 
   const int kiters = (k + 2047) / 2048;
+  int it_begin = 0, it_end = kiters;
+  if (KSPLIT) {
+    const int per_wave = (kiters + WAVES_PER_WG - 1) / WAVES_PER_WG;
+    it_begin = wave * per_wave;
+    it_end = min(kiters, it_begin + per_wave);
+  }
 
-  if (BITS == 4 && KI > 0) {
+  if (BITS == 4 && KI > 0 && !KSPLIT) {
     // Fully-unrolled k-loop: per-iteration (cx, xs) partials accumulate in
     // registers; scales are applied ONLY in the epilogue, so the hoisted
     // scale/zero preload never blocks the weight stream (probe.hip mode 2
@@ -197,7 +208,7 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
       }
     }
   } else
-  for (int it = 0; it < kiters; ++it) {
+  for (int it = it_begin; it < it_end; ++it) {
     const int k0 = it * 2048 + lane * 32;
     const int gi = k0 < k ? k0 / g : 0;  // OOB lanes read s = 0 anyway
     const bool ghi = gi >= 64;
@@ -316,6 +327,54 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
   }
 
   // ---- wave-reduce and store -----------------------------------------
+  __shared__ float part_sh[WAVES_PER_WG][COLS][M];
+  __shared__ float ss_sh[WAVES_PER_WG][M];
+  if (KSPLIT) {
+#pragma unroll
+    for (int mm = 0; mm < M; ++mm) {
+      if (mm >= m) break;
+      if (RMS) {
+        const float ssw = wave_reduce_sum(ss_acc[mm]);
+        if (lane == 0) ss_sh[wave][mm] = ssw;
+      }
+#pragma unroll
+      for (int c = 0; c < COLS; ++c) {
+        const float v = wave_reduce_sum(acc[c][mm]);
+        if (lane == 0) part_sh[wave][c][mm] = v;
+      }
+    }
+    __syncthreads();
+    if (wave != 0) return;
+#pragma unroll
+    for (int mm = 0; mm < M; ++mm) {
+      if (mm >= m) break;
+      float inv = 1.f;
+      if (RMS) {
+        float ss = 0.f;
+#pragma unroll
+        for (int w = 0; w < WAVES_PER_WG; ++w) ss += ss_sh[w][mm];
+        inv = rsqrtf(ss / (float)k + eps);
+      }
+      if (lane == 0) {
+        unsigned short* yrow = (unsigned short*)y;
+#pragma unroll
+        for (int c = 0; c < COLS; ++c) {
+          if (col[c] >= n) continue;
+          float v = 0.f;
+#pragma unroll
+          for (int w = 0; w < WAVES_PER_WG; ++w) v += part_sh[w][c][mm];
+          v *= inv;
+          if (bias != nullptr)
+            v += bf16_bits_to_f32(((const unsigned short*)bias)[col[c]]);
+          if (resid != nullptr)
+            v += bf16_bits_to_f32(
+                ((const unsigned short*)resid)[(u64)mm * n + col[c]]);
+          yrow[(u64)mm * n + col[c]] = (unsigned short)f32_to_bf16_bits_rne(v);
+        }
+      }
+    }
+    return;
+  }
 #pragma unroll
   for (int mm = 0; mm < M; ++mm) {
     if (mm >= m) break;
@@ -386,6 +445,18 @@ void launch_one(dim3 grid, dim3 block, hipStream_t stream, const void* qweight,
 #undef LFMQ_KI
 }
 
+template <int BITS, int COLS, bool GATEUP, int M, bool RMS>
+void launch_ksplit(dim3 grid, dim3 block, hipStream_t stream,
+                   const void* qweight, const void* scales, const void* qzeros,
+                   const void* x, void* y, const void* bias, const void* resid,
+                   int m, int n, int k, int g, float eps) {
+  hipLaunchKernelGGL((dq_gemv_kernel<M, BITS, COLS, GATEUP, 0, RMS, true>),
+                     grid, block, 0, stream, (const u32*)qweight,
+                     (const u32*)scales, (const u32*)qzeros, (const u32*)x,
+                     (u32*)y, (const u32*)bias, (const u32*)resid, m, n, k, g,
+                     eps);
+}
+
 template <int BITS, bool GATEUP, bool RMS>
 hipError_t launch_bits(const void* qweight, const void* scales,
                        const void* qzeros, const void* x, void* y,
@@ -398,7 +469,24 @@ hipError_t launch_bits(const void* qweight, const void* scales,
     // are occupancy/latency-bound (probe.hip: the inner loop itself
     // sustains >6 TB/s at large grids).  cols_hint 0 = auto.
     int cols = cols_hint;
-    if (cols == 0) cols = (GATEUP || n >= 16384) ? 4 : 2;
+    const int kiters = (k + 2047) / 2048;
+    // measured: wg-internal split-k does NOT beat the multi-column wave
+    // layout on the decode shapes (LDS combine + fewer weight streams
+    // offset the shorter latency chain) — kept behind hint 5 for A/B.
+    const bool want_ksplit = !GATEUP && BITS == 4 && kiters >= 2 &&
+                             cols_hint == 5;
+    if (want_ksplit) {
+      // wg-internal split-k: the 4 waves share 4 columns
+      dim3 gridk(ceil_div_u32(n, 4));
+      if (m <= 1)
+        launch_ksplit<BITS, 4, false, 1, RMS>(gridk, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
+      else if (m <= 2)
+        launch_ksplit<BITS, 4, false, 2, RMS>(gridk, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
+      else
+        launch_ksplit<BITS, 4, false, 4, RMS>(gridk, block, stream, qweight, scales, qzeros, x, y, bias, resid, m, n, k, g, eps);
+      return hipGetLastError();
+    }
+    if (cols == 0 || cols == 5) cols = (GATEUP || n >= 16384) ? 4 : 2;
     const int per_wg = WAVES_PER_WG * (GATEUP ? cols / 2 : cols);
     dim3 grid(ceil_div_u32(GATEUP ? n / 2 : n, per_wg));
 #define LFMQ_M(CC)                                                            \
