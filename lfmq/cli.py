@@ -101,9 +101,8 @@ def eval(ckpt: str = typer.Option(...),
     fp = fp.to(dev)
     r = bench_ppl(fp, qm, qm.cfg.vocab_size, seq_len=seq_len, n_seqs=n_seqs,
                   seed=seed, device=dev)
-    r["weight_mse_mean"] = (
-        sum(weight_mse(fp.cpu(), qm.cpu()).values()) /
-        max(len(weight_mse(fp, qm)), 1))
+    mse = weight_mse(fp.cpu(), qm.cpu())
+    r["weight_mse_mean"] = sum(mse.values()) / max(len(mse), 1)
     typer.echo(json.dumps(r, indent=2))
 
 
