@@ -135,14 +135,24 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void dq_gemv_kernel(
     // registers; scales are applied ONLY in the epilogue, so the hoisted
     // scale/zero preload never blocks the weight stream (probe.hip mode 2
     // shows the full dequant+dot loop sustains >6 TB/s once unserialized).
-    float cxa[COLS][M][KI], xsa[M][KI];
+    float cxa[COLS][M][KI > 0 ? KI : 1], xsa[M][KI > 0 ? KI : 1];
+    // issue EVERY iteration's weight loads up front: with per-iteration
+    // loads the compiler reuses the W registers and the loop serializes
+    // on memory latency (SQ_WAIT_ANY 62% measured on the k=14336 shape)
+    u32x4 w_all[KI > 0 ? KI : 1][COLS];
 #pragma unroll
     for (int it = 0; it < KI; ++it) {
       const u32 wbyte = (u32)(it * 1024 + lane * 16);
-      u32x4 w[COLS];
 #pragma unroll
       for (int c = 0; c < COLS; ++c)
-        w[c] = __builtin_amdgcn_raw_buffer_load_b128(w_rsrc[c], wbyte, 0, 2);
+        w_all[it][c] =
+            __builtin_amdgcn_raw_buffer_load_b128(w_rsrc[c], wbyte, 0, 2);
+    }
+#pragma unroll
+    for (int it = 0; it < KI; ++it) {
+      u32x4 w[COLS];
+#pragma unroll
+      for (int c = 0; c < COLS; ++c) w[c] = w_all[it][c];
 #pragma unroll
       for (int mm = 0; mm < M; ++mm) {
         if (mm >= m) break;

@@ -317,7 +317,8 @@ class LlamaForCausalLM(nn.Module):
         x = self.embed_tokens(input_ids)
         if getattr(self, "_fused_decode", None) is None:
             self._fused_decode = self._decode_fused_ready()
-        if self._fused_decode:
+        # the rms/silu-fused GEMV variants are m <= 4 kernels
+        if self._fused_decode and input_ids.shape[0] <= 4:
             for i, layer in enumerate(self.layers):
                 x = layer.forward_decode_fused(x, cache.k[i], cache.v[i],
                                                state, max_seq)

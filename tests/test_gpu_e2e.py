@@ -137,3 +137,38 @@ def test_mixtral_gpu_decode():
     assert rel < 0.05, float(rel)
     out = qg.generate(ids.to(DEV), max_new_tokens=6)
     assert out.shape == (1, 14)
+
+
+def test_graph_decoder_batch8():
+    """batch-8 decode (secondary BASELINE metric shape) through the graph
+    path (fused kernels are m<=4: the b=8 step must take the general
+    path, not crash)."""
+    from lfmq.models.graph import GraphDecoder
+    cfg = LlamaConfig(vocab_size=1024, hidden_size=512, intermediate_size=1024,
+                      n_layers=2, n_heads=8, n_kv_heads=2, head_dim=64,
+                      max_seq=128)
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=64, fold_rms=False)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=23), qcfg).to(DEV)
+    ids = torch.randint(0, cfg.vocab_size, (8, 10), device=DEV)
+    eager = qm.generate(ids, max_new_tokens=6)[:, 10:]
+    dec = GraphDecoder(qm, batch=8, max_seq=64, max_new_tokens=8)
+    dec.prefill(ids)
+    got = dec.decode(6)
+    assert torch.equal(got.cpu(), eager.cpu())
+
+
+def test_w8_decode_gpu():
+    """W8A16 end-to-end decode on GPU (config class: w8 kernels wired)."""
+    cfg = LlamaConfig(vocab_size=512, hidden_size=512, intermediate_size=1024,
+                      n_layers=2, n_heads=8, n_kv_heads=2, head_dim=64,
+                      max_seq=64)
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=8, group_size=128)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=27), qcfg)
+    ids = torch.randint(0, cfg.vocab_size, (1, 8))
+    l_cpu = qm(ids, qm.make_cache(1, 32)).float()
+    qg = qm.to(DEV)
+    l_gpu = qg(ids.to(DEV), qg.make_cache(1, 32)).float()
+    rel = (l_gpu.cpu() - l_cpu).abs().max() / l_cpu.abs().max()
+    assert rel < 0.05, float(rel)
+    out = qg.generate(ids.to(DEV), max_new_tokens=6)
+    assert out.shape == (1, 14)
