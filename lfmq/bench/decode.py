@@ -38,18 +38,27 @@ def build_quantized_llama(model_name: str, scheme: str, bits: int, g: int,
         cfg = LlamaConfig.llama3_8b()
     elif model_name == "llama3-70b":
         cfg = LlamaConfig.llama3_70b()
+    elif model_name == "mixtral-8x7b":
+        from ..models.mixtral import MixtralConfig
+        cfg = MixtralConfig.mixtral_8x7b()
     else:
         raise ValueError(model_name)
     if n_layers:
         cfg.n_layers = n_layers
     model = build_model(cfg)
-    model = model.to_empty(device=device) if device != "cpu" else model
+    model = model.to_empty(device=str(device)) if str(device) != "cpu" else model
     _fast_random_init(model, seed)
     qcfg = QuantConfig(scheme=scheme, bits=bits, group_size=g)
     model = quantize(model, qcfg)
     if tp_world > 1:
-        from ..dist.tp import shard_llama
-        model = shard_llama(model, tp_rank, tp_world)
+        if cfg.arch == "mixtral":
+            from ..dist.ep import apply_expert_parallel
+            model = apply_expert_parallel(model, tp_rank, tp_world)
+        else:
+            from ..dist.tp import shard_llama
+            model = shard_llama(model, tp_rank, tp_world)
+    if cfg.arch == "mixtral" and hasattr(model, "consolidate_moe"):
+        model.consolidate_moe()
     return model, cfg, qcfg
 
 
