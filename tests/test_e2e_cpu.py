@@ -87,3 +87,20 @@ def test_quantized_decode_prefill_consistency():
         outs.append(qm(ids[:, t:t + 1], cache).float())
     ld = torch.cat(outs, dim=1)
     assert torch.allclose(lp, ld, atol=2e-2, rtol=1e-2)
+
+
+def test_opt_awq_e2e(tmp_path):
+    """AWQ on the OPT arch (LayerNorm/v_proj/fc1 scale plan)."""
+    from lfmq.quant.calib import CalibStream
+    cfg = OPTConfig.tiny()
+    fp = build_model(cfg, seed=14)
+    qcfg = QuantConfig(scheme="awq", bits=4, group_size=32, awq_grid=6,
+                       calib_n_seqs=2, calib_seq_len=32)
+    calib = CalibStream(cfg.vocab_size, 32, 2, qcfg.calib_seed)
+    qm = quantize(build_model(cfg, seed=14), qcfg, calib=calib)
+    ids = torch.randint(0, cfg.vocab_size, (1, 10))
+    lf = fp(ids, fp.make_cache(1, 24)).float()
+    lq = qm(ids, qm.make_cache(1, 24)).float()
+    pf = torch.log_softmax(lf, -1); pq = torch.log_softmax(lq, -1)
+    kl = (pf.exp() * (pf - pq)).sum(-1).mean()
+    assert kl < 0.5, float(kl)

@@ -84,3 +84,24 @@ def test_row_shard_group_divisibility_guard():
     q = QLinear(96, 8, 4, 96)  # k=96, g=96: k/2=48 not divisible by 96
     with pytest.raises(AssertionError):
         _shard_row(q, 0, 2)
+
+
+@pytest.mark.parametrize("world", [4])
+def test_tp_gloo_world4(tmp_path, world):
+    """4-way TP numerics (needs n_kv_heads % 4 == 0)."""
+    cfg = LlamaConfig(vocab_size=256, hidden_size=256, intermediate_size=512,
+                      n_layers=2, n_heads=8, n_kv_heads=4, head_dim=32,
+                      max_seq=64)
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=64)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=41), qcfg)
+    path = str(tmp_path / "ckpt4")
+    lfmq.save(qm, path, qcfg)
+    ids = torch.arange(12).remainder(256).reshape(1, 12)
+    ref_logits = qm(ids, qm.make_cache(1, 32))
+    port = 29650
+    mp.spawn(_worker, args=(world, path, port, str(tmp_path)), nprocs=world,
+             join=True)
+    out = torch.load(str(tmp_path / "tp_out.pt"), weights_only=False)
+    rel = (out["logits"].float() - ref_logits.float()).abs().max() / \
+        ref_logits.float().abs().max()
+    assert rel < 0.05, f"TP4 logits rel err {float(rel)}"
