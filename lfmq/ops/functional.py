@@ -9,11 +9,23 @@ dispatch: per device there is exactly one implementation.
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional
 
 import torch
 
 from . import hip_ext, ref
+
+# Debug numeric guard: LFMQ_NANCHECK=1 asserts every op
+# output is finite — catches kernel-layout or quantization bugs at the
+# op that produced them instead of 30 layers later.
+_NANCHECK = os.environ.get("LFMQ_NANCHECK", "0") == "1"
+
+
+def _guard(t: torch.Tensor, name: str) -> torch.Tensor:
+    if _NANCHECK and not bool(torch.isfinite(t.float()).all()):
+        raise FloatingPointError(f"lfmq op {name} produced non-finite values")
+    return t
 
 
 def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5,
@@ -68,7 +80,7 @@ def dq_linear(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,
         r2 = resid.reshape(-1, resid.shape[-1]).contiguous() if resid is not None else None
         y = hip_ext().dq_linear(x2, qweight, scales, qzeros, bits, group_size,
                                 bias, r2, rms_eps, 0)
-        return y.reshape(*x.shape[:-1], y.shape[-1])
+        return _guard(y, "dq_linear").reshape(*x.shape[:-1], y.shape[-1])
     assert resid is None and rms_eps < 0, "fused rms/resid is GPU-only"
     return ref.dq_linear(x, qweight, scales, qzeros, bits, group_size, bias).to(x.dtype)
 
@@ -90,9 +102,9 @@ def silu_mul_gateup(gate_up: torch.Tensor, inter: int) -> torch.Tensor:
 
 def attn_decode(q, k_cache, v_cache, seq_len: int, scale: Optional[float] = None):
     if q.is_cuda:
-        return hip_ext().attn_decode(
+        return _guard(hip_ext().attn_decode(
             q, k_cache, v_cache, seq_len,
-            scale or 1.0 / math.sqrt(q.shape[-1]), None)
+            scale or 1.0 / math.sqrt(q.shape[-1]), None), "attn_decode")
     return ref.attn_decode(q, k_cache, v_cache, seq_len, scale)
 
 

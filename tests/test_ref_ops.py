@@ -168,3 +168,14 @@ def test_grouped_dq_linear():
     for i in range(t):
         yi = ref.dq_linear(x[i:i+1], qw[eid[i]], s[eid[i]], z[eid[i]], 4, 32)
         assert torch.allclose(y[i:i+1], yi, atol=1e-5)
+
+
+def test_nancheck_guard(monkeypatch):
+    """LFMQ_NANCHECK flag wiring (SURVEY §5.2 numeric guards)."""
+    import importlib
+    import lfmq.ops.functional as F
+    monkeypatch.setattr(F, "_NANCHECK", True)
+    t = torch.tensor([1.0, float("nan")])
+    with pytest.raises(FloatingPointError):
+        F._guard(t, "test")
+    F._guard(torch.ones(3), "ok")
