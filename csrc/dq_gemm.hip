@@ -29,7 +29,7 @@ __global__ __launch_bounds__(TPB, 1) void dq_gemm_kernel(
     const u32* __restrict__ qzeros, const u32* __restrict__ x,
     u32* __restrict__ y, const u32* __restrict__ bias, int m, int n, int k,
     int g) {
-  // LDS: A [BM][BK] bf16 + B [BN][BK] bf16 (16 KiB each), one buffer
+  // LDS: 2 x (A [BM][BK] + B [BN][BK]) bf16 (double-buffered, 64 KiB)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* ldsA = smem;
   char* ldsB = smem + BM * BK * 2;
@@ -66,59 +66,67 @@ __global__ __launch_bounds__(TPB, 1) void dq_gemm_kernel(
     for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int ksteps = k / BK;
-  for (int kt = 0; kt < ksteps; ++kt) {
-    // ---- stage A: x[arow_g][kt*64 + slot*8 .. +8] -> ldsA[ar][..] ----
-    // 128 rows x 8 slots of 16B = 1024 slot-tasks over 512 threads: each
-    // thread stages slots aslot and aslot+4 of its row.
+
+  // Double-buffered T14 schedule (guide §6 G15): per k-tile, ONE barrier;
+  // the next tile's global loads are ISSUED before the MFMAs and the
+  // LDS writes land after them into the other buffer (legal: that
+  // buffer's last readers passed this iteration's barrier).
+  char* ldsA2[2] = {ldsA, ldsA + (BM + BN) * BK * 2};
+  char* ldsB2[2] = {ldsB, ldsB + (BM + BN) * BK * 2};
+
+  auto stage_a_load = [&](int kt, u32x4 va[2]) {
 #pragma unroll
     for (int s2 = 0; s2 < 2; ++s2) {
       const int slot = aslot + 4 * s2;
       const u32 gbyte = (u32)(((u64)arow_g * k + kt * BK + slot * 8) * 2);
-      u32x4 v = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, gbyte, 0, 0);
-      *(u32x4*)(ldsA + swz((u32)(ar * BK * 2 + slot * 16), ar)) = v;
+      va[s2] = __builtin_amdgcn_raw_buffer_load_b128(x_rsrc, gbyte, 0, 0);
     }
-    // ---- stage B: dequant 16 codes of row bcol_g -> ldsB[bc][..] ------
-    {
-      const int kbase = kt * BK + bpair * 16;  // 16 k-codes per thread
-      const int gi = kbase / g;                // single group (g % 64 == 0)
-      const float s = bf16_bits_to_f32(
-          ((const unsigned short*)scales_u16)[(u64)bcol_g * G + gi]);
-      const u32 zword = qzeros[(u64)bcol_g * zw + gi / zper];
-      const float zb =
-          BITS == 4 ? (float)(unpack4(zword, gi & 7) + 128u)
-                    : (float)unpack8(zword, gi & 3);
-      u32 out[8];
-      if (BITS == 4) {
-        const u32* wrow = qweight + (u64)bcol_g * (k / 8);
-        const u32 w0 = wrow[kbase / 8], w1 = wrow[kbase / 8 + 1];
-        DequantPairs4 d0 = dequant4_pairs(w0), d1 = dequant4_pairs(w1);
-        const u32 ps[8] = {d0.p01, d0.p23, d0.p45, d0.p67,
-                           d1.p01, d1.p23, d1.p45, d1.p67};
+  };
+  auto stage_a_write = [&](int buf, const u32x4 va[2]) {
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          float lo, hi;
-          bf16x2_to_f32(ps[i], lo, hi);
-          out[i] = pack_bf16x2(fmaf(s, lo, -s * zb), fmaf(s, hi, -s * zb));
-        }
-      } else {
-        const u32* wrow = qweight + (u64)bcol_g * (k / 4);
+    for (int s2 = 0; s2 < 2; ++s2) {
+      const int slot = aslot + 4 * s2;
+      *(u32x4*)(ldsA2[buf] + swz((u32)(ar * BK * 2 + slot * 16), ar)) = va[s2];
+    }
+  };
+  auto stage_b = [&](int kt, int buf) {
+    const int kbase = kt * BK + bpair * 16;
+    const int gi = kbase / g;
+    const float s = bf16_bits_to_f32(
+        ((const unsigned short*)scales_u16)[(u64)bcol_g * G + gi]);
+    const u32 zword = qzeros[(u64)bcol_g * zw + gi / zper];
+    const float zb = BITS == 4 ? (float)(unpack4(zword, gi & 7) + 128u)
+                               : (float)unpack8(zword, gi & 3);
+    u32 out[8];
+    if (BITS == 4) {
+      const u32* wrow = qweight + (u64)bcol_g * (k / 8);
+      const u32 w0 = wrow[kbase / 8], w1 = wrow[kbase / 8 + 1];
+      DequantPairs4 d0 = dequant4_pairs(w0), d1 = dequant4_pairs(w1);
+      const u32 ps[8] = {d0.p01, d0.p23, d0.p45, d0.p67,
+                         d1.p01, d1.p23, d1.p45, d1.p67};
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          const u32 wv = wrow[kbase / 4 + i];
-          const float c0 = (float)(wv & 0xFFu), c1 = (float)((wv >> 8) & 0xFFu);
-          const float c2 = (float)((wv >> 16) & 0xFFu), c3 = (float)(wv >> 24);
-          out[2 * i] = pack_bf16x2(s * (c0 - zb), s * (c1 - zb));
-          out[2 * i + 1] = pack_bf16x2(s * (c2 - zb), s * (c3 - zb));
-        }
+      for (int i = 0; i < 8; ++i) {
+        float lo, hi;
+        bf16x2_to_f32(ps[i], lo, hi);
+        out[i] = pack_bf16x2(fmaf(s, lo, -s * zb), fmaf(s, hi, -s * zb));
       }
-      char* dst = ldsB + swz((u32)(bc * BK * 2 + bpair * 32), bc);
-      *(u32x4*)dst = u32x4{out[0], out[1], out[2], out[3]};
-      *(u32x4*)(ldsB + swz((u32)(bc * BK * 2 + bpair * 32 + 16), bc)) =
-          u32x4{out[4], out[5], out[6], out[7]};
+    } else {
+      const u32* wrow = qweight + (u64)bcol_g * (k / 4);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const u32 wv = wrow[kbase / 4 + i];
+        const float c0 = (float)(wv & 0xFFu), c1 = (float)((wv >> 8) & 0xFFu);
+        const float c2 = (float)((wv >> 16) & 0xFFu), c3 = (float)(wv >> 24);
+        out[2 * i] = pack_bf16x2(s * (c0 - zb), s * (c1 - zb));
+        out[2 * i + 1] = pack_bf16x2(s * (c2 - zb), s * (c3 - zb));
+      }
     }
-    __syncthreads();
-
-    // ---- MFMA: 2 k-steps of 32 -------------------------------------
+    char* dst0 = ldsB2[buf] + swz((u32)(bc * BK * 2 + bpair * 32), bc);
+    *(u32x4*)dst0 = u32x4{out[0], out[1], out[2], out[3]};
+    *(u32x4*)(ldsB2[buf] + swz((u32)(bc * BK * 2 + bpair * 32 + 16), bc)) =
+        u32x4{out[4], out[5], out[6], out[7]};
+  };
+  auto mfma_tile = [&](int buf) {
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       const int k8 = ks * 32 + (lane >> 4) * 8;
@@ -126,12 +134,12 @@ __global__ __launch_bounds__(TPB, 1) void dq_gemm_kernel(
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int row = wm * 64 + i * 16 + (lane & 15);
-        a[i] = *(frag_ab*)(ldsA + swz((u32)(row * BK * 2 + k8 * 2), row));
+        a[i] = *(frag_ab*)(ldsA2[buf] + swz((u32)(row * BK * 2 + k8 * 2), row));
       }
 #pragma unroll
       for (int j = 0; j < 2; ++j) {
         const int col = wn * 32 + j * 16 + (lane & 15);
-        b[j] = *(frag_ab*)(ldsB + swz((u32)(col * BK * 2 + k8 * 2), col));
+        b[j] = *(frag_ab*)(ldsB2[buf] + swz((u32)(col * BK * 2 + k8 * 2), col));
       }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
@@ -140,7 +148,25 @@ __global__ __launch_bounds__(TPB, 1) void dq_gemm_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[i], b[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
+  };
+
+  {
+    u32x4 va[2];
+    stage_a_load(0, va);
+    stage_a_write(0, va);
+    stage_b(0, 0);
+  }
+  int cur = 0;
+  for (int kt = 0; kt < ksteps; ++kt) {
+    __syncthreads();  // buf[cur] staged; buf[cur^1]'s last readers done
+    u32x4 va[2];
+    if (kt + 1 < ksteps) stage_a_load(kt + 1, va);  // issue early
+    mfma_tile(cur);
+    if (kt + 1 < ksteps) {                          // write late
+      stage_a_write(cur ^ 1, va);
+      stage_b(kt + 1, cur ^ 1);
+    }
+    cur ^= 1;
   }
 
   // ---- epilogue: C[m0+..][n0+..] (C/D map: col=l&15, row=(l>>4)*4+r) --
@@ -173,7 +199,7 @@ hipError_t lfmq_dq_gemm(const void* qweight, const void* scales,
   if (k % BK || g % BK || g > k) return hipErrorInvalidValue;
   const u32 nbx = ceil_div_u32(n, BN), nby = ceil_div_u32(m, BM);
   dim3 grid(nbx * nby), block(TPB);
-  const size_t lds = (BM + BN) * BK * 2;
+  const size_t lds = 2 * (BM + BN) * BK * 2;  // double-buffered
   if (bits == 4)
     hipLaunchKernelGGL((dq_gemm_kernel<4>), grid, block, lds, stream,
                        (const u32*)qweight, (const u32*)scales,

@@ -62,6 +62,19 @@ def main():
     print("== w8 ==")
     bench_gemv(1, 4096, 14336, 128, bits=8)
 
+    print("== dq_gemm prefill shapes ==")
+    ext = hip_ext()
+    for (m, n, k) in [(2048, 6144, 4096), (2048, 4096, 14336)]:
+        per, g = 8, 128
+        G = k // g
+        qw = torch.randint(-2**31, 2**31 - 1, (n, k // per), dtype=torch.int32, device=DEV)
+        sc = (torch.rand(n, G, device=DEV) * 0.01).to(torch.bfloat16)
+        qz = torch.randint(-2**31, 2**31 - 1, (n, (G + 7) // 8), dtype=torch.int32, device=DEV)
+        x = torch.randn(m, k, device=DEV).to(torch.bfloat16)
+        t = timeit(lambda: ext.dq_linear(x, qw, sc, qz, 4, g, None, None, -1.0, 0), iters=20, warmup=5)
+        tf = 2 * m * n * k / t / 1e12
+        print(f"gemm m={m} n={n} k={k}: {t*1e6:8.1f} us  {tf:6.1f} TF/s")
+
     print("== attn_decode (b=1, hq=32, hkv=8, d=128) ==")
     ext = hip_ext()
     for T in (256, 640, 2048, 8192):
