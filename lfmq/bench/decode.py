@@ -81,6 +81,10 @@ def bench_decode(model, cfg, batch: int, prompt: int, steps: int, warmup: int,
             import sys
             print(f"[bench] hipGraph capture failed ({e}); eager decode",
                   file=sys.stderr)
+            try:
+                torch.cuda.synchronize()  # leave no half-captured stream state
+            except Exception:
+                pass
     if not graphed:
         cache = model.make_cache(batch, max_seq=prompt + warmup + steps + 8,
                                  device=device)

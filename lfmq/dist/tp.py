@@ -177,9 +177,10 @@ def load_tp(path: str, meta: dict, device: str, rank: int, world: int,
         field = tname.rsplit(".", 1)[1]
         if mod_name in col_parts:
             if field in ("qweight", "scales", "qzeros", "bias", "weight"):
+                # safetensors lazy slices need contiguous ranges: single-part
+                # shards slice at read time; multi-part (fused qkv/gate_up)
+                # read whole and slice in the fix-up pass below
                 sls = _row_slices(col_parts[mod_name], rank, world)
-                idx = torch.cat([torch.arange(s.start, s.stop) for s in sls])
-                # safetensors slices need contiguous ranges; do per-part reads
                 return None if len(sls) > 1 else (sls[0],)
         if mod_name in row_mods:
             mod = get_module(model, mod_name)
