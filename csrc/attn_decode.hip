@@ -145,6 +145,8 @@ __global__ __launch_bounds__(TPB) void attn_decode_partial(
 // depends on another workgroup's cache write), then runs the same
 // chunked flash-decoding as attn_decode_partial.  Replaces the separate
 // qkv_rope_append launch on the decode path.
+constexpr int FCH = 64;  // fused-path chunk: short PV chains, big grid
+
 template <int D>
 __global__ __launch_bounds__(TPB) void attn_decode_fused(
     const u32* __restrict__ qkv, u32* __restrict__ kc, u32* __restrict__ vc,
@@ -154,7 +156,7 @@ __global__ __launch_bounds__(TPB) void attn_decode_fused(
   if (T_dev != nullptr) T = *T_dev;
   const int bh = blockIdx.x;
   const int chunk = blockIdx.y;
-  if (chunk * TPB >= T) return;
+  if (chunk * FCH >= T) return;
   const int bb = bh / hq;
   const int h = bh % hq;
   const int rep = hq / hkv;
@@ -208,7 +210,7 @@ __global__ __launch_bounds__(TPB) void attn_decode_fused(
       slast_sh = t;
     }
     // designated writer appends the new k/v row to the cache
-    if (h % rep == 0 && chunk == pos / TPB) {
+    if (h % rep == 0 && chunk == pos / FCH) {
       if (threadIdx.x < dw) {
         const unsigned short* ks = k_sh;
         krow_cache[(u64)pos * dw + threadIdx.x] =
@@ -219,11 +221,13 @@ __global__ __launch_bounds__(TPB) void attn_decode_fused(
     __syncthreads();
   }
 
-  // ---- q·K over this chunk (q from LDS; t == pos uses the reduction) --
-  const int t = chunk * TPB + threadIdx.x;
-  const int t_clamped = t < tmax ? t : tmax - 1;
-  float s = -INFINITY;
-  {
+  // ---- q·K over this 64-token chunk (wave 0 only; q from LDS) --------
+  const int wave = threadIdx.x / LFMQ_WAVE;
+  const int lane = threadIdx.x % LFMQ_WAVE;
+  const int t = chunk * FCH + lane;
+  if (wave == 0) {
+    const int t_clamped = t < tmax ? t : tmax - 1;
+    float s = -INFINITY;
     const u32* krow = (const u32*)krow_cache + (u64)t_clamped * dw;
     const u32* q32 = (const u32*)q_sh;
     float acc = 0.f;
@@ -237,41 +241,26 @@ __global__ __launch_bounds__(TPB) void attn_decode_fused(
     }
     if (t == pos) acc = slast_sh;          // never trust the stale cache row
     if (t < T) s = acc * scale;
-  }
-
-  // ---- chunk softmax (identical to attn_decode_partial) ---------------
-  const int wave = threadIdx.x / LFMQ_WAVE;
-  const int lane = threadIdx.x % LFMQ_WAVE;
-  float wm = wave_reduce_max(s);
-  if (lane == 0) red[wave] = wm;
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    float m = red[0];
-#pragma unroll
-    for (int i = 1; i < NWAVE; ++i) m = fmaxf(m, red[i]);
-    m_sh = m;
+    // single-wave softmax over the 64 scores
+    const float m = wave_reduce_max(s);
+    const float p = (t < T) ? __expf(s - m) : 0.f;
+    p_sh[lane] = p;
+    const float l = wave_reduce_sum(p);
+    if (lane == 0) {
+      m_sh = m;
+      l_sh = l;
+    }
   }
   __syncthreads();
   const float m = m_sh;
-  const float p = (t < T) ? __expf(s - m) : 0.f;
-  p_sh[threadIdx.x] = p;
-  float wl = wave_reduce_sum(p);
-  if (lane == 0) red[wave] = wl;
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    float l = 0.f;
-#pragma unroll
-    for (int i = 0; i < NWAVE; ++i) l += red[i];
-    l_sh = l;
-  }
 
-  // ---- PV (v of t == pos comes from LDS) ------------------------------
+  // ---- PV: 4 waves x 16 tokens each (short dependent chains) ----------
   float o0 = 0.f, o1 = 0.f;
-  const int tbase = chunk * TPB + wave * LFMQ_WAVE;
-  const int nt = min(LFMQ_WAVE, T - tbase);
+  const int tbase = chunk * FCH + wave * (FCH / NWAVE);
+  const int nt = min(FCH / NWAVE, T - tbase);
   if (lane < dw) {
-#pragma unroll 8
-    for (int tt = 0; tt < LFMQ_WAVE; ++tt) {
+#pragma unroll
+    for (int tt = 0; tt < FCH / NWAVE; ++tt) {
       if (tt >= nt) break;
       const int tv = tbase + tt;
       const int tvc = tv < tmax ? tv : tmax - 1;
@@ -279,7 +268,7 @@ __global__ __launch_bounds__(TPB) void attn_decode_fused(
                                  : vrow_cache[(u64)tvc * dw + lane];
       float v0, v1;
       bf16x2_to_f32(vd, v0, v1);
-      const float pv = p_sh[wave * LFMQ_WAVE + tt];
+      const float pv = p_sh[wave * (FCH / NWAVE) + tt];
       o0 = fmaf(pv, v0, o0);
       o1 = fmaf(pv, v1, o1);
     }
@@ -318,10 +307,10 @@ __global__ __launch_bounds__(TPB) void attn_decode_fused(
 template <int D>
 __global__ __launch_bounds__(128) void attn_decode_combine(
     const float* __restrict__ scratch, u32* __restrict__ o, int nchunks,
-    int T, const i32* __restrict__ T_dev) {
+    int T, const i32* __restrict__ T_dev, int csz) {
   if (T_dev != nullptr) T = *T_dev;
   const int bh = blockIdx.x;
-  const int used = (T + TPB - 1) / TPB;
+  const int used = (T + csz - 1) / csz;
   const float* base = scratch + (u64)bh * nchunks * (D + 2);
 
   float M = -INFINITY;
@@ -366,7 +355,7 @@ hipError_t lfmq_attn_decode(const void* q, const void* kc, const void* vc,
     if (nchunks > 1)                                                          \
       hipLaunchKernelGGL((attn_decode_combine<DD>), gridB, blockB, 0, stream, \
                          (const float*)scratch, (u32*)o, nchunks, T,          \
-                         (const i32*)T_dev);                                  \
+                         (const i32*)T_dev, TPB);                             \
   } while (0)
   if (d == 128)
     LFMQ_ATTN(128);
@@ -386,7 +375,7 @@ hipError_t lfmq_attn_decode_fused(const void* qkv, void* kc, void* vc,
                                   int d, float scale, hipStream_t stream) {
   if (d != 128 && d != 64) return hipErrorInvalidValue;
   if (hq % hkv) return hipErrorInvalidValue;
-  const int nchunks = (T + TPB - 1) / TPB;
+  const int nchunks = (T + 63) / 64;  // FCH = 64
   dim3 gridA(b * hq, nchunks), blockA(TPB);
   dim3 gridB(b * hq), blockB(128);
 #define LFMQ_ATTNF(DD)                                                        \
@@ -398,7 +387,7 @@ hipError_t lfmq_attn_decode_fused(const void* qkv, void* kc, void* vc,
     if (nchunks > 1)                                                          \
       hipLaunchKernelGGL((attn_decode_combine<DD>), gridB, blockB, 0, stream, \
                          (const float*)scratch, (u32*)o, nchunks, T,          \
-                         (const i32*)T_dev);                                  \
+                         (const i32*)T_dev, 64);                              \
   } while (0)
   if (d == 128)
     LFMQ_ATTNF(128);

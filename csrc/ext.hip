@@ -362,7 +362,7 @@ torch::Tensor attn_decode_fused(torch::Tensor qkv, torch::Tensor kc,
   check_t(inv_freq, torch::kFloat32, "inv_freq");
   const int b = (int)kc.size(0), hkv = (int)kc.size(1);
   const int tmax = (int)kc.size(2), d = (int)kc.size(3);
-  const int nchunks = ((int)seq_len + 255) / 256;
+  const int nchunks = ((int)seq_len + 63) / 64;  // fused path: FCH = 64
   auto scratch = torch::empty({(long)b * hq * nchunks * (d + 2)},
                               qkv.options().dtype(torch::kFloat32));
   auto o = torch::empty({(long)b, 1L, hq * (long)d}, qkv.options());
