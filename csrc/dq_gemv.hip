@@ -529,12 +529,24 @@ hipError_t launch_bits(const void* qweight, const void* scales,
 // pre-folded into the quantized weights); resid != null fuses a residual
 // add into the epilogue (invalid under TP row-parallel world > 1 — the
 // caller adds after the all-reduce instead).
+hipError_t lfmq_dq_gemv_mfma(const void* qweight, const void* scales,
+                             const void* qzeros, const void* x, void* y,
+                             const void* bias, int m, int n, int k, int g,
+                             hipStream_t stream);
+
 hipError_t lfmq_dq_gemv(const void* qweight, const void* scales,
                         const void* qzeros, const void* x, void* y,
                         const void* bias, const void* resid, int m, int n,
                         int k, int g, int bits, float rms_eps, int cols_hint,
                         hipStream_t stream) {
   if (m > 16 || k % 8 != 0 || g % 32 != 0 || g > k) return hipErrorInvalidValue;
+  // MFMA-tiled batched variant (csrc/dq_gemv_mfma.hip): measured BELOW
+  // the lane-dot path at m = 8 (group-serial staging chain) — kept
+  // behind hint 6 for continued tuning, not auto-routed.
+  if (bits == 4 && m > 4 && rms_eps < 0.f && resid == nullptr && g <= 128 &&
+      k % g == 0 && k / g <= 128 && cols_hint == 6)
+    return lfmq_dq_gemv_mfma(qweight, scales, qzeros, x, y, bias, m, n, k, g,
+                             stream);
   const bool rms = rms_eps >= 0.f;
   if (rms && m > 4) return hipErrorInvalidValue;
   if (bits == 4) {

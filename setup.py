@@ -19,6 +19,7 @@ SOURCES = [
     "csrc/rope.hip",
     "csrc/elementwise.hip",
     "csrc/dq_gemv.hip",
+    "csrc/dq_gemv_mfma.hip",
     "csrc/dq_gemm.hip",
     "csrc/hessian.hip",
     "csrc/attn_decode.hip",
