@@ -49,10 +49,22 @@ class QLinear(nn.Module):
     def load_quantized(self, codes: torch.Tensor, scales: torch.Tensor,
                        zeros: torch.Tensor, bias: Optional[torch.Tensor] = None):
         """codes int32 [n,k], scales fp32/bf16 [n,G], zeros int32 [n,G].
-        Device-agnostic: pack runs on whatever device the codes live on."""
-        self.qweight.copy_(ref.pack_codes(codes, self.bits))
+        On GPU the hand-written HIP pack kernel (csrc/pack.hip) runs; on
+        CPU the exact reference twin."""
+        if codes.is_cuda:
+            from ..ops import hip_ext
+            ext = hip_ext()
+            self.qweight.copy_(ext.pack(codes.contiguous(), self.bits))
+            z = zeros.to(torch.int32).contiguous()
+            per = 32 // self.bits
+            pad = (-z.shape[1]) % per
+            if pad:
+                z = torch.cat([z, z.new_zeros(z.shape[0], pad)], dim=1)
+            self.qzeros.copy_(ext.pack(z, self.bits))
+        else:
+            self.qweight.copy_(ref.pack_codes(codes, self.bits))
+            self.qzeros.copy_(ref.pack_zeros(zeros.to(torch.int32), self.bits))
         self.scales.copy_(scales.to(self.scales.dtype))
-        self.qzeros.copy_(ref.pack_zeros(zeros.to(torch.int32), self.bits))
         if bias is not None:
             self.bias.copy_(bias.to(self.scales.dtype))
 
