@@ -61,6 +61,36 @@ def search_scale(
     return best_s
 
 
+@torch.no_grad()
+def search_clip(x: torch.Tensor, w: torch.Tensor, cfg: QuantConfig,
+                grid=(1.0, 0.9, 0.8, 0.7, 0.6)) -> torch.Tensor:
+    """Per-output-channel weight-clip search (AWQ's optional second
+    stage): shrink each row's quantization range by the ratio that
+    minimizes the activation-weighted reconstruction error.  Returns the
+    clip ratio per row [n]."""
+    xw = x.float().pow(2).mean(0)                    # per-in-channel energy
+    n = w.shape[0]
+    best_err = torch.full((n,), float("inf"))
+    best_r = torch.ones(n)
+    for r in grid:
+        wc = torch.clamp(w, min=w.amin(dim=1, keepdim=True) * r,
+                         max=w.amax(dim=1, keepdim=True) * r)
+        wq = _pseudo_quant(wc, cfg.bits, cfg.group_size, cfg.sym)
+        err = ((w - wq).pow(2) * xw).sum(dim=1)
+        better = err < best_err
+        best_err = torch.where(better, err, best_err)
+        best_r = torch.where(better, torch.full_like(best_r, r), best_r)
+    return best_r
+
+
+@torch.no_grad()
+def apply_clip(lin: nn.Linear, ratios: torch.Tensor) -> None:
+    w = lin.weight.data.float()
+    lo = w.amin(dim=1, keepdim=True) * ratios.unsqueeze(1)
+    hi = w.amax(dim=1, keepdim=True) * ratios.unsqueeze(1)
+    lin.weight.data = torch.clamp(w, min=lo, max=hi).to(lin.weight.dtype)
+
+
 def _scale_plan(model) -> List[dict]:
     """Per-block scale groups: {prev: (kind, name, slice), linears: [...]}"""
     arch = getattr(model.cfg, "arch", "llama")
@@ -165,6 +195,10 @@ class AWQQuantizer:
                 apply_scale(block, group["prev"], [linears[n] for n in names], s)
 
             for lname, lin in linears.items():
+                if cfg.awq_clip and samples.get(lname):
+                    xcat = torch.cat(samples[lname], dim=0).to(device)
+                    ratios = search_clip(xcat, lin.weight.float(), cfg)
+                    apply_clip(lin, ratios.to(lin.weight.device))
                 q = QLinear.from_rtn(lin, cfg).to(device)
                 replace_module(block, lname, q)
 

@@ -180,3 +180,38 @@ def test_gptq_resume(tmp_path):
         assert k1 == k2
         if "layers." in k1:
             assert torch.equal(v1, v2), k1
+
+
+def test_awq_clip_search_reduces_error():
+    from lfmq.quant.awq import search_clip, apply_clip
+    torch.manual_seed(2)
+    k, n = 128, 32
+    w = torch.randn(n, k)
+    w[:, 0] *= 30.0  # an outlier column stretches every row's range
+    x = torch.randn(500, k)
+    x[:, 0] *= 0.01  # ...but that channel barely matters to the output
+    cfg = QuantConfig(scheme="awq", bits=4, group_size=-1)
+    ratios = search_clip(x, w, cfg)
+    assert (ratios < 1.0).any()  # clipping must engage
+    xw = x.pow(2).mean(0)
+    err_clip = 0.0
+    lin = nn.Linear(k, n, bias=False)
+    lin.weight.data = w.clone()
+    apply_clip(lin, ratios)
+    from lfmq.quant.awq import _pseudo_quant
+    err_c = ((w - _pseudo_quant(lin.weight.float(), 4, -1, False)).pow(2) * xw).sum()
+    err_0 = ((w - _pseudo_quant(w, 4, -1, False)).pow(2) * xw).sum()
+    assert err_c < err_0
+
+
+def test_awq_clip_e2e():
+    import lfmq
+    from lfmq.models.llama import LlamaConfig
+    from lfmq.quant.calib import CalibStream
+    cfg = LlamaConfig.tiny()
+    qcfg = QuantConfig(scheme="awq", bits=4, group_size=64, awq_clip=True,
+                       awq_grid=6, calib_n_seqs=2, calib_seq_len=32)
+    calib = CalibStream(cfg.vocab_size, 32, 2, qcfg.calib_seed)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=15), qcfg, calib=calib)
+    ids = torch.randint(0, cfg.vocab_size, (1, 8))
+    assert torch.isfinite(qm(ids, qm.make_cache(1, 16)).float()).all()
