@@ -31,6 +31,8 @@ def main():
     ap.add_argument("--group-size", type=int, default=128)
     ap.add_argument("--layers", type=int, default=0,
                     help="override layer count (debug only; 0 = full model)")
+    ap.add_argument("--quant-lm-head", action="store_true",
+                    help="quantize the lm_head too (default: bf16 head)")
     args = ap.parse_args()
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -51,7 +53,7 @@ def main():
     model, cfg, qcfg = build_quantized_llama(
         args.model, "rtn", args.bits, args.group_size, device,
         tp_rank=rank, tp_world=world, seed=0,
-        n_layers=args.layers or None)
+        n_layers=args.layers or None, quant_lm_head=args.quant_lm_head)
 
     r = bench_decode(model, cfg, args.batch, args.prompt, args.steps,
                      args.warmup, device)
@@ -72,7 +74,8 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": args.model,
-                "quant": f"w{args.bits}a16-g{args.group_size}-rtn",
+                "quant": f"w{args.bits}a16-g{args.group_size}-rtn"
+                         + ("-qhead" if args.quant_lm_head else ""),
                 "global_batch": args.batch,
                 "seq_len": args.prompt,
                 "gen_steps": r["steps"],

@@ -48,9 +48,11 @@ __global__ __launch_bounds__(TPB, 1) void dq_gemv_mfma_kernel(
   // W tiles [16 cols][20-dword padded rows] + xgs [16][G<=128] fp32 +
   // cross-wave accumulator tiles.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* xtile = smem + wave * 16 * 128 * 2;            // 4 tiles x 4 KiB
-  u32* wtile = (u32*)(smem + 4 * 16 * 128 * 2) + wave * 16 * 20;
-  float* xgs = (float*)(smem + 4 * 16 * 128 * 2 + 4 * 16 * 20 * 4);
+  // banks: [2][NW] x-tiles (4 KiB each), then [2][NW] padded W tiles,
+  // then xgs [16][G], then the combine tiles
+  char* xtile = smem + wave * 16 * 128 * 2;            // bank 0 slot
+  u32* wtile = (u32*)(smem + 2 * 4 * 16 * 128 * 2) + wave * 16 * 20;
+  float* xgs = (float*)(smem + 2 * 4 * 16 * 128 * 2 + 2 * 4 * 16 * 20 * 4);
   float* comb = xgs + 16 * 128;                        // [NW][16][16]
 
   auto x_rsrc = make_rsrc(x, (u32)(m * k * 2));
@@ -88,6 +90,62 @@ __global__ __launch_bounds__(TPB, 1) void dq_gemv_mfma_kernel(
   const int g_end = min(G, g_begin + per_wave);
 
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+
+  if (gdw == 16) {
+    // g == 128 fast path: 2-deep wave-local pipeline — group gg+1's
+    // global loads ISSUE before gg's MFMAs and their LDS writes land
+    // after (T14 issue-early/write-late; no barriers: all LDS traffic
+    // is wave-local, double-buffered in the wave's two tile slots).
+    char* xt[2] = {xtile, xtile + NW * 16 * 128 * 2};   // bank 1 slot
+    u32* wt[2] = {wtile, wtile + NW * 16 * 20};
+    u32x4 xr[4], wr;
+    auto issue = [&](int gg) {
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const int task = lane + 64 * t;
+        const int row = task / 16, slot = task % 16;
+        xr[t] = __builtin_amdgcn_raw_buffer_load_b128(
+            x_rsrc, (u32)((row * k + gg * g) * 2 + slot * 16), 0, 0);
+      }
+      wr = *(const u32x4*)(wst_row + gg * 16 + (lane % 4) * 4);
+    };
+    auto write = [&](int buf) {
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const int task = lane + 64 * t;
+        const int row = task / 16, slot = task % 16;
+        *(u32x4*)(xt[buf] + aswz((u32)(row * 256 + slot * 16), row)) = xr[t];
+      }
+      *(u32x4*)(wt[buf] + (lane / 4) * 20 + (lane % 4) * 4) = wr;
+    };
+    issue(g_begin);
+    write(0);
+    for (int gg = g_begin; gg < g_end; ++gg) {
+      const int buf = (gg - g_begin) & 1;
+      if (gg + 1 < g_end) issue(gg + 1);   // loads in flight across MFMAs
+      f32x4 accg = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        const u32 wd = wt[buf][lc * 20 + ks * 4 + (lane >> 4)];
+        const DequantPairs4 dq = dequant4_pairs(wd);
+        frag_ab b = __builtin_bit_cast(
+            frag_ab, u32x4{dq.p01, dq.p23, dq.p45, dq.p67});
+        const int krow = lane & 15;
+        const int koff = ks * 32 + (lane >> 4) * 8;
+        frag_ab a = *(frag_ab*)(xt[buf] + aswz((u32)(krow * 256 + koff * 2), krow));
+        accg = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, accg, 0, 0, 0);
+      }
+      if (gg + 1 < g_end) write(buf ^ 1);  // write-late into the other slot
+      const float s = bf16_bits_to_f32(srow[gg]);
+      const float zb = s * (float)(unpack4(zrow[gg / 8], gg % 8) + 128u);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = (lane >> 4) * 4 + r;
+        acc[r] = fmaf(s, accg[r], acc[r]);
+        acc[r] = fmaf(-zb, xgs[row * G + gg], acc[r]);
+      }
+    }
+  } else
   for (int gg = g_begin; gg < g_end; ++gg) {
     // stage this wave's x k-tile [16][g]: 16 rows x (g/8) 16-B slots
     {
@@ -169,8 +227,8 @@ hipError_t lfmq_dq_gemv_mfma(const void* qweight, const void* scales,
   if (m > 16 || g % 32 != 0 || g > 128 || k % g != 0 || k / g > 128)
     return hipErrorInvalidValue;
   dim3 grid(ceil_div_u32(n, 16)), block(TPB);
-  const size_t lds =
-      4 * 16 * 128 * 2 + 4 * 16 * 20 * 4 + 16 * 128 * 4 + NW * 16 * 16 * 4;
+  const size_t lds = 2 * (4 * 16 * 128 * 2 + 4 * 16 * 20 * 4) +
+                     16 * 128 * 4 + NW * 16 * 16 * 4;
   hipLaunchKernelGGL(dq_gemv_mfma_kernel, grid, block, lds, stream,
                      (const u32*)qweight, (const u32*)scales,
                      (const u32*)qzeros, (const u32*)x, (u32*)y,

@@ -31,7 +31,8 @@ def _fast_random_init(model, seed: int):
 @torch.no_grad()
 def build_quantized_llama(model_name: str, scheme: str, bits: int, g: int,
                           device, tp_rank: int = 0, tp_world: int = 1,
-                          seed: int = 0, n_layers: Optional[int] = None):
+                          seed: int = 0, n_layers: Optional[int] = None,
+                          quant_lm_head: bool = False):
     """Random-init on device, quantize (RTN weights for bench speed: the
     decode kernel path and memory traffic are scheme-independent), shard."""
     if model_name == "llama3-8b":
@@ -48,7 +49,8 @@ def build_quantized_llama(model_name: str, scheme: str, bits: int, g: int,
     model = build_model(cfg)
     model = model.to_empty(device=str(device)) if str(device) != "cpu" else model
     _fast_random_init(model, seed)
-    qcfg = QuantConfig(scheme=scheme, bits=bits, group_size=g)
+    qcfg = QuantConfig(scheme=scheme, bits=bits, group_size=g,
+                       skip_modules=() if quant_lm_head else ("lm_head",))
     model = quantize(model, qcfg)
     if tp_world > 1:
         if cfg.arch == "mixtral":

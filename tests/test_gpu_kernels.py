@@ -379,3 +379,20 @@ def test_attn_decode_fused_multichunk():
     o = lfmq_hip.attn_decode_fused(qkv.to(DEV), kc.to(DEV), vc.to(DEV), T,
                                    1.0 / _m.sqrt(d), inv, hq, None)
     assert rel_err(o.reshape(b, hq, 1, d), o_ref) < 2e-2
+
+
+@pytest.mark.parametrize("m", [5, 8, 16])
+@pytest.mark.parametrize("n,k,g", [(256, 2048, 128), (100, 512, 64),
+                                   (48, 4096, 128)])
+def test_dq_gemv_mfma_variant(m, n, k, g):
+    """The MFMA-tiled batched GEMV (cols_hint=6) vs reference."""
+    torch.manual_seed(m * 10 + n)
+    w = torch.randn(n, k)
+    codes, scales, zeros = ref.quant_groupwise(w, 4, g, False)
+    qw, qz = ref.pack_codes(codes, 4), ref.pack_zeros(zeros, 4)
+    x = torch.randn(m, k).to(torch.bfloat16)
+    y_ref = ref.dq_linear(x.float(), qw, scales.to(torch.bfloat16).float(),
+                          qz, 4, g)
+    y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
+                        qz.to(DEV), 4, g, None, None, -1.0, 6)
+    assert rel_err(y, y_ref) < 2e-2, f"rel={rel_err(y, y_ref)}"
