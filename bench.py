@@ -80,6 +80,7 @@ def main():
                 "seq_len": args.prompt,
                 "gen_steps": r["steps"],
                 "parallelism": f"tp{world}",
+                "hip_graph": bool(r.get("hip_graph")),
             },
         }
         print(json.dumps(out))

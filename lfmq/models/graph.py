@@ -72,8 +72,10 @@ class GraphDecoder:
 
     @torch.no_grad()
     def decode(self, n_tokens: int) -> torch.Tensor:
-        """Replay the captured step n_tokens-1 more times; returns
-        [batch, n_tokens] including the first post-prefill token."""
+        """Replay the captured step n_tokens times; returns
+        [batch, n_tokens] starting with the first post-prefill token
+        (each replay records the token it consumed and feeds the newly
+        sampled one back into the graph's input buffer)."""
         if self.graph is None:
             # capture AFTER a real prefill so shapes/addresses are final
             saved = (self.state.clone(), self.tok.clone(),
