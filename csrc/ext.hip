@@ -314,7 +314,7 @@ torch::Tensor grouped_dq_gateup(torch::Tensor x, torch::Tensor src_tok,
   check_t(qweight, torch::kInt32, "qweight");
   check_t(src_tok, torch::kInt32, "src_tok");
   check_t(offsets, torch::kInt32, "offsets");
-  const int E = (int)qweight.size(0), n = (int)qweight.size(1) == 0 ? 0 : (int)qweight.size(1);
+  const int E = (int)qweight.size(0), n = (int)qweight.size(1);
   const int k = (int)x.size(1);
   const long slots = src_tok.numel();
   auto y = torch::empty({slots, (long)(n / 2)}, x.options());
