@@ -74,3 +74,24 @@ def summarize_db(db_path: str, top: int = 20, tail_frac: float = 0.5) -> list:
          f"ORDER BY 3 DESC LIMIT {top}")
     return [{"kernel": n, "calls": c, "total_ms": ms, "us_per_call": us}
             for n, c, ms, us in cur.execute(q)]
+
+
+def torch_profile(fn, steps: int = 8, out_trace: str = None):
+    """torch.profiler wrapper for launch-gap analysis:
+    runs fn() `steps` times under the ROCm-aware profiler and returns the
+    key-averages table; optionally exports a chrome trace."""
+    import torch
+    from torch.profiler import ProfilerActivity, profile
+    acts = [ProfilerActivity.CPU]
+    if torch.cuda.is_available():
+        acts.append(ProfilerActivity.CUDA)
+    with profile(activities=acts) as prof:
+        for _ in range(steps):
+            fn()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+    if out_trace:
+        prof.export_chrome_trace(out_trace)
+    return prof.key_averages().table(sort_by="self_cuda_time_total"
+                                     if torch.cuda.is_available()
+                                     else "self_cpu_time_total", row_limit=25)

@@ -131,6 +131,37 @@ def shard_llama(model: nn.Module, rank: int, world: int) -> nn.Module:
     return model
 
 
+@torch.no_grad()
+def shard_mixtral(model: nn.Module, rank: int, world: int) -> nn.Module:
+    """TP-only Mixtral fallback: attention column/row
+    sharded exactly like Llama, experts REPLICATED (the mandated
+    expert-parallel path is lfmq.dist.ep; this keeps memory higher but
+    needs no all-to-all).  The row-parallel o_proj all-reduce makes the
+    hidden stream identical on every rank, so the replicated MoE stays
+    consistent."""
+    if world == 1:
+        return model
+    cfg = model.cfg
+    d, hq, hkv = cfg.head_dim, cfg.n_heads, cfg.n_kv_heads
+    assert hq % world == 0 and hkv % world == 0
+    for layer in model.layers:
+        attn = layer.self_attn
+        qkv_parts = [(0, hq * d), (hq * d, hkv * d), ((hq + hkv) * d, hkv * d)]
+        replace_module(attn, "qkv_proj",
+                       _shard_col(attn.qkv_proj, qkv_parts, rank, world))
+        replace_module(attn, "o_proj", _shard_row(attn.o_proj, rank, world))
+        attn.n_heads = hq // world
+        attn.n_kv_heads = hkv // world
+    lm = model.lm_head
+    if not isinstance(lm, QLinear):
+        assert lm.out_features % world == 0
+        vs = lm.out_features // world
+        new = nn.Linear(lm.in_features, vs, bias=False, dtype=lm.weight.dtype)
+        new.weight.data.copy_(lm.weight.data[rank * vs:(rank + 1) * vs])
+        replace_module(model, "lm_head", _GatheredLinear(new))
+    return model
+
+
 class _GatheredLinear(nn.Module):
     """Dense column-parallel linear + all-gather (the bf16 lm_head)."""
 

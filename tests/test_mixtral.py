@@ -75,3 +75,30 @@ def _ep_worker(rank, world, port, tmpdir):
 
 def test_ep_gloo_matches_dense(tmp_path):
     mp.spawn(_ep_worker, args=(2, 29712, str(tmp_path)), nprocs=2, join=True)
+
+
+def _tp_worker(rank, world, port, tmpdir):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        cfg = MixtralConfig.tiny()
+        qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+        qm = lfmq.quantize(lfmq.build_model(cfg, seed=44), qcfg)
+        ids = torch.arange(10).remainder(cfg.vocab_size).reshape(1, 10)
+        ref_logits = qm(ids, qm.make_cache(1, 32))
+        from lfmq.dist.tp import shard_mixtral
+        shard_mixtral(qm, rank, world)
+        tp_logits = qm(ids, qm.make_cache(1, 32))
+        rel = (tp_logits.float() - ref_logits.float()).abs().max() / \
+            ref_logits.float().abs().max()
+        assert rel < 0.03, f"mixtral TP rel={float(rel)}"
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_mixtral_tp_fallback_gloo(tmp_path):
+    """TP-only Mixtral (attention sharded, experts replicated)."""
+    mp.spawn(_tp_worker, args=(2, 29733, str(tmp_path)), nprocs=2, join=True)
