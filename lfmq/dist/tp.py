@@ -184,6 +184,11 @@ def load_tp(path: str, meta: dict, device: str, rank: int, world: int,
 
     model = _build_quantized_skeleton(meta, dtype=dtype)
     cfg = model.cfg
+    if getattr(cfg, "arch", "llama") != "llama":
+        raise NotImplementedError(
+            f"TP load for arch {cfg.arch!r} is not implemented: use "
+            "lfmq.load(..., tp_world=1) + lfmq.dist.ep.apply_expert_parallel "
+            "(Mixtral) or lfmq.dist.tp.shard_mixtral on the loaded model")
     d, hq, hkv = cfg.head_dim, cfg.n_heads, cfg.n_kv_heads
     inter, vocab = cfg.intermediate_size, cfg.vocab_size
 

@@ -102,3 +102,13 @@ def _tp_worker(rank, world, port, tmpdir):
 def test_mixtral_tp_fallback_gloo(tmp_path):
     """TP-only Mixtral (attention sharded, experts replicated)."""
     mp.spawn(_tp_worker, args=(2, 29733, str(tmp_path)), nprocs=2, join=True)
+
+
+def test_mixtral_tp_load_raises_cleanly(tmp_path):
+    cfg = MixtralConfig.tiny()
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=9), qcfg)
+    p = str(tmp_path / "mx")
+    lfmq.save(qm, p, qcfg)
+    with pytest.raises(NotImplementedError):
+        lfmq.load(p, tp_rank=0, tp_world=2)
