@@ -41,6 +41,9 @@ def _cat_rows(t: torch.Tensor, slices: List[slice]) -> torch.Tensor:
 
 def _shard_col(q: QLinear, parts: List[Tuple[int, int]], rank: int, world: int,
                gather_output: bool = False) -> ColumnParallelQLinear:
+    assert not hasattr(q, "g_idx"), (
+        "act-order (g_idx) layers cannot be TP-sharded in this round — "
+        "re-quantize with act_order=False for TP")
     sl = _row_slices(parts, rank, world)
     out = ColumnParallelQLinear(q.in_features, q.out_features // world, q.bits,
                                 q.group_size, bias=q.bias is not None,
@@ -54,6 +57,9 @@ def _shard_col(q: QLinear, parts: List[Tuple[int, int]], rank: int, world: int,
 
 
 def _shard_row(q: QLinear, rank: int, world: int) -> RowParallelQLinear:
+    assert not hasattr(q, "g_idx"), (
+        "act-order (g_idx) layers cannot be TP-sharded in this round — "
+        "re-quantize with act_order=False for TP")
     k = q.in_features
     assert k % world == 0, f"in_features {k} % tp {world} != 0"
     ks = k // world

@@ -105,3 +105,31 @@ def test_tp_gloo_world4(tmp_path, world):
     rel = (out["logits"].float() - ref_logits.float()).abs().max() / \
         ref_logits.float().abs().max()
     assert rel < 0.05, f"TP4 logits rel err {float(rel)}"
+
+
+def test_tp_gloo_quantized_lm_head(tmp_path):
+    """TP with a QUANTIZED lm_head (skip_modules=()): column-shard +
+    all-gather over the packed head."""
+    cfg = _tiny_cfg()
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32,
+                            skip_modules=())
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=51), qcfg)
+    path = str(tmp_path / "ckpt_qh")
+    lfmq.save(qm, path, qcfg)
+    ids = torch.arange(12).remainder(256).reshape(1, 12)
+    ref_logits = qm(ids, qm.make_cache(1, 32))
+    mp.spawn(_worker, args=(2, path, 29741, str(tmp_path)), nprocs=2,
+             join=True)
+    out = torch.load(str(tmp_path / "tp_out.pt"), weights_only=False)
+    rel = (out["logits"].float() - ref_logits.float()).abs().max() / \
+        ref_logits.float().abs().max()
+    assert rel < 0.05, f"quantized-head TP rel err {float(rel)}"
+
+
+def test_act_order_tp_shard_guard(tmp_path):
+    from lfmq.dist.tp import _shard_col
+    from lfmq.models.qlinear import QLinear
+    q = QLinear(64, 32, 4, 32)
+    q.register_buffer("g_idx", torch.zeros(64, dtype=torch.int32))
+    with pytest.raises(AssertionError):
+        _shard_col(q, [(0, 32)], 0, 2)
