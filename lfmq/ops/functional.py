@@ -110,20 +110,37 @@ def attn_decode(q, k_cache, v_cache, seq_len: int, scale: Optional[float] = None
 
 def attn_prefill(q, k, v, causal: bool = True, scale: Optional[float] = None):
     """Prefill attention. GPU: explicit bf16 matmul chain via rocBLAS (the
-    PyTorch-ROCm layer; never SDPA, which can route to aotriton/Triton)."""
+    PyTorch-ROCm layer; never SDPA, which can route to aotriton/Triton).
+    GQA is handled by folding the q-head repeat into the row dimension
+    (no K/V repeat_interleave copies), and the query axis is processed in
+    chunks so the fp32 score tile stays bounded (a full 2048-seq layer
+    would otherwise materialize a 537 MB score tensor)."""
     if q.is_cuda:
         b, hq, s, d = q.shape
         hkv = k.shape[1]
         rep = hq // hkv
         scale = scale or 1.0 / math.sqrt(d)
-        kf = k.repeat_interleave(rep, dim=1) if rep > 1 else k
-        vf = v.repeat_interleave(rep, dim=1) if rep > 1 else v
-        sc = torch.matmul(q, kf.transpose(-1, -2)).float() * scale
-        if causal:
-            mask = torch.full((s, s), float("-inf"), device=q.device).triu(1)
-            sc = sc + mask
-        p = torch.softmax(sc, dim=-1).to(q.dtype)
-        return torch.matmul(p, vf)
+        kt = k.transpose(-1, -2)                      # [b, hkv, d, s]
+        out = torch.empty_like(q)
+        CH = 512 if s > 512 else s
+        for q0 in range(0, s, CH):
+            q1 = min(s, q0 + CH)
+            # [b, hkv, rep*(q1-q0), d]: repeat folded into the row dim
+            qc = (q.view(b, hkv, rep, s, d)[:, :, :, q0:q1]
+                  .reshape(b, hkv, rep * (q1 - q0), d))
+            sc = torch.matmul(qc, kt).float() * scale  # [b,hkv,rep*cq,s]
+            if causal:
+                t_idx = torch.arange(s, device=q.device)
+                rows = torch.arange(q0, q1, device=q.device)
+                m = (t_idx[None, :] > rows[:, None])   # [cq, s]
+                sc = sc.view(b, hkv, rep, q1 - q0, s).masked_fill(
+                    m[None, None, None], float("-inf")).view(
+                    b, hkv, rep * (q1 - q0), s)
+            p = torch.softmax(sc, dim=-1).to(q.dtype)
+            oc = torch.matmul(p, v)                    # [b,hkv,rep*cq,d]
+            out.view(b, hkv, rep, s, d)[:, :, :, q0:q1] = (
+                oc.view(b, hkv, rep, q1 - q0, d))
+        return out
     return ref.attn_prefill(q, k, v, causal, scale)
 
 
