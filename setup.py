@@ -12,6 +12,13 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
+# LFMQ_ASAN=1: address-sanitized build of the HOST side of the extension
+# (binding/launch logic).  Run with
+# LD_PRELOAD=$(clang -print-file-name=libclang_rt.asan-x86_64.so).
+_ASAN = os.environ.get("LFMQ_ASAN", "0") == "1"
+_CXX_FLAGS = ["-O3", "-std=c++17"] + (
+    ["-fsanitize=address", "-fno-omit-frame-pointer"] if _ASAN else [])
+
 SOURCES = [
     "csrc/ext.hip",
     "csrc/pack.hip",
@@ -36,7 +43,7 @@ setup(
             name="lfmq_hip",
             sources=SOURCES,
             extra_compile_args={
-                "cxx": ["-O3", "-std=c++17"],
+                "cxx": _CXX_FLAGS,
                 "nvcc": ["-O3", "-std=c++17"],
             },
         )
