@@ -29,6 +29,9 @@ hipError_t lfmq_attn_decode(const void*, const void*, const void*, void*,
 hipError_t lfmq_attn_decode_fused(const void*, void*, void*, void*, void*,
                                   const void*, int, int, int, int, int,
                                   const void*, int, float, hipStream_t);
+hipError_t lfmq_attn_prefill(const void*, const void*, const void*, void*,
+                             int, int, int, int, int, int, int, float,
+                             hipStream_t);
 hipError_t lfmq_hessian_accum(const void*, void*, int, int, hipStream_t);
 hipError_t lfmq_dq_gemv_gateup(const void*, const void*, const void*,
                                const void*, void*, int, int, int, int, int,
@@ -245,6 +248,26 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   return o;
 }
 
+// EXPERIMENTAL flash prefill (csrc/attn_prefill.hip): q [b,s,hq*d] roped,
+// caches [b,hkv,tmax,d] holding rows [0, pos0+s).  d == 128 only.
+torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor kc,
+                           torch::Tensor vc, long hq, long pos0,
+                           double scale) {
+  check_t(q, torch::kBFloat16, "q");
+  check_t(kc, torch::kBFloat16, "k_cache");
+  check_t(vc, torch::kBFloat16, "v_cache");
+  const int b = (int)q.size(0), s = (int)q.size(1);
+  const int hkv = (int)kc.size(1), tmax = (int)kc.size(2);
+  const int d = (int)kc.size(3);
+  TORCH_CHECK(q.size(2) == hq * d, "q last dim != hq*d");
+  TORCH_CHECK(pos0 + s <= tmax, "cache too small");
+  auto o = torch::empty_like(q);
+  CHECK_HIP(lfmq_attn_prefill(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                              o.data_ptr(), b, s, (int)hq, hkv, tmax,
+                              (int)pos0, d, (float)scale, stream()));
+  return o;
+}
+
 torch::Tensor dq_gemv_gateup(torch::Tensor x, torch::Tensor qweight,
                              torch::Tensor scales, torch::Tensor qzeros,
                              long bits, long g, double rms_eps,
@@ -398,6 +421,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("residual_add", &residual_add, "bf16 add");
   m.def("kv_append", &kv_append, "KV cache append");
   m.def("attn_decode", &attn_decode, "GQA decode attention");
+  m.def("attn_prefill", &attn_prefill,
+        "EXPERIMENTAL flash prefill attention (LFMQ_FLASH_PREFILL)");
   m.def("hessian_accum", &hessian_accum, "H += 2 X^T X");
   m.def("dq_gemv_gateup", &dq_gemv_gateup, "fused gate_up GEMV + silu*mul");
   m.def("qkv_rope_append", &qkv_rope_append, "fused rope + KV append");

@@ -106,6 +106,10 @@ class LlamaAttention(nn.Module):
                 o = F.attn_decode(q.view(b, hq, 1, d), cache_k, cache_v,
                                   seq_len_after, self.scale)
                 o = o.view(b, s, hq * d)
+            elif F.USE_FLASH_PREFILL and d == 128:
+                # experimental one-kernel MFMA prefill (LFMQ_FLASH_PREFILL=1)
+                o = F.attn_prefill_fused(q, cache_k, cache_v, pos0,
+                                         self.scale).reshape(b, s, hq * d)
             else:
                 qh = q.transpose(1, 2)
                 kc = cache_k[:, :, pos0:seq_len_after]

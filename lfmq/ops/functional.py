@@ -144,6 +144,25 @@ def attn_prefill(q, k, v, causal: bool = True, scale: Optional[float] = None):
     return ref.attn_prefill(q, k, v, causal, scale)
 
 
+def attn_prefill_fused(q: torch.Tensor, k_cache: torch.Tensor,
+                       v_cache: torch.Tensor, pos0: int,
+                       scale: Optional[float] = None) -> torch.Tensor:
+    """EXPERIMENTAL GPU flash prefill (csrc/attn_prefill.hip): one MFMA
+    kernel instead of the rocBLAS matmul chain.  q [b, s, hq, d] already
+    roped (the qkv_rope_append output); caches [b, hkv, Tmax, d] hold
+    rows [0, pos0+s).  d == 128 only.  The model uses it only under
+    LFMQ_FLASH_PREFILL=1 — not yet validated on hardware."""
+    b, s, hq, d = q.shape
+    scale = scale or 1.0 / math.sqrt(d)
+    o = _guard(hip_ext().attn_prefill(q.reshape(b, s, hq * d), k_cache,
+                                      v_cache, hq, pos0, scale),
+               "attn_prefill_fused")
+    return o.view(b, s, hq, d)
+
+
+USE_FLASH_PREFILL = os.environ.get("LFMQ_FLASH_PREFILL", "0") == "1"
+
+
 def qkv_rope_append(qkv: torch.Tensor, k_cache: torch.Tensor,
                     v_cache: torch.Tensor, positions: torch.Tensor,
                     pos0: int, hq: int, theta: float = 500000.0) -> torch.Tensor:

@@ -30,6 +30,7 @@ SOURCES = [
     "csrc/dq_gemm.hip",
     "csrc/hessian.hip",
     "csrc/attn_decode.hip",
+    "csrc/attn_prefill.hip",
     "csrc/qkv_rope_append.hip",
     "csrc/sample.hip",
     "csrc/probe.hip",

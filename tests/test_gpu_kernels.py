@@ -396,3 +396,49 @@ def test_dq_gemv_mfma_variant(m, n, k, g):
     y = ext().dq_linear(x.to(DEV), qw.to(DEV), scales.to(torch.bfloat16).to(DEV),
                         qz.to(DEV), 4, g, None, None, -1.0, 6)
     assert rel_err(y, y_ref) < 2e-2, f"rel={rel_err(y, y_ref)}"
+
+
+# --------------------------------------- experimental flash prefill
+# Opt-in (LFMQ_EXPERIMENTAL=1): csrc/attn_prefill.hip is round-2 work,
+# not yet hardware-validated — kept out of the default gpu suite.
+import os as _os  # noqa: E402
+
+EXPERIMENTAL = _os.environ.get("LFMQ_EXPERIMENTAL", "0") == "1"
+
+
+def _prefill_oracle(q, kc, vc, pos0, T, scale):
+    """fp32 attention of queries at positions [pos0, T) against cache
+    rows [0, T) with causal mask.  q [b,s,hq,d]; caches [b,hkv,Tmax,d]."""
+    b, s, hq, d = q.shape
+    hkv = kc.shape[1]
+    rep = hq // hkv
+    kf = kc[:, :, :T].float().repeat_interleave(rep, dim=1)
+    vf = vc[:, :, :T].float().repeat_interleave(rep, dim=1)
+    qh = q.float().permute(0, 2, 1, 3)                       # [b,hq,s,d]
+    sc = torch.einsum("bhqd,bhtd->bhqt", qh, kf) * scale
+    t_idx = torch.arange(T)
+    rows = torch.arange(pos0, pos0 + s)
+    sc = sc.masked_fill((t_idx[None, :] > rows[:, None])[None, None],
+                        float("-inf"))
+    o = torch.einsum("bhqt,bhtd->bhqd", torch.softmax(sc, dim=-1), vf)
+    return o.permute(0, 2, 1, 3)                             # [b,s,hq,d]
+
+
+@pytest.mark.skipif(not EXPERIMENTAL, reason="opt-in: LFMQ_EXPERIMENTAL=1")
+@pytest.mark.parametrize("b,s,pos0", [(1, 128, 0), (1, 96, 0), (2, 64, 0),
+                                      (1, 64, 100), (1, 1, 37), (1, 300, 13)])
+def test_attn_prefill_flash(b, s, pos0):
+    torch.manual_seed(s * 7 + pos0)
+    hq, hkv, d = 8, 2, 128
+    T = pos0 + s
+    tmax = T + 5
+    scale = 1.0 / math.sqrt(d)
+    q = torch.randn(b, s, hq, d, dtype=torch.bfloat16) * 0.5
+    kc = torch.zeros(b, hkv, tmax, d, dtype=torch.bfloat16)
+    vc = torch.zeros(b, hkv, tmax, d, dtype=torch.bfloat16)
+    kc[:, :, :T] = torch.randn(b, hkv, T, d, dtype=torch.bfloat16) * 0.5
+    vc[:, :, :T] = torch.randn(b, hkv, T, d, dtype=torch.bfloat16)
+    o_ref = _prefill_oracle(q, kc, vc, pos0, T, scale)
+    from lfmq.ops import functional as F
+    o = F.attn_prefill_fused(q.to(DEV), kc.to(DEV), vc.to(DEV), pos0, scale)
+    assert rel_err(o, o_ref) < 3e-2
