@@ -107,7 +107,8 @@ class LlamaAttention(nn.Module):
                                   seq_len_after, self.scale)
                 o = o.view(b, s, hq * d)
             elif F.USE_FLASH_PREFILL and d == 128:
-                # experimental one-kernel MFMA prefill (LFMQ_FLASH_PREFILL=1)
+                # one-kernel MFMA flash prefill; handles pos0 > 0 too
+                # (LFMQ_FLASH_PREFILL=0 restores the matmul-chain path)
                 o = F.attn_prefill_fused(q, cache_k, cache_v, pos0,
                                          self.scale).reshape(b, s, hq * d)
             else:

@@ -398,14 +398,7 @@ def test_dq_gemv_mfma_variant(m, n, k, g):
     assert rel_err(y, y_ref) < 2e-2, f"rel={rel_err(y, y_ref)}"
 
 
-# --------------------------------------- experimental flash prefill
-# Opt-in (LFMQ_EXPERIMENTAL=1): csrc/attn_prefill.hip is round-2 work,
-# not yet hardware-validated — kept out of the default gpu suite.
-import os as _os  # noqa: E402
-
-EXPERIMENTAL = _os.environ.get("LFMQ_EXPERIMENTAL", "0") == "1"
-
-
+# ------------------------------------------------ flash prefill
 def _prefill_oracle(q, kc, vc, pos0, T, scale):
     """fp32 attention of queries at positions [pos0, T) against cache
     rows [0, T) with causal mask.  q [b,s,hq,d]; caches [b,hkv,Tmax,d]."""
@@ -424,7 +417,6 @@ def _prefill_oracle(q, kc, vc, pos0, T, scale):
     return o.permute(0, 2, 1, 3)                             # [b,s,hq,d]
 
 
-@pytest.mark.skipif(not EXPERIMENTAL, reason="opt-in: LFMQ_EXPERIMENTAL=1")
 @pytest.mark.parametrize("b,s,pos0", [(1, 128, 0), (1, 96, 0), (2, 64, 0),
                                       (1, 64, 100), (1, 1, 37), (1, 300, 13)])
 def test_attn_prefill_flash(b, s, pos0):
