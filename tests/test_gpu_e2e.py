@@ -213,3 +213,28 @@ def test_cli_quantize_eval_gpu(tmp_path):
     assert r.returncode == 0, r.stderr[-1500:]
     out = json.loads(r.stdout[r.stdout.index("{"):])
     assert "delta_ppl" in out and "logit_kl" in out
+
+
+def test_decode_matches_prefill_gpu_d128():
+    """Same consistency check at head_dim 128 — the flash-prefill
+    (csrc/attn_prefill.hip) default path vs the fused decode kernels."""
+    cfg = LlamaConfig(vocab_size=1024, hidden_size=512, intermediate_size=1024,
+                      n_layers=2, n_heads=4, n_kv_heads=2, head_dim=128,
+                      max_seq=256)
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=64)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=9), qcfg).to(DEV)
+    ids = torch.randint(0, cfg.vocab_size, (1, 72), device=DEV)
+    lp = qm(ids, qm.make_cache(1, 128)).float()          # flash prefill
+    cache = qm.make_cache(1, 128)
+    outs = [qm(ids[:, t:t + 1], cache).float() for t in range(72)]
+    ld = torch.cat(outs, dim=1)                          # decode kernels
+    rel = (lp - ld).abs().max() / lp.abs().max()
+    assert rel < 0.05, f"decode/flash-prefill mismatch rel={float(rel)}"
+    # chunked continuation (pos0 > 0) through the same kernel
+    more = torch.randint(0, cfg.vocab_size, (1, 40), device=DEV)
+    cont = qm(more, cache).float()
+    cache2 = qm.make_cache(1, 128)
+    qm(torch.cat([ids, more], dim=1)[:, :72], cache2)
+    outs2 = [qm(more[:, t:t + 1], cache2).float() for t in range(40)]
+    rel2 = (cont - torch.cat(outs2, dim=1)).abs().max() / cont.abs().max()
+    assert rel2 < 0.05, f"continuation mismatch rel={float(rel2)}"
