@@ -133,3 +133,46 @@ def test_act_order_tp_shard_guard(tmp_path):
     q.register_buffer("g_idx", torch.zeros(64, dtype=torch.int32))
     with pytest.raises(AssertionError):
         _shard_col(q, [(0, 32)], 0, 2)
+
+
+def _worker_bench_path(rank, world, port, out_dir):
+    """Mirror of the bench.py N-GPU route: build+quantize in every rank
+    (deterministic), shard_llama in memory, forward under gloo."""
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from lfmq.dist.tp import shard_llama
+        cfg = LlamaConfig(vocab_size=256, hidden_size=256,
+                          intermediate_size=512, n_layers=2, n_heads=8,
+                          n_kv_heads=8, head_dim=32, max_seq=64)
+        qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+        qm = lfmq.quantize(lfmq.build_model(cfg, seed=77), qcfg)
+        qm = shard_llama(qm, rank, world)
+        ids = torch.arange(12).remainder(256).reshape(1, 12)
+        logits = qm(ids, qm.make_cache(1, 32))
+        if rank == 0:
+            torch.save({"logits": logits}, os.path.join(out_dir, "tp8.pt"))
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_tp_gloo_world8_bench_path(tmp_path):
+    """8-way TP through the exact bench.py shard route (the width the
+    driver's SCALE run uses on llama3-8b: hkv 8 => kv head per rank)."""
+    cfg = LlamaConfig(vocab_size=256, hidden_size=256, intermediate_size=512,
+                      n_layers=2, n_heads=8, n_kv_heads=8, head_dim=32,
+                      max_seq=64)
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=77), qcfg)
+    ids = torch.arange(12).remainder(256).reshape(1, 12)
+    ref = qm(ids, qm.make_cache(1, 32))
+    mp.spawn(_worker_bench_path, args=(8, 29780, str(tmp_path)), nprocs=8,
+             join=True)
+    out = torch.load(str(tmp_path / "tp8.pt"), weights_only=False)
+    rel = (out["logits"].float() - ref.float()).abs().max() / \
+        ref.float().abs().max()
+    assert rel < 0.05, f"TP8 logits rel err {float(rel)}"
