@@ -107,10 +107,17 @@ class ExpertParallelMoE(nn.Module):
         yl.index_add_(0, src_tok, back_rows.float() * slot_w.unsqueeze(-1))
         yl = yl.to(x2.dtype)
 
-        # re-replicate: all_gather the shards
-        parts = [x2.new_empty(sizes[r], h) for r in range(world)]
-        dist.all_gather(parts, yl.contiguous())
-        return torch.cat(parts, dim=0).reshape(b, s, h)
+        # re-replicate the shards.  all_gather needs EQUAL sizes on every
+        # rank (gloo and RCCL both reject ragged lists), but the token
+        # shards are uneven whenever t % world != 0 — pad to the max
+        # shard and slice per rank after the collective.
+        mx = max(sizes)
+        buf = x2.new_zeros(mx, h)
+        buf[:yl.shape[0]] = yl
+        parts = [x2.new_empty(mx, h) for _ in range(world)]
+        dist.all_gather(parts, buf)
+        return torch.cat([parts[r][:sizes[r]] for r in range(world)],
+                         dim=0).reshape(b, s, h)
 
 
 @torch.no_grad()

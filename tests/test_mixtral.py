@@ -112,3 +112,31 @@ def test_mixtral_tp_load_raises_cleanly(tmp_path):
     lfmq.save(qm, p, qcfg)
     with pytest.raises(NotImplementedError):
         lfmq.load(p, tp_rank=0, tp_world=2)
+
+
+def _ep_worker8(rank, world, port, tmpdir):
+    """EP at the 8-rank width the 8-GPU node would use: 8 experts, one
+    per rank (the mixtral 8x7B deployment shape)."""
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        cfg = MixtralConfig.tiny(n_experts=8)
+        qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+        qm = lfmq.quantize(lfmq.build_model(cfg, seed=53), qcfg)
+        ids = torch.arange(10).remainder(cfg.vocab_size).reshape(1, 10)
+        ref_logits = qm(ids, qm.make_cache(1, 32))
+        from lfmq.dist.ep import apply_expert_parallel
+        apply_expert_parallel(qm, rank, world)
+        ep_logits = qm(ids, qm.make_cache(1, 32))
+        rel = (ep_logits.float() - ref_logits.float()).abs().max() / \
+            ref_logits.float().abs().max()
+        assert rel < 0.03, f"EP8 mismatch rel={float(rel)}"
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ep_gloo_world8_one_expert_per_rank(tmp_path):
+    mp.spawn(_ep_worker8, args=(8, 29791, str(tmp_path)), nprocs=8, join=True)
