@@ -42,6 +42,12 @@ def build_quantized_llama(model_name: str, scheme: str, bits: int, g: int,
     elif model_name == "mixtral-8x7b":
         from ..models.mixtral import MixtralConfig
         cfg = MixtralConfig.mixtral_8x7b()
+    elif model_name == "tiny-debug":
+        # contract-test shape only (tests/test_bench_contract.py);
+        # never a reportable benchmark config
+        cfg = LlamaConfig(vocab_size=512, hidden_size=256,
+                          intermediate_size=512, n_layers=2, n_heads=4,
+                          n_kv_heads=2, head_dim=64, max_seq=256)
     else:
         raise ValueError(model_name)
     if n_layers:
