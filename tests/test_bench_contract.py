@@ -41,3 +41,10 @@ def test_report_rows_to_markdown():
     md = rows_to_markdown(rows)
     assert "GPTQ" in md and "506.8" in md and "8.5e-05" in md
     assert md.count("\n") == 2  # header + separator + one row
+
+
+def test_scaling_runner_single_point():
+    from lfmq.bench.scaling import run_point
+    r = run_point(1, steps=2, warmup=1,
+                  extra=["--model", "tiny-debug", "--prompt", "8"])
+    assert r["metric"] == "decode tok/s" and r["n_gpus"] == 1
