@@ -85,7 +85,11 @@ def quantize(model: str = typer.Option("llama3-8b"),
 def eval(ckpt: str = typer.Option(...),
          seq_len: int = typer.Option(2048),
          n_seqs: int = typer.Option(4),
-         seed: int = typer.Option(777)):
+         seed: int = typer.Option(777),
+         model_seed: int = typer.Option(0, help="seed the checkpointed "
+                                        "model was built with (quantize "
+                                        "--seed); the bf16 reference must "
+                                        "match it or delta-ppl is noise")):
     """Delta-perplexity vs the bf16 model (BASELINE.md protocol):
     teacher-forced NLL on an identical fixed-seed synthetic stream,
     plus logit KL and per-layer weight MSE (random-init caveat)."""
@@ -96,8 +100,7 @@ def eval(ckpt: str = typer.Option(...),
     dev = _device()
     meta = load_meta(ckpt)
     qm = lfmq.load(ckpt, device=dev)
-    fp = lfmq.build_model(meta["model"], seed=0)
-    fp.init_random(0)
+    fp = lfmq.build_model(meta["model"], seed=model_seed)
     fp = fp.to(dev)
     r = bench_ppl(fp, qm, qm.cfg.vocab_size, seq_len=seq_len, n_seqs=n_seqs,
                   seed=seed, device=dev)

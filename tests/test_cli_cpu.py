@@ -1,0 +1,30 @@
+"""CPU CLI coverage: quantize -> eval round trip on the tiny preset
+(the GPU variant lives in test_gpu_e2e.py::test_cli_quantize_eval_gpu)."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run(args, **kw):
+    return subprocess.run([sys.executable, "-m", "lfmq.cli"] + args,
+                          capture_output=True, text=True, cwd=REPO,
+                          timeout=300, **kw)
+
+
+def test_cli_quantize_eval_cpu(tmp_path):
+    ck = str(tmp_path / "ck")
+    r = _run(["quantize", "--model", "tiny-llama", "--scheme", "rtn",
+              "--group-size", "32", "--out", ck, "--seed", "3"])
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert os.path.exists(os.path.join(ck, "model.safetensors"))
+    r = _run(["eval", "--ckpt", ck, "--seq-len", "32", "--n-seqs", "1",
+              "--model-seed", "3"])
+    assert r.returncode == 0, r.stderr[-1500:]
+    out = json.loads(r.stdout[r.stdout.index("{"):])
+    assert "delta_ppl" in out and "logit_kl" in out
+    # quantization noise only: the matching-seed bf16 reference keeps the
+    # divergence small; a wrong model_seed would give KL ~ ln(vocab)
+    assert out["logit_kl"] < 1.0, out
