@@ -93,6 +93,23 @@ def bench_decode(model, cfg, batch: int, prompt: int, steps: int, warmup: int,
                 torch.cuda.synchronize()  # leave no half-captured stream state
             except Exception:
                 pass
+        if lfmq_dist.get_world_size() > 1:
+            # All ranks must agree on the mode: a graphed rank replays its
+            # captured collectives while an eager rank issues fresh ones —
+            # a split decision deadlocks the TP all-reduces.  MIN-reduce
+            # the outcome and drop to eager everywhere if any rank failed.
+            import sys
+
+            import torch.distributed as dist
+            ok = torch.tensor([1.0 if graphed else 0.0], dtype=torch.float32)
+            ok = ok.to(device)
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if graphed and float(ok.cpu()[0]) < 1.0:
+                print("[bench] peer rank failed graph capture; all ranks "
+                      "fall back to eager", file=sys.stderr)
+                graphed = False
+                del dec
+                torch.cuda.synchronize()
     if not graphed:
         cache = model.make_cache(batch, max_seq=prompt + warmup + steps + 8,
                                  device=device)
