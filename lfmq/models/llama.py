@@ -338,16 +338,8 @@ class LlamaForCausalLM(nn.Module):
 
     @torch.no_grad()
     def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
-                 cache: Optional[KVCache] = None) -> torch.Tensor:
-        b, s = input_ids.shape
-        if cache is None:
-            cache = self.make_cache(b, max_seq=s + max_new_tokens + 8)
-        logits = self.forward(input_ids, cache)
-        out = [input_ids]
-        tok = F.greedy_sample(logits[:, -1]).unsqueeze(1)
-        for _ in range(max_new_tokens - 1):
-            out.append(tok)
-            logits = self.forward(tok, cache)
-            tok = F.greedy_sample(logits[:, -1]).unsqueeze(1)
-        out.append(tok)
-        return torch.cat(out, dim=1)
+                 cache: Optional[KVCache] = None, gen_cfg=None,
+                 generator=None) -> torch.Tensor:
+        from .generation import autoregressive_generate
+        return autoregressive_generate(self, input_ids, max_new_tokens,
+                                       cache, gen_cfg, generator)

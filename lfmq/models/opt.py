@@ -135,16 +135,8 @@ class OPTForCausalLM(nn.Module):
         return logits
 
     @torch.no_grad()
-    def generate(self, input_ids, max_new_tokens: int = 16, cache=None):
-        b, s = input_ids.shape
-        if cache is None:
-            cache = self.make_cache(b, max_seq=s + max_new_tokens + 8)
-        logits = self.forward(input_ids, cache)
-        out = [input_ids]
-        tok = logits[:, -1:].argmax(dim=-1)
-        for _ in range(max_new_tokens - 1):
-            out.append(tok)
-            logits = self.forward(tok, cache)
-            tok = logits[:, -1:].argmax(dim=-1)
-        out.append(tok)
-        return torch.cat(out, dim=1)
+    def generate(self, input_ids, max_new_tokens: int = 16, cache=None,
+                 gen_cfg=None, generator=None):
+        from .generation import autoregressive_generate
+        return autoregressive_generate(self, input_ids, max_new_tokens,
+                                       cache, gen_cfg, generator)

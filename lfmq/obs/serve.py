@@ -26,6 +26,10 @@ TOKS_PER_S = Gauge("lfmq_decode_tok_per_s", "last request decode tok/s")
 class GenRequest(BaseModel):
     input_ids: List[List[int]]
     max_new_tokens: int = 32
+    greedy: bool = True
+    temperature: float = 1.0
+    eos_token_id: int | None = None
+    seed: int | None = None
 
 
 def make_app(model) -> FastAPI:
@@ -41,9 +45,18 @@ def make_app(model) -> FastAPI:
     def generate(req: GenRequest):
         REQS.inc()
         ids = torch.tensor(req.input_ids, dtype=torch.int64, device=device)
+        from ..config import GenerationConfig
+        gc = GenerationConfig(max_new_tokens=req.max_new_tokens,
+                              greedy=req.greedy, temperature=req.temperature,
+                              eos_token_id=req.eos_token_id)
+        gen = None
+        if req.seed is not None:
+            # generator device must match the logits device (multinomial)
+            gen = torch.Generator(device=str(device)).manual_seed(req.seed)
         t0 = time.perf_counter()
         with torch.no_grad():
-            out = model.generate(ids, max_new_tokens=req.max_new_tokens)
+            out = model.generate(ids, max_new_tokens=req.max_new_tokens,
+                                 gen_cfg=gc, generator=gen)
         if device.type == "cuda":
             torch.cuda.synchronize()
         dt = time.perf_counter() - t0

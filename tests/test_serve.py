@@ -21,3 +21,25 @@ def test_serve_generate_and_metrics():
     assert len(body["output_ids"][0]) == 8
     r = client.get("/metrics")
     assert r.status_code == 200 and b"lfmq_tokens_total" in r.content
+
+
+def test_serve_sampling_options():
+    import torch
+    from fastapi.testclient import TestClient
+
+    import lfmq
+    from lfmq.models.llama import LlamaConfig
+    from lfmq.obs.serve import make_app
+
+    cfg = LlamaConfig.tiny()
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=2),
+                       lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32))
+    client = TestClient(make_app(qm))
+    body = {"input_ids": [[1, 2, 3, 4]], "max_new_tokens": 6,
+            "greedy": False, "temperature": 0.7, "seed": 9}
+    a = client.post("/generate", json=body).json()
+    b = client.post("/generate", json=body).json()
+    assert a["output_ids"] == b["output_ids"]      # seeded => reproducible
+    assert a["new_tokens"] == 6
+    toks = a["output_ids"][0][4:]
+    assert all(0 <= t < cfg.vocab_size for t in toks)
