@@ -56,9 +56,13 @@ def quantize(model: str = typer.Option("llama3-8b"),
              calib_seqs: int = typer.Option(16),
              calib_len: int = typer.Option(512),
              resume: bool = typer.Option(False, help="resume a partial run"),
-             act_order: bool = typer.Option(False)):
-    """Random-init a model (no network => synthetic weights), calibrate,
-    quantize and write the packed checkpoint."""
+             act_order: bool = typer.Option(False),
+             hf_dir: Optional[str] = typer.Option(
+                 None, help="local HF-transformers checkpoint dir "
+                 "(safetensors shards) to load into the preset before "
+                 "quantizing; default is random-init")):
+    """Build a model (random-init, or --hf-dir pretrained weights),
+    calibrate, quantize and write the packed checkpoint."""
     import lfmq
     from lfmq.obs import phase_timer
 
@@ -68,6 +72,9 @@ def quantize(model: str = typer.Option("llama3-8b"),
                             act_order=act_order)
     with phase_timer("build model"):
         m, cfg = _build_preset(model, seed, layers)
+        if hf_dir is not None:
+            from lfmq.models.import_hf import import_hf_state
+            import_hf_state(m, hf_dir)
         m = m.to(dev)
     with phase_timer(f"{scheme} quantize"):
         if scheme == "gptq" and resume:
