@@ -89,6 +89,32 @@ def quantize(model: str = typer.Option("llama3-8b"),
 
 
 @app.command()
+def info(ckpt: str = typer.Option(...)):
+    """Inspect a packed checkpoint: quant config, model shape, size."""
+    import os
+
+    from lfmq.checkpoint import MODEL_FILE, load_meta
+    meta = load_meta(ckpt)
+    q = meta["quant"]
+    mc = meta["model"]
+    man = meta.get("manifest", {})
+    size = os.path.getsize(os.path.join(ckpt, MODEL_FILE))
+    bits_total = sum(s["in"] * s["out"] for s in man.values())
+    typer.echo(json.dumps({
+        "format_version": meta["format_version"],
+        "scheme": q["scheme"], "bits": q["bits"],
+        "group_size": q["group_size"], "sym": q["sym"],
+        "arch": mc.get("arch"), "n_layers": mc.get("n_layers"),
+        "hidden_size": mc.get("hidden_size"),
+        "vocab_size": mc.get("vocab_size"),
+        "quantized_linears": len(man),
+        "quantized_params": bits_total,
+        "file_bytes": size,
+        "file_gib": round(size / (1 << 30), 3),
+    }, indent=2))
+
+
+@app.command()
 def eval(ckpt: str = typer.Option(...),
          seq_len: int = typer.Option(2048),
          n_seqs: int = typer.Option(4),

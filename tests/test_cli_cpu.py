@@ -28,3 +28,15 @@ def test_cli_quantize_eval_cpu(tmp_path):
     # quantization noise only: the matching-seed bf16 reference keeps the
     # divergence small; a wrong model_seed would give KL ~ ln(vocab)
     assert out["logit_kl"] < 1.0, out
+
+
+def test_cli_info(tmp_path):
+    ck = str(tmp_path / "ck")
+    r = _run(["quantize", "--model", "tiny-llama", "--scheme", "rtn",
+              "--group-size", "32", "--out", ck])
+    assert r.returncode == 0, r.stderr[-1500:]
+    r = _run(["info", "--ckpt", ck])
+    assert r.returncode == 0, r.stderr[-1500:]
+    out = json.loads(r.stdout[r.stdout.index("{"):])
+    assert out["scheme"] == "rtn" and out["bits"] == 4
+    assert out["quantized_linears"] > 0 and out["file_bytes"] > 0
