@@ -166,7 +166,14 @@ class AWQQuantizer:
         target_names = {n for n, _ in iter_target_linears(model, cfg)}
         plan = _scale_plan(model)
 
-        for bname, block in decoder_blocks(model):
+        blocks = list(decoder_blocks(model))
+        from ..obs import get_logger
+        log = get_logger()
+        import time as _time
+        t_run = _time.perf_counter()
+        for bi, (bname, block) in enumerate(blocks):
+            log.info("awq block %d/%d (%s) [%.1fs elapsed]", bi + 1,
+                     len(blocks), bname, _time.perf_counter() - t_run)
             block.to(device)
             linears = {n: m for n, m in block.named_modules()
                        if isinstance(m, nn.Linear)

@@ -148,7 +148,14 @@ class GPTQQuantizer:
             os.makedirs(os.path.join(resume_dir, "blocks"), exist_ok=True)
 
         target_names = {n for n, _ in iter_target_linears(model, cfg)}
-        for bi, (bname, block) in enumerate(decoder_blocks(model)):
+        blocks = list(decoder_blocks(model))
+        from ..obs import get_logger
+        log = get_logger()
+        import time as _time
+        t_run = _time.perf_counter()
+        for bi, (bname, block) in enumerate(blocks):
+            log.info("gptq block %d/%d (%s) [%.1fs elapsed]", bi + 1,
+                     len(blocks), bname, _time.perf_counter() - t_run)
             block.to(device)
             if resume_dir:
                 bpath = os.path.join(resume_dir, "blocks", f"{bi:04d}.pt")
