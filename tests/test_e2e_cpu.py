@@ -104,3 +104,24 @@ def test_opt_awq_e2e(tmp_path):
     pf = torch.log_softmax(lf, -1); pq = torch.log_softmax(lq, -1)
     kl = (pf.exp() * (pf - pq)).sum(-1).mean()
     assert kl < 0.5, float(kl)
+
+
+def test_sym_quant_e2e_round_trip(tmp_path):
+    """sym=True end-to-end: quantize -> save -> load -> decode; the
+    packed layout stores the constant sym zero like any other zero."""
+    import torch
+
+    import lfmq
+    from lfmq.models.llama import LlamaConfig
+    cfg = LlamaConfig.tiny()
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32, sym=True)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=47), qcfg)
+    p = str(tmp_path / "sym")
+    lfmq.save(qm, p, qcfg)
+    back = lfmq.load(p)
+    ids = torch.arange(8).remainder(cfg.vocab_size).reshape(1, 8)
+    a = qm(ids, qm.make_cache(1, 16))
+    b = back(ids, back.make_cache(1, 16))
+    assert torch.equal(a, b)
+    out = back.generate(ids[:, :4], max_new_tokens=4)
+    assert out.shape == (1, 8)
