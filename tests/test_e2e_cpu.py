@@ -125,3 +125,21 @@ def test_sym_quant_e2e_round_trip(tmp_path):
     assert torch.equal(a, b)
     out = back.generate(ids[:, :4], max_new_tokens=4)
     assert out.shape == (1, 8)
+
+
+def test_per_channel_quant_e2e(tmp_path):
+    """group_size=-1 (per-channel: one group = the whole row) through
+    quantize -> save -> load -> forward."""
+    import torch
+
+    import lfmq
+    from lfmq.models.llama import LlamaConfig
+    cfg = LlamaConfig.tiny()
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=-1)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=49), qcfg)
+    p = str(tmp_path / "pc")
+    lfmq.save(qm, p, qcfg)
+    back = lfmq.load(p)
+    ids = torch.arange(8).remainder(cfg.vocab_size).reshape(1, 8)
+    assert torch.equal(qm(ids, qm.make_cache(1, 16)),
+                       back(ids, back.make_cache(1, 16)))
