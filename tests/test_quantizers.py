@@ -40,13 +40,14 @@ def _naive_gptq(w, h, cfg):
     return codes, scales, zeros
 
 
-def test_gptq_blocked_equals_naive():
+@pytest.mark.parametrize("bits", [4, 8])
+def test_gptq_blocked_equals_naive(bits):
     torch.manual_seed(3)
     n, k = 24, 128
     w = torch.randn(n, k)
     x = torch.randn(300, k)
     h = 2 * x.t() @ x
-    cfg = QuantConfig(scheme="gptq", bits=4, group_size=32, block_size=32)
+    cfg = QuantConfig(scheme="gptq", bits=bits, group_size=32, block_size=32)
     c1, s1, z1 = _naive_gptq(w, h, cfg)
     c2, s2, z2, gi = gptq_quantize_layer(w, h, cfg)
     assert gi is None
