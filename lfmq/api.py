@@ -115,8 +115,18 @@ def load(path: str, device: str = "cpu", tp_rank: int = 0, tp_world: int = 1,
     # module tree (no CPU materialization of a second copy): a 70B W4
     # checkpoint (~35 GB packed) loads with one host pass, not two.
     state = load_packed(path, device=str(device))
+    if dtype is not None:
+        # assign=True would otherwise let the checkpoint dtype override an
+        # explicitly requested one: cast float tensors here.
+        state = {k: (v.to(dtype) if v.is_floating_point() else v)
+                 for k, v in state.items()}
     model.load_state_dict(state, strict=True, assign=True)
     model = model.to(device)  # anything not in the checkpoint (none today)
+    # assign=True unties tied embeddings (two independent tensors after the
+    # load): re-tie so memory and semantics match build time.
+    if getattr(model.cfg, "tie_embeddings", False) and \
+            isinstance(getattr(model, "lm_head", None), nn.Linear):
+        model.lm_head.weight = model.embed_tokens.weight
     if getattr(model.cfg, "arch", None) == "llama":
         from .quant.fold import refresh_norm_fold_flags
         refresh_norm_fold_flags(model)

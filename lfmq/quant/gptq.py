@@ -165,6 +165,12 @@ class GPTQQuantizer:
                         lin = block.get_submodule(lname)
                         q = QLinear(meta["in"], meta["out"], meta["bits"],
                                     meta["g"], bias=meta["bias"]).to(device)
+                        if meta.get("g_idx"):
+                            # act_order checkpoints carry a g_idx buffer;
+                            # register it so load_state_dict finds the key
+                            # (resume of act_order=True failed).
+                            q.register_buffer("g_idx", torch.zeros(
+                                meta["in"], dtype=torch.int32, device=device))
                         replace_module(block, lname, q)
                     block.load_state_dict(saved["state"])
                     xs = [run_block(block, x.to(device)).to(xs[0].device)
@@ -207,7 +213,8 @@ class GPTQQuantizer:
                         man[lname] = {"in": mod.in_features,
                                       "out": mod.out_features,
                                       "bits": mod.bits, "g": mod.group_size,
-                                      "bias": mod.bias is not None}
+                                      "bias": mod.bias is not None,
+                                      "g_idx": hasattr(mod, "g_idx")}
                 fd, tmp = tempfile.mkstemp(dir=os.path.join(resume_dir, "blocks"))
                 os.close(fd)
                 torch.save({"state": {k: v.cpu() for k, v in

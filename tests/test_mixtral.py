@@ -140,3 +140,47 @@ def _ep_worker8(rank, world, port, tmpdir):
 
 def test_ep_gloo_world8_one_expert_per_rank(tmp_path):
     mp.spawn(_ep_worker8, args=(8, 29791, str(tmp_path)), nprocs=8, join=True)
+
+
+def _ep_paths_worker(rank, world, port, tmpdir):
+    """Same model through BOTH dispatch protocols: the static
+    (capacity-padded, sync-free) decode path and the dynamic ragged path
+    must both match the unsharded reference, including decode t=1."""
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import lfmq.dist.ep as ep
+        cfg = MixtralConfig.tiny()
+        qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+        qm = lfmq.quantize(lfmq.build_model(cfg, seed=23), qcfg)
+        ids = torch.arange(9).remainder(cfg.vocab_size).reshape(1, 9)
+        one = torch.tensor([[3]])
+        ref_logits = qm(ids, qm.make_cache(1, 32))
+        ref_one = qm(one, qm.make_cache(1, 32))
+        ep.apply_expert_parallel(qm, rank, world)
+
+        saved = ep.STATIC_SLOT_LIMIT
+        try:
+            ep.STATIC_SLOT_LIMIT = 10 ** 9        # force static
+            st = qm(ids, qm.make_cache(1, 32))
+            st1 = qm(one, qm.make_cache(1, 32))   # decode shape t=1
+            ep.STATIC_SLOT_LIMIT = 0              # force dynamic
+            dy = qm(ids, qm.make_cache(1, 32))
+        finally:
+            ep.STATIC_SLOT_LIMIT = saved
+        for got, want, tag in ((st, ref_logits, "static"),
+                               (st1, ref_one, "static-t1"),
+                               (dy, ref_logits, "dynamic")):
+            rel = (got.float() - want.float()).abs().max() / \
+                want.float().abs().max()
+            assert rel < 0.03, f"EP {tag} mismatch rel={float(rel)}"
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ep_gloo_static_and_dynamic_paths(tmp_path):
+    mp.spawn(_ep_paths_worker, args=(2, 29815, str(tmp_path)), nprocs=2,
+             join=True)
