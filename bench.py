@@ -22,8 +22,8 @@ import torch
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=64)
-    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--steps", type=int, default=256)
+    ap.add_argument("--warmup", type=int, default=16)
     ap.add_argument("--model", default="llama3-8b")
     ap.add_argument("--batch", type=int, default=1)
     ap.add_argument("--prompt", type=int, default=128)
@@ -33,6 +33,8 @@ def main():
                     help="override layer count (debug only; 0 = full model)")
     ap.add_argument("--quant-lm-head", action="store_true",
                     help="quantize the lm_head too (default: bf16 head)")
+    ap.add_argument("--no-fidelity", action="store_true",
+                    help="skip the delta-ppl/logit-KL fidelity pass")
     args = ap.parse_args()
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -58,6 +60,25 @@ def main():
     r = bench_decode(model, cfg, args.batch, args.prompt, args.steps,
                      args.warmup, device)
 
+    # Fidelity field of the contract metric ("decode tok/s + delta-ppl vs
+    # fp16", BASELINE.json:2): teacher-forced ppl + logit KL of the
+    # quantized model vs the same-seed bf16 reference on a fixed synthetic
+    # stream.  Measured at world==1 (the sharded model's fidelity is the
+    # same weights; TP changes only the execution plan).
+    delta_ppl = logit_kl = None
+    if world == 1 and not args.no_fidelity:
+        from lfmq.bench.decode import build_bf16_llama
+        from lfmq.bench.ppl import bench_ppl
+        fp_model, _ = build_bf16_llama(args.model, device, seed=0,
+                                       n_layers=args.layers or None)
+        seq = min(512, cfg.max_seq)
+        pr = bench_ppl(fp_model, model, cfg.vocab_size, seq_len=seq,
+                       n_seqs=2, device=device)
+        delta_ppl, logit_kl = pr["delta_ppl"], pr["logit_kl"]
+        del fp_model
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+
     if rank == 0:
         out = {
             "metric": "decode tok/s",
@@ -72,6 +93,8 @@ def main():
             "vs_baseline": None,
             "dtype": "bf16",
             "data": "synthetic",
+            "delta_ppl": delta_ppl,
+            "logit_kl": logit_kl,
             "config": {
                 "model": args.model,
                 "quant": f"w{args.bits}a16-g{args.group_size}-rtn"

@@ -28,13 +28,7 @@ def _fast_random_init(model, seed: int):
     return model
 
 
-@torch.no_grad()
-def build_quantized_llama(model_name: str, scheme: str, bits: int, g: int,
-                          device, tp_rank: int = 0, tp_world: int = 1,
-                          seed: int = 0, n_layers: Optional[int] = None,
-                          quant_lm_head: bool = False):
-    """Random-init on device, quantize (RTN weights for bench speed: the
-    decode kernel path and memory traffic are scheme-independent), shard."""
+def _bench_config(model_name: str, n_layers: Optional[int] = None):
     if model_name == "llama3-8b":
         cfg = LlamaConfig.llama3_8b()
     elif model_name == "llama3-70b":
@@ -52,6 +46,30 @@ def build_quantized_llama(model_name: str, scheme: str, bits: int, g: int,
         raise ValueError(model_name)
     if n_layers:
         cfg.n_layers = n_layers
+    return cfg
+
+
+@torch.no_grad()
+def build_bf16_llama(model_name: str, device, seed: int = 0,
+                     n_layers: Optional[int] = None):
+    """The un-quantized reference with the SAME init as the bench model
+    (same cfg/seed/init path as build_quantized_llama before quantize) —
+    the fp baseline for the contract's delta-ppl/logit-KL fidelity field."""
+    cfg = _bench_config(model_name, n_layers)
+    model = build_model(cfg)
+    model = model.to_empty(device=str(device)) if str(device) != "cpu" else model
+    _fast_random_init(model, seed)
+    return model, cfg
+
+
+@torch.no_grad()
+def build_quantized_llama(model_name: str, scheme: str, bits: int, g: int,
+                          device, tp_rank: int = 0, tp_world: int = 1,
+                          seed: int = 0, n_layers: Optional[int] = None,
+                          quant_lm_head: bool = False):
+    """Random-init on device, quantize (RTN weights for bench speed: the
+    decode kernel path and memory traffic are scheme-independent), shard."""
+    cfg = _bench_config(model_name, n_layers)
     model = build_model(cfg)
     model = model.to_empty(device=str(device)) if str(device) != "cpu" else model
     _fast_random_init(model, seed)
