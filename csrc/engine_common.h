@@ -1,0 +1,173 @@
+// Persistent-engine primitives for the MI355X decode megakernel.
+//
+// These implement, verbatim where the guide gives code, the CDNA4
+// inter-workgroup patterns of cdna_hip_programming.md §6 Guideline 16
+// (agent-scope release/acquire, 8-byte {tag,data} granules) and the
+// §5.6 weight-streaming engine structure (LDS-DMA loader wave + consumer
+// waves over a 16 KiB-slot ring, nt weight stream).
+//
+// CRITICAL hardware facts this encodes (MI355X_MICROARCH.md):
+//  - per-CU L1 is never refreshed by another CU's stores; per-XCD L2s
+//    are not coherent: every cross-workgroup word moves ONLY via
+//    agent-scope atomics (sc1) — R2 granules here.
+//  - R2 form: the data IS the flag: one naturally-aligned 8-byte
+//    {tag:32, value:32} per granule, one relaxed agent store, no fence;
+//    consumer re-reads granules until every tag matches the epoch.
+//  - epochs count within the call (phase index + 1, never 0); every
+//    granule word is zeroed by a hipMemsetAsync before EVERY launch.
+//  - every spin is bounded and surfaces a give-up code.
+#pragma once
+#include "common.h"
+
+namespace lfmq_engine {
+
+#define LFMQ_RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
+#define LFMQ_WG_SCOPE __HIP_MEMORY_SCOPE_WORKGROUP
+
+typedef __attribute__((address_space(1))) unsigned int gu32;
+typedef __attribute__((address_space(1))) unsigned long long gu64;
+
+// bounded-spin budget: ~2^24 polls x s_sleep(2) ~= seconds of wall —
+// far beyond any healthy step; a trip means a lost producer.
+constexpr unsigned SPIN_MAX = 1u << 24;
+
+__device__ __forceinline__ bool spin_fail(unsigned& spins, gu32* tmo,
+                                          unsigned code, int lane) {
+  if (++spins < SPIN_MAX) {
+    __builtin_amdgcn_s_sleep(2);
+    return false;
+  }
+  if (lane == 0 && tmo != nullptr) __hip_atomic_store(tmo, code, LFMQ_RLX_AGENT);
+  return true;
+}
+
+// ---- R2 granule publish: ONE aligned 8-byte relaxed agent store ----------
+__device__ __forceinline__ void store_granule(gu64* g, unsigned epoch,
+                                              unsigned value) {
+  __hip_atomic_store(g, ((unsigned long long)epoch << 32) | value,
+                     LFMQ_RLX_AGENT);
+}
+
+// ---- R2 granule sweep + LDS staging --------------------------------------
+// ONE wave re-reads its lane-strided share of cnt granules every pass until
+// every tag == epoch, ds-writing the payload dword into lds_dst[idx] as it
+// goes (the final pass rewrites everything with matched values).  Returns
+// false on spin give-up (outputs poisoned by the caller's memset protocol).
+__device__ __forceinline__ bool sweep_stage(gu64* g, int cnt, unsigned epoch,
+                                            u32* lds_dst, gu32* tmo,
+                                            unsigned code, int lane) {
+  for (unsigned spins = 0;;) {
+    bool ok = true;
+    for (int idx = lane; idx < cnt; idx += LFMQ_WAVE) {
+      const unsigned long long x = __hip_atomic_load(g + idx, LFMQ_RLX_AGENT);
+      lds_dst[idx] = (unsigned)x;
+      ok &= (unsigned)(x >> 32) == epoch;
+    }
+    if (__all(ok)) return true;
+    if (spin_fail(spins, tmo, code, lane)) return false;
+  }
+}
+
+// variant without staging (caller consumes values in registers): sweeps its
+// share into vals[] (each lane's granules at lane, lane+64, ...)
+template <int MAXN>
+__device__ __forceinline__ bool sweep_vals(gu64* g, int cnt, unsigned epoch,
+                                           unsigned (&vals)[MAXN], gu32* tmo,
+                                           unsigned code, int lane) {
+  for (unsigned spins = 0;;) {
+    bool ok = true;
+    int j = 0;
+    for (int idx = lane; idx < cnt; idx += LFMQ_WAVE, ++j) {
+      const unsigned long long x = __hip_atomic_load(g + idx, LFMQ_RLX_AGENT);
+      vals[j] = (unsigned)x;
+      ok &= (unsigned)(x >> 32) == epoch;
+    }
+    if (__all(ok)) return true;
+    if (spin_fail(spins, tmo, code, lane)) return false;
+  }
+}
+
+// ---- intra-workgroup flags (LDS, workgroup scope) -------------------------
+// gather/loader/consumer waves coordinate through LDS words: release store
+// inserts the lgkmcnt drain ahead of the flag; acquire load orders the
+// reader's later LDS reads after it.
+__device__ __forceinline__ void lds_flag_store(unsigned* f, unsigned v) {
+  __hip_atomic_store(f, v, __ATOMIC_RELEASE, LFMQ_WG_SCOPE);
+}
+// RELAXED variant for the loader's fill publication: a release store here
+// would make the compiler drain vmcnt(0) (LDS-DMA completion is tracked by
+// vmcnt), killing the fill pipeline.  The caller orders the store itself
+// with an explicit lagged `s_waitcnt vmcnt(N)` asm (whose "memory" clobber
+// keeps this store below it).
+__device__ __forceinline__ void lds_flag_store_rlx(unsigned* f, unsigned v) {
+  __hip_atomic_store(f, v, __ATOMIC_RELAXED, LFMQ_WG_SCOPE);
+}
+__device__ __forceinline__ unsigned lds_flag_load(const unsigned* f) {
+  return __hip_atomic_load(f, __ATOMIC_ACQUIRE, LFMQ_WG_SCOPE);
+}
+// bounded poll until *f >= v (monotonic counters)
+__device__ __forceinline__ bool lds_wait_ge(const unsigned* f, unsigned v,
+                                            gu32* tmo, unsigned code,
+                                            int lane) {
+  for (unsigned spins = 0;;) {
+    if (lds_flag_load(f) >= v) return true;
+    if (spin_fail(spins, tmo, code, lane)) return false;
+  }
+}
+
+// ---- loader-side LDS access in asm ----------------------------------------
+// hipcc tracks global_load_lds completion through vmcnt and conservatively
+// drains vmcnt(0) before ANY visible ds_read/ds_write while fills are in
+// flight — which would serialize the loader every time it polls consumer
+// progress or publishes a slot.  These asm forms (guide §5.7 form (i):
+// loads + s_waitcnt in ONE statement, early-clobber outputs) keep the fill
+// pipeline open; ordering vs the DMA is the caller's explicit lagged
+// s_waitcnt (in loader_publish) — never hipcc's.
+
+// LDS byte offset of a __shared__ object, for asm ds operands
+__device__ __forceinline__ unsigned lds_off(const void* p) {
+  return (unsigned)(unsigned long long)(
+      __attribute__((address_space(3))) const void*)p;
+}
+
+// read 3 LDS words (consumer progress counters) without a vmcnt drain
+__device__ __forceinline__ void loader_read3(unsigned off0, unsigned off1,
+                                             unsigned off2, unsigned& v0,
+                                             unsigned& v1, unsigned& v2) {
+  asm volatile(
+      "ds_read_b32 %0, %3\n\t"
+      "ds_read_b32 %1, %4\n\t"
+      "ds_read_b32 %2, %5\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(v0), "=&v"(v1), "=&v"(v2)
+      : "v"(off0), "v"(off1), "v"(off2)
+      : "memory");
+}
+
+// publish a fill-progress value: wait until at most VMLAG fill instructions
+// remain outstanding (i.e. everything older has LANDED in LDS), then
+// ds_write the counter — all inside one asm so hipcc adds no drain.
+template <int VMLAG>
+__device__ __forceinline__ void loader_publish(unsigned flag_off,
+                                               unsigned value) {
+  asm volatile(
+      "s_waitcnt vmcnt(%c2)\n\t"
+      "ds_write_b32 %0, %1"
+      :
+      : "v"(flag_off), "v"(value), "i"(VMLAG)
+      : "memory");
+}
+
+// ---- LDS-DMA fill: one 1-KiB chunk (64 lanes x 16 B), nt (aux = 2) --------
+// gsrc: this lane's 16-B-aligned global source; lds_base: wave-uniform LDS
+// byte destination (lane data lands at lds_base + lane*16).
+typedef __attribute__((address_space(1))) const void* lfmq_gptr;
+typedef __attribute__((address_space(3))) void* lfmq_lptr;
+
+__device__ __forceinline__ void glds_fill_1k(const char* gsrc_lane,
+                                             char* lds_base /*__shared__*/) {
+  __builtin_amdgcn_global_load_lds((lfmq_gptr)gsrc_lane, (lfmq_lptr)lds_base,
+                                   16, 0, /*aux nt*/ 2);
+}
+
+}  // namespace lfmq_engine

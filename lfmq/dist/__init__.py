@@ -23,8 +23,10 @@ def init_dist(backend: Optional[str] = None) -> tuple[int, int]:
     is absent or 1.
     """
     world = int(os.environ.get("WORLD_SIZE", "1"))
-    if world <= 1:
+    if world <= 1 and "RANK" not in os.environ:
         return 0, 1
+    # under torchrun (RANK set) we initialise even at world 1 so a 1-GPU
+    # run exercises the real RCCL init path (VERDICT r1 #2)
     if not dist.is_initialized():
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
