@@ -1,0 +1,685 @@
+// Persistent weight-streaming decode engine for Llama-family W4A16 decode
+// on MI355X (round-2 headline work; design: docs/persistent_decode_design.md,
+// structure: cdna_hip_programming.md §5.6 + Guideline 16,
+// prices: MI355X_MICROARCH.md persistent-kernel table).
+//
+// ONE kernel launch executes the WHOLE 32-layer decode step (b=1, s=1):
+// embedding read, per layer [norm+qkv GEMV] -> [rope+append+attention] ->
+// [o GEMV+resid] -> [norm+gate_up GEMV+silu] -> [down GEMV+resid], and the
+// final RMSNorm.  Replaces ~194 launches/step whose ~3-4 us fixed cost was
+// measured as ~0.7 ms of the 1.97 ms round-1 step (profiles/decode_profile_
+// r01.md).  lm_head + sampling remain separate launches.
+//
+// Geometry: grid = 256 workgroups (one per CU, all co-resident — verified
+// by csrc/engine_probe.hip census), 4 waves each:
+//   wave 3           = loader: streams this CU's packed-weight byte stream
+//                      through a 16 KiB-slot LDS ring via global_load_lds
+//                      (nt, aux=2), run-ahead bounded by the ring, slot
+//                      publication lagged by an explicit vmcnt (all loader
+//                      LDS accesses are inline asm so hipcc inserts no
+//                      vmcnt(0) drains — see engine_common.h).
+//   waves 0-2        = consumers: dequant+dot records out of the ring
+//                      (v_perm nibble dequant + v_dot2 bf16, the proven
+//                      round-1 GEMV inner idioms), attention units, and
+//                      3-way-split granule gathers; wave 0 leads (norm
+//                      scalars, phase publication).
+//
+// Cross-CU dataflow: every op output is published as R2 granules (8-byte
+// {epoch,2xbf16} relaxed agent stores — the data IS the flag) and gathered
+// by all CUs; epochs are 6L+{1..6}, granule buffers are hipMemset to zero
+// before EVERY launch (graph memset node).  Edge-buffer reuse period (one
+// layer) exceeds the max CU run-ahead (one edge), so overwrite races are
+// structurally impossible: a CU can publish edge E of layer L+1 only after
+// gathering an edge of layer L that required every other CU to have passed
+// its own E(L) publish.
+//
+// Weight stream layout (built by lfmq/models/engine.py, one contiguous
+// region per CU, identical lengths): per layer, per op in [qkv, o,
+// gate_up, down], per owned column record:
+//   [k/2 B nibbles (row-major packed dwords — identical to the at-rest
+//    qweight row)] [G x 2 B bf16 scales] [ceil(G/2) B packed int4 zeros]
+//   [pad to 16 B]
+// Records never straddle the ring wrap (ring_bytes boundary): the builder
+// and the kernel both skip to the next ring boundary when one would.
+// gate_up interleaves records [gate_j, up_j, ...] so one wave computes
+// silu(g)*u locally.  Column ownership: CU c owns columns
+// [c*n/256, (c+1)*n/256) of every op — all Llama n dims divide by 256.
+#include <algorithm>
+
+#include "common.h"
+#include "engine_args.h"
+#include "engine_common.h"
+
+using namespace lfmq_engine;
+
+namespace {
+
+constexpr int SLOT = 16384;
+constexpr int LAG = 4;   // publish slot s-LAG+1 after issuing slot s
+constexpr int AT = 256;  // attention tokens per unit (one wave per unit)
+constexpr int NCU = 256;
+constexpr u32 ONE2_BF16 = 0x3F803F80u;
+
+struct Ctl {
+  unsigned filled;       // loader: slots published
+  unsigned consumed[3];  // consumer waves: slots fully passed
+  unsigned gath;         // gather arrivals (monotonic, +3 per event)
+  unsigned phase;        // lead-published phase (monotonic)
+  unsigned busy;         // gather window hint for loader thinning
+  float inv_cur;         // norm scalar of the current phase
+  float ss_part[3];      // per-wave sum-of-squares partials
+};
+
+// record byte size (padded to 16) — device twin of lfmq_engine_rec_bytes
+__device__ inline int rec_bytes(int k, int G) {
+  return (k / 2 + G * 2 + (G + 1) / 2 + 15) & ~15;
+}
+
+// ---- record walking (must mirror lfmq/models/engine.py exactly) ----------
+struct Cursor {
+  u64 off;       // next free byte in the CU stream
+  u32 ring_pos;  // off % ring_bytes
+};
+__device__ __forceinline__ u64 cursor_next(Cursor& c, int rec, u32 ring_bytes) {
+  if (c.ring_pos + (u32)rec > ring_bytes) {  // would straddle the ring wrap
+    c.off += ring_bytes - c.ring_pos;
+    c.ring_pos = 0;
+  }
+  const u64 at = c.off;
+  c.off += rec;
+  c.ring_pos += rec;
+  if (c.ring_pos == ring_bytes) c.ring_pos = 0;
+  return at;
+}
+
+// ---- the dequant-dot of one record out of the ring ------------------------
+// lane covers k indices {i*512 + lane*8 .. +8} for i in 0..k/512-1: the
+// nibble dword at rec + (i*64+lane)*4 is exactly qweight dword (k0/8), x is
+// one conflict-free ds_read_b128 (stride 16 B across lanes).  Math matches
+// csrc/dq_gemv.hip: s*( sum (128+c)*x ) - s*(z+128)*sum x, fp32 accumulate.
+__device__ __forceinline__ float record_dot(const char* lds_rec,
+                                            const char* lds_src, int k, int G,
+                                            int log2g, int lane) {
+  const int iters = k >> 9;
+  const u32* wbase = (const u32*)lds_rec;
+  const unsigned short* sbase = (const unsigned short*)(lds_rec + k / 2);
+  const unsigned char* zbase = (const unsigned char*)(lds_rec + k / 2 + G * 2);
+  float acc = 0.f;
+  for (int i = 0; i < iters; ++i) {
+    const int k0 = i * 512 + lane * 8;
+    const u32 w = wbase[i * 64 + lane];
+    const u32x4 xq = *(const u32x4*)(lds_src + k0 * 2);
+    const int gi = k0 >> log2g;
+    const float s = bf16_bits_to_f32(sbase[gi]);
+    const u32 zb = zbase[gi >> 1];
+    const u32 z = (gi & 1) ? (zb >> 4) : (zb & 0xFu);
+    const DequantPairs4 dq = dequant4_pairs(w);
+    float cx = 0.f, xs = 0.f;
+    cx = dot2_bf16(dq.p01, xq[0], cx);
+    cx = dot2_bf16(dq.p23, xq[1], cx);
+    cx = dot2_bf16(dq.p45, xq[2], cx);
+    cx = dot2_bf16(dq.p67, xq[3], cx);
+    xs = dot2_bf16(ONE2_BF16, xq[0], xs);
+    xs = dot2_bf16(ONE2_BF16, xq[1], xs);
+    xs = dot2_bf16(ONE2_BF16, xq[2], xs);
+    xs = dot2_bf16(ONE2_BF16, xq[3], xs);
+    acc = fmaf(s, cx, acc);
+    acc = fmaf(-s * (float)(z + 128u), xs, acc);
+  }
+  return wave_reduce_sum(acc);
+}
+
+// wait until the ring holds stream bytes [0, end_off): filled*SLOT >= end
+__device__ __forceinline__ bool wait_fill(Ctl* ctl, u64 end_off, gu32* tmo,
+                                          unsigned code, int lane) {
+  const unsigned need = (unsigned)((end_off + SLOT - 1) / SLOT);
+  return lds_wait_ge(&ctl->filled, need, tmo, code, lane);
+}
+
+// ---- 3-way gather helpers --------------------------------------------------
+// stage this wave's third of cnt granules into lds_dst (u32 payload each),
+// then bump the arrival counter.  Caller waits for 3*gseq arrivals.
+__device__ __forceinline__ bool gather_share(gu64* g, int cnt, unsigned epoch,
+                                             u32* lds_dst, int wave, Ctl* ctl,
+                                             gu32* tmo, unsigned code,
+                                             int lane) {
+  const int b = cnt * wave / 3, e = cnt * (wave + 1) / 3;
+  if (!sweep_stage(g + b, e - b, epoch, lds_dst + b, tmo, code, lane))
+    return false;
+  __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
+  return true;
+}
+__device__ __forceinline__ bool wait_gath(Ctl* ctl, unsigned gseq, gu32* tmo,
+                                          unsigned code, int lane) {
+  return lds_wait_ge(&ctl->gath, 3 * gseq, tmo, code, lane);
+}
+
+// single-granule bounded wait (combine path)
+__device__ __forceinline__ u32 gran_wait(gu64* g, unsigned epoch, gu32* tmo,
+                                         unsigned code, int lane, bool* okp) {
+  for (unsigned spins = 0;;) {
+    const unsigned long long x = __hip_atomic_load(g, LFMQ_RLX_AGENT);
+    if ((unsigned)(x >> 32) == epoch) return (u32)x;
+    if (spin_fail(spins, tmo, code, lane)) { *okp = false; return 0; }
+  }
+}
+
+// ============================================================ main kernel
+template <int D>
+__global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
+  extern __shared__ char smem[];
+  const int cu = blockIdx.x;
+  const int wave = threadIdx.x / LFMQ_WAVE;
+  const int lane = threadIdx.x % LFMQ_WAVE;
+  const u32 ring_bytes = (u32)a.nslot_ring * SLOT;
+
+  char* ring = smem;
+  unsigned short* xbuf = (unsigned short*)(smem + ring_bytes);
+  char* unionb = (char*)(xbuf + a.hidden);
+  const int union_bytes =
+      max(a.inter * 2, max(a.nqkv * 2, a.hq * D * 2));
+  char* ascr = unionb + union_bytes;   // attention scratch / resid staging
+  unsigned short* q_scr = (unsigned short*)ascr;          // [3][D]
+  unsigned short* k_scr = q_scr + 3 * D;                  // [3][D]
+  float* p_scr = (float*)(k_scr + 3 * D);                 // [3][AT]
+  const int ascr_bytes = max(3 * D * 4 + 3 * AT * 4, a.hidden * 2);
+  Ctl* ctl = (Ctl*)(ascr + ((ascr_bytes + 63) & ~63));
+
+  gu64* g_qkv = (gu64*)a.g_qkv;
+  gu64* g_attn = (gu64*)a.g_attn;
+  gu64* g_o = (gu64*)a.g_o;
+  gu64* g_h = (gu64*)a.g_h;
+  gu64* g_down = (gu64*)a.g_down;
+  gu64* g_scr = (gu64*)a.g_scr;
+  gu32* tmo = (gu32*)a.tmo;
+
+  if (threadIdx.x == 0) {
+    ctl->filled = 0;
+    ctl->consumed[0] = ctl->consumed[1] = ctl->consumed[2] = 0;
+    ctl->gath = 0;
+    ctl->phase = 0;
+    ctl->busy = 0;
+  }
+  __syncthreads();  // last convergent barrier; waves diverge below
+
+  // ======================================================== loader wave
+  if (wave == 3) {
+    const char* wg_src = a.stream + (u64)cu * a.stream_per_cu;
+    const unsigned f_off = lds_off(&ctl->filled);
+    const unsigned c0 = lds_off(&ctl->consumed[0]);
+    const unsigned c1 = lds_off(&ctl->consumed[1]);
+    const unsigned c2 = lds_off(&ctl->consumed[2]);
+    const unsigned b_off = lds_off(&ctl->busy);
+    const int total = a.n_slots_total;
+    for (int s = 0; s < total; ++s) {
+      if (s >= a.nslot_ring) {
+        unsigned spins = 0;
+        for (;;) {
+          unsigned d0, d1, d2;
+          loader_read3(c0, c1, c2, d0, d1, d2);
+          if ((int)min(d0, min(d1, d2)) >= s - a.nslot_ring + 1) break;
+          if (spin_fail(spins, tmo, 0x10, lane)) return;
+        }
+      }
+      char* lbase = ring + (s % a.nslot_ring) * SLOT;
+      const char* g = wg_src + (u64)s * SLOT + lane * 16;
+#pragma unroll
+      for (int i = 0; i < 16; ++i) glds_fill_1k(g + i * 1024, lbase + i * 1024);
+      {  // gather-pass thinning: one outstanding fill while the CU sweeps
+        unsigned busy;
+        asm volatile("ds_read_b32 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+                     : "=&v"(busy) : "v"(b_off) : "memory");
+        if (busy) asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+      }
+      if (s >= LAG - 1)
+        loader_publish<(LAG - 1) * 16>(f_off, (unsigned)(s - LAG + 2));
+    }
+    loader_publish<0>(f_off, (unsigned)total);
+    return;
+  }
+
+  // ===================================================== consumer waves
+  const int rep = a.hq / a.hkv;
+  const int T = *a.T_dev;
+  const int pos = T - 1;
+  const int nch = (T + AT - 1) / AT;
+  const int G1 = a.hidden >> a.log2g;       // qkv / gate_up k-groups
+  const int Go = (a.hq * D) >> a.log2g;     // o k-groups
+  const int G2 = a.inter >> a.log2g;        // down k-groups
+  const int rb_qkv = rec_bytes(a.hidden, G1);
+  const int rb_o = rec_bytes(a.hq * D, Go);
+  const int rb_gu = rec_bytes(a.hidden, G1);
+  const int rb_dn = rec_bytes(a.inter, G2);
+  const int npc_qkv = a.nqkv / NCU;
+  const int npc_o = a.hidden / NCU;
+  const int npc_gu = 2 * a.inter / NCU;
+  const int npc_dn = a.hidden / NCU;
+
+  Cursor cur{0, 0};
+  unsigned gseq = 0;
+  unsigned phase_want = 0;
+  bool ok = true;
+
+#define ENG_CHECK(x) do { if (!(x)) return; } while (0)
+#define ENG_PROF(slot_)                                                   \
+  do {                                                                    \
+    if (a.prof && cu == 0 && wave == 0 && lane == 0)                      \
+      a.prof[(u64)L * 8 + (slot_)] = __builtin_amdgcn_s_memtime();        \
+  } while (0)
+
+  // ---- prologue: stage the embedding row into xbuf, 3-way -------------
+  {
+    const long long tk = *a.tok;
+    const u32* erow = (const u32*)(a.embed + (u64)tk * a.hidden);
+    u32* x32 = (u32*)xbuf;
+    const int nd = a.hidden / 2;
+    const int b = nd * wave / 3, e = nd * (wave + 1) / 3;
+    float ss = 0.f;
+    for (int j = b + lane; j < e; j += LFMQ_WAVE) {
+      const u32 v = erow[j];
+      x32[j] = v;
+      ss = dot2_bf16(v, v, ss);
+    }
+    ss = wave_reduce_sum(ss);
+    if (lane == 0) ctl->ss_part[wave] = ss;
+    __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
+    ++gseq;
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x20, lane));
+    if (wave == 0) {
+      const float sst = ctl->ss_part[0] + ctl->ss_part[1] + ctl->ss_part[2];
+      ctl->inv_cur = rsqrtf(sst / (float)a.hidden + a.rms_eps);
+      lds_flag_store(&ctl->phase, ++phase_want);
+    } else {
+      ++phase_want;
+      ENG_CHECK(lds_wait_ge(&ctl->phase, phase_want, tmo, 0x21, lane));
+    }
+  }
+
+  for (int L = 0; L < a.n_layers; ++L) {
+    const unsigned e_qkv = 6 * L + 1, e_scr = 6 * L + 2, e_attn = 6 * L + 3,
+                   e_o = 6 * L + 4, e_h = 6 * L + 5, e_down = 6 * L + 6;
+    ENG_PROF(0);
+
+    // ---------------- P1: qkv records (norm-folded, scale inv_cur) -----
+    {
+      const float inv = ctl->inv_cur;
+      float v0 = 0.f;
+      for (int r = 0; r < npc_qkv; ++r) {
+        const u64 at = cursor_next(cur, rb_qkv, ring_bytes);
+        if ((r >> 1) % 3 != wave) continue;
+        ENG_CHECK(wait_fill(ctl, at + rb_qkv, tmo, 0x30, lane));
+        const float v = inv * record_dot(ring + (u32)(at % ring_bytes),
+                                         (const char*)xbuf, a.hidden, G1,
+                                         a.log2g, lane);
+        if (r & 1) {
+          if (lane == 0)
+            store_granule(g_qkv + (cu * npc_qkv + r) / 2, e_qkv,
+                          pack_bf16x2(v0, v));
+        } else {
+          v0 = v;
+        }
+        lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
+      }
+      lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
+    }
+    ENG_PROF(1);
+
+    // ---------------- P2: gather qkv into union ------------------------
+    if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
+    ENG_CHECK(gather_share(g_qkv, a.nqkv / 2, e_qkv, (u32*)unionb, wave, ctl,
+                           tmo, 0x31, lane));
+    ++gseq;
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x32, lane));
+    if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
+    ENG_PROF(2);
+
+    // ---------------- P3: attention units ------------------------------
+    {
+      const unsigned short* qkv16 = (const unsigned short*)unionb;
+      const u32* qkv32 = (const u32*)unionb;
+      unsigned short* qs = q_scr + wave * D;
+      unsigned short* ks = k_scr + wave * D;
+      float* ps = p_scr + wave * AT;
+      const int total_units = a.hq * nch;
+      for (int u = cu + NCU * wave; u < total_units; u += NCU * 3) {
+        const int h = u / nch, c = u % nch;
+        const int hk = h / rep;
+        const unsigned short* kcache =
+            a.kc + (u64)L * a.kv_layer_stride + (u64)hk * a.tmax * D;
+        const unsigned short* vcache =
+            a.vc + (u64)L * a.kv_layer_stride + (u64)hk * a.tmax * D;
+        // rope q (head h) and k (head hk); lane i holds dims i and i+D/2
+        const int i = lane;
+        float sv, cv;
+        __sincosf((float)pos * a.inv_freq[i], &sv, &cv);
+        const float q1 = bf16_bits_to_f32(qkv16[h * D + i]);
+        const float q2 = bf16_bits_to_f32(qkv16[h * D + i + D / 2]);
+        const float k1 = bf16_bits_to_f32(qkv16[(a.hq + hk) * D + i]);
+        const float k2 = bf16_bits_to_f32(qkv16[(a.hq + hk) * D + i + D / 2]);
+        const float qlo = q1 * cv - q2 * sv, qhi = q1 * sv + q2 * cv;
+        const float klo = k1 * cv - k2 * sv, khi = k1 * sv + k2 * cv;
+        qs[i] = (unsigned short)f32_to_bf16_bits_rne(qlo);
+        qs[i + D / 2] = (unsigned short)f32_to_bf16_bits_rne(qhi);
+        ks[i] = (unsigned short)f32_to_bf16_bits_rne(klo);
+        ks[i + D / 2] = (unsigned short)f32_to_bf16_bits_rne(khi);
+        const float s_last = wave_reduce_sum(qlo * klo + qhi * khi);
+        // cache append by the designated unit (plain stores: nothing reads
+        // row `pos` from the cache until the next kernel launch)
+        if (h % rep == 0 && c == pos / AT) {
+          u32* kdst = (u32*)(kcache + (u64)pos * D);
+          u32* vdst = (u32*)(vcache + (u64)pos * D);
+          kdst[lane] = (u32)ks[2 * lane] | ((u32)ks[2 * lane + 1] << 16);
+          vdst[lane] = qkv32[((u64)a.hq + a.hkv + hk) * (D / 2) + lane];
+        }
+        // scores: 4 tokens per lane (t = c*AT + j*64 + lane)
+        const u32* q32 = (const u32*)qs;
+        float sc[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int t = c * AT + j * LFMQ_WAVE + lane;
+          float s = -INFINITY;
+          if (t < T) {
+            if (t == pos) {
+              s = s_last * a.attn_scale;
+            } else {
+              const u32* krow = (const u32*)(kcache + (u64)t * D);
+              float acc = 0.f;
+#pragma unroll
+              for (int q = 0; q < D / 8; ++q) {
+                const u32x4 kv4 = *(const u32x4*)(krow + 4 * q);
+                acc = dot2_bf16(kv4[0], q32[4 * q + 0], acc);
+                acc = dot2_bf16(kv4[1], q32[4 * q + 1], acc);
+                acc = dot2_bf16(kv4[2], q32[4 * q + 2], acc);
+                acc = dot2_bf16(kv4[3], q32[4 * q + 3], acc);
+              }
+              s = acc * a.attn_scale;
+            }
+          }
+          sc[j] = s;
+        }
+        float m = fmaxf(fmaxf(sc[0], sc[1]), fmaxf(sc[2], sc[3]));
+        m = wave_reduce_max(m);
+        float l = 0.f;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const float p = sc[j] == -INFINITY ? 0.f : __expf(sc[j] - m);
+          ps[j * LFMQ_WAVE + lane] = p;
+          l += p;
+        }
+        l = wave_reduce_sum(l);
+        // PV: lane = output dword (dims 2*lane, 2*lane+1)
+        float o0 = 0.f, o1 = 0.f;
+        const int nt = min(AT, T - c * AT);
+        for (int tt = 0; tt < nt; ++tt) {
+          const int t = c * AT + tt;
+          const u32 vd = (t == pos)
+                             ? qkv32[((u64)a.hq + a.hkv + hk) * (D / 2) + lane]
+                             : ((const u32*)(vcache + (u64)t * D))[lane];
+          float v0, v1;
+          bf16x2_to_f32(vd, v0, v1);
+          const float p = ps[tt];
+          o0 = fmaf(p, v0, o0);
+          o1 = fmaf(p, v1, o1);
+        }
+        if (nch == 1) {  // direct output, no combine
+          const float invl = 1.f / l;
+          store_granule(g_attn + h * (D / 2) + lane, e_attn,
+                        pack_bf16x2(o0 * invl, o1 * invl));
+        } else {  // publish fp32 partial {o[D], m, l} to scratch
+          gu64* base = g_scr + ((u64)h * a.nch_max + c) * (D + 2);
+          store_granule(base + 2 * lane, e_scr, __builtin_bit_cast(u32, o0));
+          store_granule(base + 2 * lane + 1, e_scr,
+                        __builtin_bit_cast(u32, o1));
+          if (lane == 0) {
+            store_granule(base + D, e_scr, __builtin_bit_cast(u32, m));
+            store_granule(base + D + 1, e_scr, __builtin_bit_cast(u32, l));
+          }
+        }
+      }
+      // combines: this wave merges every head whose chunk-0 unit it ran
+      if (nch > 1) {
+        for (int u = cu + NCU * wave; u < total_units; u += NCU * 3) {
+          if (u % nch != 0) continue;
+          const int h = u / nch;
+          gu64* hb = g_scr + (u64)h * a.nch_max * (D + 2);
+          float mc = -INFINITY, lc = 0.f;
+          if (lane < nch) {
+            mc = __builtin_bit_cast(float, gran_wait(hb + lane * (D + 2) + D,
+                                                     e_scr, tmo, 0x40, lane,
+                                                     &ok));
+            lc = __builtin_bit_cast(float, gran_wait(
+                     hb + lane * (D + 2) + D + 1, e_scr, tmo, 0x41, lane,
+                     &ok));
+          }
+          ENG_CHECK(ok);
+          const float M = wave_reduce_max(mc);
+          const float wgt = lane < nch ? __expf(mc - M) : 0.f;
+          const float Lsum = wave_reduce_sum(wgt * lc);
+          float a0 = 0.f, a1 = 0.f;
+          for (int c = 0; c < nch; ++c) {
+            const float wc = __shfl(wgt, c, 64);
+            gu64* cb = hb + c * (D + 2);
+            const float p0 = __builtin_bit_cast(
+                float, gran_wait(cb + 2 * lane, e_scr, tmo, 0x42, lane, &ok));
+            const float p1 = __builtin_bit_cast(
+                float,
+                gran_wait(cb + 2 * lane + 1, e_scr, tmo, 0x43, lane, &ok));
+            a0 = fmaf(wc, p0, a0);
+            a1 = fmaf(wc, p1, a1);
+          }
+          ENG_CHECK(ok);
+          const float invL = 1.f / Lsum;
+          store_granule(g_attn + h * (D / 2) + lane, e_attn,
+                        pack_bf16x2(a0 * invL, a1 * invL));
+        }
+      }
+    }
+    ENG_PROF(3);
+
+    // ---------------- P4: gather attention output ----------------------
+    if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
+    ENG_CHECK(gather_share(g_attn, a.hq * D / 2, e_attn, (u32*)unionb, wave,
+                           ctl, tmo, 0x50, lane));
+    ++gseq;
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x51, lane));
+    if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
+    ENG_PROF(4);
+
+    // ---------------- P5: o records (src = attnbuf) --------------------
+    {
+      float v0 = 0.f;
+      for (int r = 0; r < npc_o; ++r) {
+        const u64 at = cursor_next(cur, rb_o, ring_bytes);
+        if ((r >> 1) % 3 != wave) continue;
+        ENG_CHECK(wait_fill(ctl, at + rb_o, tmo, 0x52, lane));
+        const float v = record_dot(ring + (u32)(at % ring_bytes), unionb,
+                                   a.hq * D, Go, a.log2g, lane);
+        if (r & 1) {
+          if (lane == 0)
+            store_granule(g_o + (cu * npc_o + r) / 2, e_o,
+                          pack_bf16x2(v0, v));
+        } else {
+          v0 = v;
+        }
+      }
+      lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
+    }
+
+    // ---------------- P6: gather o, residual-add into xbuf, inv2 -------
+    if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
+    ENG_CHECK(gather_share(g_o, a.hidden / 2, e_o, (u32*)ascr, wave, ctl, tmo,
+                           0x53, lane));
+    ++gseq;
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x54, lane));
+    {
+      u32* x32 = (u32*)xbuf;
+      const u32* add32 = (const u32*)ascr;
+      const int nd = a.hidden / 2;
+      const int b = nd * wave / 3, e = nd * (wave + 1) / 3;
+      float ss = 0.f;
+      for (int j = b + lane; j < e; j += LFMQ_WAVE) {
+        float x0, x1, a0, a1;
+        bf16x2_to_f32(x32[j], x0, x1);
+        bf16x2_to_f32(add32[j], a0, a1);
+        x0 += a0;
+        x1 += a1;
+        const u32 nv = pack_bf16x2(x0, x1);
+        x32[j] = nv;
+        ss = dot2_bf16(nv, nv, ss);
+      }
+      ss = wave_reduce_sum(ss);
+      if (lane == 0) ctl->ss_part[wave] = ss;
+      __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
+      ++gseq;
+      ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x55, lane));
+      if (wave == 0) {
+        lds_flag_store_rlx(&ctl->busy, 0u);
+        const float sst = ctl->ss_part[0] + ctl->ss_part[1] + ctl->ss_part[2];
+        ctl->inv_cur = rsqrtf(sst / (float)a.hidden + a.rms_eps);
+        lds_flag_store(&ctl->phase, ++phase_want);
+      } else {
+        ++phase_want;
+        ENG_CHECK(lds_wait_ge(&ctl->phase, phase_want, tmo, 0x56, lane));
+      }
+    }
+    ENG_PROF(5);
+
+    // ---------------- P7: gate_up records (scale inv2, silu pairs) -----
+    {
+      const float inv = ctl->inv_cur;
+      float vg = 0.f, h0 = 0.f;
+      for (int r = 0; r < npc_gu; ++r) {
+        const u64 at = cursor_next(cur, rb_gu, ring_bytes);
+        if ((r >> 2) % 3 != wave) continue;
+        ENG_CHECK(wait_fill(ctl, at + rb_gu, tmo, 0x60, lane));
+        const float v = inv * record_dot(ring + (u32)(at % ring_bytes),
+                                         (const char*)xbuf, a.hidden, G1,
+                                         a.log2g, lane);
+        switch (r & 3) {
+          case 0: vg = v; break;
+          case 1: h0 = (vg / (1.f + __expf(-vg))) * v; break;
+          case 2: vg = v; break;
+          default: {
+            const float h1 = (vg / (1.f + __expf(-vg))) * v;
+            if (lane == 0)
+              store_granule(g_h + (cu * (npc_gu / 2) + (r >> 1)) / 2, e_h,
+                            pack_bf16x2(h0, h1));
+          }
+        }
+        lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
+      }
+      lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
+    }
+    ENG_PROF(6);
+
+    // ---------------- P8: gather h ------------------------------------
+    if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
+    ENG_CHECK(gather_share(g_h, a.inter / 2, e_h, (u32*)unionb, wave, ctl,
+                           tmo, 0x61, lane));
+    ++gseq;
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x62, lane));
+    if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
+
+    // ---------------- P9: down records (src = hbuf) --------------------
+    {
+      float v0 = 0.f;
+      for (int r = 0; r < npc_dn; ++r) {
+        const u64 at = cursor_next(cur, rb_dn, ring_bytes);
+        if ((r >> 1) % 3 != wave) continue;
+        ENG_CHECK(wait_fill(ctl, at + rb_dn, tmo, 0x63, lane));
+        const float v = record_dot(ring + (u32)(at % ring_bytes), unionb,
+                                   a.inter, G2, a.log2g, lane);
+        if (r & 1) {
+          if (lane == 0)
+            store_granule(g_down + (cu * npc_dn + r) / 2, e_down,
+                          pack_bf16x2(v0, v));
+        } else {
+          v0 = v;
+        }
+        lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
+      }
+      lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
+    }
+    ENG_PROF(7);
+
+    // ---------------- P10: gather down, resid-add, inv1 of next layer --
+    if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
+    ENG_CHECK(gather_share(g_down, a.hidden / 2, e_down, (u32*)ascr, wave,
+                           ctl, tmo, 0x64, lane));
+    ++gseq;
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x65, lane));
+    {
+      u32* x32 = (u32*)xbuf;
+      const u32* add32 = (const u32*)ascr;
+      const int nd = a.hidden / 2;
+      const int b = nd * wave / 3, e = nd * (wave + 1) / 3;
+      float ss = 0.f;
+      for (int j = b + lane; j < e; j += LFMQ_WAVE) {
+        float x0, x1, a0, a1;
+        bf16x2_to_f32(x32[j], x0, x1);
+        bf16x2_to_f32(add32[j], a0, a1);
+        x0 += a0;
+        x1 += a1;
+        const u32 nv = pack_bf16x2(x0, x1);
+        x32[j] = nv;
+        ss = dot2_bf16(nv, nv, ss);
+      }
+      ss = wave_reduce_sum(ss);
+      if (lane == 0) ctl->ss_part[wave] = ss;
+      __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
+      ++gseq;
+      ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x66, lane));
+      if (wave == 0) {
+        lds_flag_store_rlx(&ctl->busy, 0u);
+        const float sst = ctl->ss_part[0] + ctl->ss_part[1] + ctl->ss_part[2];
+        ctl->inv_cur = rsqrtf(sst / (float)a.hidden + a.rms_eps);
+        lds_flag_store(&ctl->phase, ++phase_want);
+      } else {
+        ++phase_want;
+        ENG_CHECK(lds_wait_ge(&ctl->phase, phase_want, tmo, 0x67, lane));
+      }
+    }
+  }
+
+  // ---- epilogue: CU 0 wave 0 writes the final-norm output -------------
+  if (cu == 0 && wave == 0) {
+    const float inv = ctl->inv_cur;
+    const u32* x32 = (const u32*)xbuf;
+    const u32* w32 = (const u32*)a.final_w;
+    u32* out32 = (u32*)a.xout;
+    for (int j = lane; j < a.hidden / 2; j += LFMQ_WAVE) {
+      float x0, x1, w0, w1;
+      bf16x2_to_f32(x32[j], x0, x1);
+      bf16x2_to_f32(w32[j], w0, w1);
+      out32[j] = pack_bf16x2(x0 * inv * w0, x1 * inv * w1);
+    }
+  }
+#undef ENG_CHECK
+#undef ENG_PROF
+}
+
+}  // namespace
+
+// host-side launcher; returns required dynamic-LDS bytes via *lds_bytes
+// when kernel launch is skipped (query mode: stream == nullptr).
+extern "C" hipError_t lfmq_engine_step(const EngineArgs* args, int d,
+                                       hipStream_t stream) {
+  if (d != 128) return hipErrorInvalidValue;
+  const u32 ring_bytes = (u32)args->nslot_ring * SLOT;
+  const int union_bytes =
+      std::max(args->inter * 2, std::max(args->nqkv * 2, args->hq * d * 2));
+  const int ascr_bytes = std::max(3 * d * 4 + 3 * AT * 4, args->hidden * 2);
+  const size_t lds = ring_bytes + args->hidden * 2 + union_bytes +
+                     ((ascr_bytes + 63) & ~63) + sizeof(Ctl) + 64;
+  static size_t lds_set = 0;
+  if (lds > lds_set) {
+    hipError_t e = hipFuncSetAttribute(
+        (const void*)lfmq_engine_kernel<128>,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    if (e != hipSuccess) return e;
+    lds_set = lds;
+  }
+  hipLaunchKernelGGL(lfmq_engine_kernel<128>, dim3(NCU), dim3(256), lds,
+                     stream, *args);
+  return hipGetLastError();
+}

@@ -49,6 +49,7 @@ hipError_t lfmq_argmax_rows(const void*, void*, void*, void*, int, int,
                             hipStream_t);
 hipError_t lfmq_read_probe(const void*, void*, long, long, int, int, int,
                            hipStream_t);
+#include "engine_args.h"
 
 namespace {
 
@@ -399,6 +400,69 @@ torch::Tensor attn_decode_fused(torch::Tensor qkv, torch::Tensor kc,
   return o;
 }
 
+// persistent decode engine: one launch = one whole decode step (layers
+// only; lm_head/sampling stay separate).  Granule buffers must be zeroed
+// by the CALLER before every call (lfmq/models/engine.py does, inside the
+// captured graph).  Returns nothing; output lands in xout.
+void engine_step(torch::Tensor stream_t, long stream_per_cu, long nslot_ring,
+                 long n_layers, long hidden, long inter, long nqkv, long hq,
+                 long hkv, long log2g, double rms_eps, double attn_scale,
+                 torch::Tensor inv_freq, torch::Tensor kc, torch::Tensor vc,
+                 torch::Tensor state, torch::Tensor tok, torch::Tensor embed,
+                 torch::Tensor final_w, torch::Tensor xout,
+                 torch::Tensor g_qkv, torch::Tensor g_attn, torch::Tensor g_o,
+                 torch::Tensor g_h, torch::Tensor g_down, torch::Tensor g_scr,
+                 torch::Tensor tmo, c10::optional<torch::Tensor> prof) {
+  check_t(kc, torch::kBFloat16, "k_cache");
+  check_t(vc, torch::kBFloat16, "v_cache");
+  check_t(embed, torch::kBFloat16, "embed");
+  check_t(final_w, torch::kBFloat16, "final_w");
+  check_t(xout, torch::kBFloat16, "xout");
+  check_t(inv_freq, torch::kFloat32, "inv_freq");
+  check_t(state, torch::kInt32, "state");
+  TORCH_CHECK(tok.scalar_type() == torch::kInt64, "tok must be int64");
+  TORCH_CHECK(stream_t.scalar_type() == torch::kUInt8, "stream uint8");
+  // kc: [L, 1, hkv, tmax, d] (stacked) or list-stacked contiguous
+  const int tmax = (int)kc.size(-2), d = (int)kc.size(-1);
+  EngineArgs a{};
+  a.stream = (const char*)stream_t.data_ptr();
+  a.stream_per_cu = (unsigned long long)stream_per_cu;
+  a.n_slots_total = (int)((stream_per_cu + 16383) / 16384);
+  a.nslot_ring = (int)nslot_ring;
+  a.n_layers = (int)n_layers;
+  a.hidden = (int)hidden;
+  a.inter = (int)inter;
+  a.nqkv = (int)nqkv;
+  a.hq = (int)hq;
+  a.hkv = (int)hkv;
+  a.log2g = (int)log2g;
+  a.rms_eps = (float)rms_eps;
+  a.attn_scale = (float)attn_scale;
+  a.inv_freq = (const float*)inv_freq.data_ptr();
+  a.kc = (const unsigned short*)kc.data_ptr();
+  a.vc = (const unsigned short*)vc.data_ptr();
+  a.kv_layer_stride = (unsigned long long)hkv * tmax * d;
+  a.tmax = tmax;
+  a.nch_max = (tmax + 255) / 256;
+  TORCH_CHECK(a.nch_max <= 64, "engine needs tmax <= 16384");
+  a.T_dev = (const int*)state.data_ptr() + 1;  // state = {pos0, T, step}
+  a.tok = (const long long*)tok.data_ptr();
+  a.embed = (const unsigned short*)embed.data_ptr();
+  a.final_w = (const unsigned short*)final_w.data_ptr();
+  a.xout = (unsigned short*)xout.data_ptr();
+  a.g_qkv = g_qkv.data_ptr();
+  a.g_attn = g_attn.data_ptr();
+  a.g_o = g_o.data_ptr();
+  a.g_h = g_h.data_ptr();
+  a.g_down = g_down.data_ptr();
+  a.g_scr = g_scr.data_ptr();
+  a.tmo = tmo.data_ptr();
+  a.prof = prof.has_value()
+               ? (unsigned long long*)prof->data_ptr()
+               : nullptr;
+  CHECK_HIP(lfmq_engine_step(&a, d, stream()));
+}
+
 void hessian_accum(torch::Tensor h, torch::Tensor x) {
   check_t(h, torch::kFloat32, "H");
   check_t(x, torch::kBFloat16, "x");
@@ -433,4 +497,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused rope+KV-append+attention decode");
   m.def("grouped_dq_gateup", &grouped_dq_gateup, "MoE grouped gate_up+silu");
   m.def("grouped_dq_linear", &grouped_dq_linear, "MoE grouped linear");
+  m.def("engine_step", &engine_step, "persistent decode engine step");
 }

@@ -89,4 +89,7 @@ class GraphDecoder:
             self.graph.replay()
         self.step_count += n_tokens
         self.cache.seq_len += n_tokens
+        eng = getattr(self.model, "_engine", None)
+        if eng is not None:
+            eng.check_health()  # surfaces engine spin give-up codes
         return self.ring[:, start:start + n_tokens].clone()

@@ -307,6 +307,21 @@ class LlamaForCausalLM(nn.Module):
         cache.advance(s)
         return logits
 
+    def _engine_for(self, cache, batch: int) -> bool:
+        """Build/bind the persistent decode engine for this cache (lazy;
+        one stream build per cache).  False => launched-kernel path."""
+        eng = getattr(self, "_engine", None)
+        if eng is not None and eng.cache is cache:
+            return True
+        if getattr(self, "_engine_unsupported", False):
+            return False
+        from .engine import DecodeEngine, engine_supported
+        if not engine_supported(self, batch):
+            self._engine_unsupported = True
+            return False
+        self._engine = DecodeEngine(self, cache, cache.max_seq)
+        return True
+
     def _decode_fused_ready(self) -> bool:
         from .qlinear import QLinear
         for layer in self.layers:
@@ -319,6 +334,11 @@ class LlamaForCausalLM(nn.Module):
 
     def forward_decode_step(self, input_ids, cache, state, max_seq):
         """One hipGraph-capturable decode step: [b,1] ids -> [b,1,vocab]."""
+        # persistent weight-streaming engine: the whole layer stack in ONE
+        # kernel launch (csrc/engine.hip); lm_head stays a separate launch.
+        if input_ids.is_cuda and self._engine_for(cache, input_ids.shape[0]):
+            x = self._engine.step(input_ids, state)
+            return self.lm_head(x)
         x = self.embed_tokens(input_ids)
         if getattr(self, "_fused_decode", None) is None:
             self._fused_decode = self._decode_fused_ready()

@@ -35,6 +35,7 @@ SOURCES = [
     "csrc/sample.hip",
     "csrc/probe.hip",
     "csrc/grouped_dq_gemm.hip",
+    "csrc/engine.hip",
 ]
 
 setup(

@@ -1,0 +1,101 @@
+"""Persistent-engine weight-stream builder: layout invariants on CPU.
+
+The kernel (csrc/engine.hip) walks records with the documented rule; this
+re-derives the walk independently from the spec in lfmq/models/engine.py's
+docstring and checks the built bytes.  GPU execution parity lives in
+tests/test_gpu_engine.py.
+"""
+import torch
+
+import lfmq
+from lfmq.models.llama import LlamaConfig
+from lfmq.models.engine import (DecodeEngine, engine_supported, _rec_bytes,
+                                SLOT, NCU)
+
+
+def _engine_model():
+    # smallest config the engine supports: d=128, all dims % 512 == 0
+    cfg = LlamaConfig(vocab_size=1024, hidden_size=512, intermediate_size=512,
+                      n_layers=2, n_heads=4, n_kv_heads=2, head_dim=128,
+                      max_seq=512)
+    m = lfmq.build_model(cfg, seed=3)
+    q = lfmq.quantize(m, lfmq.QuantConfig(scheme="rtn", bits=4, group_size=128))
+    return q
+
+
+def test_engine_supported_gates():
+    q = _engine_model()
+    assert engine_supported(q, batch=1)
+    assert not engine_supported(q, batch=2)
+    # unsupported: group too small for the lane-chunk layout
+    cfg = LlamaConfig(vocab_size=256, hidden_size=512, intermediate_size=512,
+                      n_layers=1, n_heads=4, n_kv_heads=2, head_dim=128,
+                      max_seq=256)
+    q2 = lfmq.quantize(lfmq.build_model(cfg, seed=1),
+                       lfmq.QuantConfig(scheme="rtn", bits=4, group_size=16))
+    assert not engine_supported(q2, batch=1)
+    # head_dim != 128 unsupported
+    cfg3 = LlamaConfig.tiny()
+    q3 = lfmq.quantize(lfmq.build_model(cfg3, seed=1),
+                       lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32))
+    assert not engine_supported(q3, batch=1)
+
+
+def test_stream_layout_roundtrip():
+    q = _engine_model()
+    cache = q.make_cache(1, 256)
+    eng = DecodeEngine(q, cache, 256)
+    cfg = q.cfg
+    g = eng.g
+    ring = eng.nslot * SLOT
+    stream = eng.stream.reshape(NCU, eng.per_cu)
+
+    assert eng.per_cu % SLOT == 0
+
+    # independent offset walk per the documented spec
+    def walk():
+        cur = 0
+        for layer in q.layers:
+            for op, inter in ((layer.self_attn.qkv_proj, False),
+                              (layer.self_attn.o_proj, False),
+                              (layer.mlp.gate_up_proj, True),
+                              (layer.mlp.down_proj, False)):
+                n, k = op.out_features, op.in_features
+                G = k // g
+                rb = _rec_bytes(k, G)
+                npc = n // NCU
+                for r in range(npc):
+                    if cur % ring + rb > ring:
+                        cur += ring - cur % ring
+                    yield op, inter, r, cur, rb, k, G
+                    cur += rb
+
+    for op, inter_leaved, r, off, rb, k, G in walk():
+        # never straddles the ring wrap
+        assert off % ring + rb <= ring
+        assert off % 16 == 0
+        # spot-check CU 0 and CU 17
+        for cu in (0, 17):
+            if inter_leaved:
+                m = op.out_features // 2 // NCU
+                row = (cu * m + r // 2) + (op.out_features // 2) * (r % 2)
+            else:
+                row = cu * (op.out_features // NCU) + r
+            got = stream[cu, off:off + rb]
+            qw = op.qweight.view(torch.uint8).reshape(op.out_features, k // 2)
+            sc = op.scales.view(torch.uint8).reshape(op.out_features, 2 * G)
+            zz = op.qzeros.view(torch.uint8).reshape(op.out_features, -1)
+            assert torch.equal(got[:k // 2], qw[row])
+            assert torch.equal(got[k // 2:k // 2 + 2 * G], sc[row])
+            assert torch.equal(got[k // 2 + 2 * G:k // 2 + 2 * G + (G + 1) // 2],
+                               zz[row][:(G + 1) // 2])
+
+
+def test_kvcache_stacked_views():
+    q = _engine_model()
+    cache = q.make_cache(1, 64)
+    # per-layer views alias the stacked tensor (static addresses for graphs
+    # and a single base pointer + stride for the engine)
+    cache.k[1][0, 0, 3, 5] = 7.0
+    assert float(cache.k_all[1, 0, 0, 3, 5]) == 7.0
+    assert cache.k[0].is_contiguous()

@@ -1,0 +1,108 @@
+"""GPU parity tests for the persistent decode engine (csrc/engine.hip).
+
+Engine path vs the launched-kernel path on identical state: full-width
+(8B-shaped) layers so every record/attention geometry matches production.
+Every test is bit-level-independent (fresh caches per path).
+"""
+import os
+
+import pytest
+import torch
+
+import lfmq
+from lfmq.models.llama import LlamaConfig
+
+pytestmark = pytest.mark.gpu
+
+needs_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
+
+
+def _model8b2(n_layers=2, vocab=1024, max_seq=1024):
+    cfg = LlamaConfig(vocab_size=vocab, hidden_size=4096,
+                      intermediate_size=14336, n_layers=n_layers, n_heads=32,
+                      n_kv_heads=8, head_dim=128, max_seq=max_seq)
+    m = lfmq.build_model(cfg).to_empty(device="cuda")
+    torch.manual_seed(7)
+    for p in m.parameters():
+        p.data.normal_(0.0, 0.02)
+    q = lfmq.quantize(m, lfmq.QuantConfig(scheme="rtn", bits=4,
+                                          group_size=128))
+    return q, cfg
+
+
+def _decode_steps(model, cfg, ids, nsteps, max_seq, use_engine):
+    """Prefill eagerly, then nsteps of forward_decode_step; returns
+    stacked logits [nsteps, vocab]."""
+    model._engine = None
+    model._engine_unsupported = False
+    os.environ["LFMQ_ENGINE"] = "1" if use_engine else "0"
+    try:
+        cache = model.make_cache(1, max_seq=max_seq, device="cuda")
+        logits = model(ids, cache)
+        tok = logits[:, -1:].argmax(dim=-1)
+        state = torch.tensor([ids.shape[1], ids.shape[1] + 1, 0],
+                             dtype=torch.int32, device="cuda")
+        outs = []
+        for _ in range(nsteps):
+            lg = model.forward_decode_step(tok, cache, state, max_seq)
+            outs.append(lg[:, -1].float().clone())
+            tok = lg[:, -1:].argmax(dim=-1)
+            state += torch.tensor([1, 1, 1], dtype=torch.int32,
+                                  device="cuda")
+            cache.seq_len += 1
+        if use_engine:
+            assert model._engine is not None, "engine did not activate"
+            model._engine.check_health()
+        else:
+            assert getattr(model, "_engine", None) is None
+        return torch.cat(outs)
+    finally:
+        os.environ.pop("LFMQ_ENGINE", None)
+        model._engine = None
+        model._engine_unsupported = False
+
+
+@needs_gpu
+def test_engine_parity_short_T():
+    """T < 256 the whole run: the no-combine (nch==1) attention path."""
+    q, cfg = _model8b2()
+    ids = torch.randint(0, cfg.vocab_size, (1, 64), device="cuda")
+    ref = _decode_steps(q, cfg, ids, 8, 512, use_engine=False)
+    got = _decode_steps(q, cfg, ids, 8, 512, use_engine=True)
+    rel = (got - ref).abs().max() / ref.abs().max()
+    assert float(rel) < 2e-2, f"engine parity rel={float(rel)}"
+
+
+@needs_gpu
+def test_engine_parity_across_combine_boundary():
+    """Prefill 250 + 12 steps: T crosses 256 so the scratch-partial +
+    in-engine combine path runs (nch 1 -> 2)."""
+    q, cfg = _model8b2()
+    ids = torch.randint(0, cfg.vocab_size, (1, 250), device="cuda")
+    ref = _decode_steps(q, cfg, ids, 12, 512, use_engine=False)
+    got = _decode_steps(q, cfg, ids, 12, 512, use_engine=True)
+    rel = (got - ref).abs().max() / ref.abs().max()
+    assert float(rel) < 2e-2, f"combine parity rel={float(rel)}"
+
+
+@needs_gpu
+def test_engine_parity_long_T():
+    """T ~ 600: 3 chunks per head, uneven unit counts across CUs."""
+    q, cfg = _model8b2(max_seq=1024)
+    ids = torch.randint(0, cfg.vocab_size, (1, 590), device="cuda")
+    ref = _decode_steps(q, cfg, ids, 4, 1024, use_engine=False)
+    got = _decode_steps(q, cfg, ids, 4, 1024, use_engine=True)
+    rel = (got - ref).abs().max() / ref.abs().max()
+    assert float(rel) < 2e-2, f"long-T parity rel={float(rel)}"
+
+
+@needs_gpu
+def test_engine_greedy_tokens_match():
+    """Token-level agreement over a longer greedy run (argmax is the
+    bench-visible output; logit noise must not flip it often)."""
+    q, cfg = _model8b2()
+    ids = torch.randint(0, cfg.vocab_size, (1, 32), device="cuda")
+    ref = _decode_steps(q, cfg, ids, 24, 512, use_engine=False)
+    got = _decode_steps(q, cfg, ids, 24, 512, use_engine=True)
+    agree = (ref.argmax(-1) == got.argmax(-1)).float().mean()
+    assert float(agree) >= 0.9, f"greedy agreement {float(agree)}"
