@@ -97,34 +97,55 @@ __device__ __forceinline__ u64 cursor_next(Cursor& c, int rec, u32 ring_bytes) {
 // nibble dword at rec + (i*64+lane)*4 is exactly qweight dword (k0/8), x is
 // one conflict-free ds_read_b128 (stride 16 B across lanes).  Math matches
 // csrc/dq_gemv.hip: s*( sum (128+c)*x ) - s*(z+128)*sum x, fp32 accumulate.
+// Chunked two-phase pipeline: issue a whole chunk's LDS loads up front
+// (independent), then compute — the round-1 GEMV lesson (csrc/dq_gemv.hip
+// KI-unroll): per-iteration loads make the compiler reuse registers and
+// serialize the loop on LDS latency.  CHUNK=8 covers k=4096 ops in one
+// chunk; k=14336 (down) runs 3.5 chunks.
 __device__ __forceinline__ float record_dot(const char* lds_rec,
                                             const char* lds_src, int k, int G,
                                             int log2g, int lane) {
+  constexpr int CH = 8;
   const int iters = k >> 9;
   const u32* wbase = (const u32*)lds_rec;
   const unsigned short* sbase = (const unsigned short*)(lds_rec + k / 2);
   const unsigned char* zbase = (const unsigned char*)(lds_rec + k / 2 + G * 2);
   float acc = 0.f;
-  for (int i = 0; i < iters; ++i) {
-    const int k0 = i * 512 + lane * 8;
-    const u32 w = wbase[i * 64 + lane];
-    const u32x4 xq = *(const u32x4*)(lds_src + k0 * 2);
-    const int gi = k0 >> log2g;
-    const float s = bf16_bits_to_f32(sbase[gi]);
-    const u32 zb = zbase[gi >> 1];
-    const u32 z = (gi & 1) ? (zb >> 4) : (zb & 0xFu);
-    const DequantPairs4 dq = dequant4_pairs(w);
-    float cx = 0.f, xs = 0.f;
-    cx = dot2_bf16(dq.p01, xq[0], cx);
-    cx = dot2_bf16(dq.p23, xq[1], cx);
-    cx = dot2_bf16(dq.p45, xq[2], cx);
-    cx = dot2_bf16(dq.p67, xq[3], cx);
-    xs = dot2_bf16(ONE2_BF16, xq[0], xs);
-    xs = dot2_bf16(ONE2_BF16, xq[1], xs);
-    xs = dot2_bf16(ONE2_BF16, xq[2], xs);
-    xs = dot2_bf16(ONE2_BF16, xq[3], xs);
-    acc = fmaf(s, cx, acc);
-    acc = fmaf(-s * (float)(z + 128u), xs, acc);
+  for (int c0 = 0; c0 < iters; c0 += CH) {
+    const int n = min(CH, iters - c0);
+    u32 w[CH];
+    u32x4 xq[CH];
+    float s[CH], sz[CH];
+#pragma unroll
+    for (int j = 0; j < CH; ++j) {
+      if (j >= n) break;
+      const int i = c0 + j;
+      const int k0 = i * 512 + lane * 8;
+      w[j] = wbase[i * 64 + lane];
+      xq[j] = *(const u32x4*)(lds_src + k0 * 2);
+      const int gi = k0 >> log2g;
+      const u32 sraw = sbase[gi];
+      const u32 zb = zbase[gi >> 1];
+      const u32 z = (gi & 1) ? (zb >> 4) : (zb & 0xFu);
+      s[j] = bf16_bits_to_f32(sraw);
+      sz[j] = s[j] * (float)(z + 128u);
+    }
+#pragma unroll
+    for (int j = 0; j < CH; ++j) {
+      if (j >= n) break;
+      const DequantPairs4 dq = dequant4_pairs(w[j]);
+      float cx = 0.f, xs = 0.f;
+      cx = dot2_bf16(dq.p01, xq[j][0], cx);
+      cx = dot2_bf16(dq.p23, xq[j][1], cx);
+      cx = dot2_bf16(dq.p45, xq[j][2], cx);
+      cx = dot2_bf16(dq.p67, xq[j][3], cx);
+      xs = dot2_bf16(ONE2_BF16, xq[j][0], xs);
+      xs = dot2_bf16(ONE2_BF16, xq[j][1], xs);
+      xs = dot2_bf16(ONE2_BF16, xq[j][2], xs);
+      xs = dot2_bf16(ONE2_BF16, xq[j][3], xs);
+      acc = fmaf(s[j], cx, acc);
+      acc = fmaf(-sz[j], xs, acc);
+    }
   }
   return wave_reduce_sum(acc);
 }
@@ -213,6 +234,16 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     const int total = a.n_slots_total;
     for (int s = 0; s < total; ++s) {
       if (s >= a.nslot_ring) {
+        // about to stall for ring space: ladder-publish the issued slots
+        // as they land (otherwise up to LAG-1 landed slots stay invisible
+        // and consumers lock-step with the loader across every stall);
+        // progressive vmcnt steps hand the oldest slots over immediately
+        // instead of waiting for the full in-flight burst to land.
+        if (s >= 3) {
+          loader_publish<32>(f_off, (unsigned)(s - 2));
+          loader_publish<16>(f_off, (unsigned)(s - 1));
+        }
+        loader_publish<0>(f_off, (unsigned)s);
         unsigned spins = 0;
         for (;;) {
           unsigned d0, d1, d2;
@@ -407,20 +438,31 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
           l += p;
         }
         l = wave_reduce_sum(l);
-        // PV: lane = output dword (dims 2*lane, 2*lane+1)
-        float o0 = 0.f, o1 = 0.f;
+        // PV: lane = output dword (dims 2*lane, 2*lane+1).  4 independent
+        // partial pairs + unroll keep ~8 V-row loads in flight (a single
+        // accumulate chain serializes on global-load latency — the same
+        // structure as csrc/attn_decode.hip's PV pass).
+        float oa[4][2] = {{0.f, 0.f}, {0.f, 0.f}, {0.f, 0.f}, {0.f, 0.f}};
         const int nt = min(AT, T - c * AT);
-        for (int tt = 0; tt < nt; ++tt) {
-          const int t = c * AT + tt;
-          const u32 vd = (t == pos)
-                             ? qkv32[((u64)a.hq + a.hkv + hk) * (D / 2) + lane]
-                             : ((const u32*)(vcache + (u64)t * D))[lane];
-          float v0, v1;
-          bf16x2_to_f32(vd, v0, v1);
-          const float p = ps[tt];
-          o0 = fmaf(p, v0, o0);
-          o1 = fmaf(p, v1, o1);
+        const u32 v_last = qkv32[((u64)a.hq + a.hkv + hk) * (D / 2) + lane];
+#pragma unroll 2
+        for (int tb = 0; tb < nt; tb += 4) {
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            const int tt = tb + q;
+            if (tt >= nt) break;
+            const int t = c * AT + tt;
+            const u32 vd =
+                (t == pos) ? v_last : ((const u32*)(vcache + (u64)t * D))[lane];
+            float v0, v1;
+            bf16x2_to_f32(vd, v0, v1);
+            const float p = ps[tt];
+            oa[q][0] = fmaf(p, v0, oa[q][0]);
+            oa[q][1] = fmaf(p, v1, oa[q][1]);
+          }
         }
+        const float o0 = (oa[0][0] + oa[1][0]) + (oa[2][0] + oa[3][0]);
+        const float o1 = (oa[0][1] + oa[1][1]) + (oa[2][1] + oa[3][1]);
         if (nch == 1) {  // direct output, no combine
           const float invl = 1.f / l;
           store_granule(g_attn + h * (D / 2) + lane, e_attn,

@@ -30,23 +30,36 @@ def _model8b2(n_layers=2, vocab=1024, max_seq=1024):
     return q, cfg
 
 
-def _decode_steps(model, cfg, ids, nsteps, max_seq, use_engine):
-    """Prefill eagerly, then nsteps of forward_decode_step; returns
-    stacked logits [nsteps, vocab]."""
+
+
+def _per_step_rel(got, ref):
+    """max over steps of (max |diff| / max |ref|) per step — deep
+    random-init logits shrink with depth, so a global-max normalization
+    understates the reference scale of early steps."""
+    d = (got - ref).abs().amax(dim=-1)
+    r = ref.abs().amax(dim=-1)
+    return float((d / r).max())
+
+def _decode_steps(model, cfg, ids, nsteps, max_seq, use_engine, seed=11):
+    """Prefill eagerly, then nsteps of forward_decode_step with a FIXED
+    (teacher-forced) token sequence — random-init logits are nearly flat,
+    so free-running greedy would diverge on sub-tolerance noise and
+    compare different sequences.  Returns stacked logits [nsteps, vocab]."""
     model._engine = None
     model._engine_unsupported = False
     os.environ["LFMQ_ENGINE"] = "1" if use_engine else "0"
+    g = torch.Generator().manual_seed(seed)
+    forced = torch.randint(0, cfg.vocab_size, (nsteps,), generator=g)
     try:
         cache = model.make_cache(1, max_seq=max_seq, device="cuda")
-        logits = model(ids, cache)
-        tok = logits[:, -1:].argmax(dim=-1)
+        model(ids, cache)
         state = torch.tensor([ids.shape[1], ids.shape[1] + 1, 0],
                              dtype=torch.int32, device="cuda")
         outs = []
-        for _ in range(nsteps):
+        for i in range(nsteps):
+            tok = forced[i].reshape(1, 1).to("cuda")
             lg = model.forward_decode_step(tok, cache, state, max_seq)
             outs.append(lg[:, -1].float().clone())
-            tok = lg[:, -1:].argmax(dim=-1)
             state += torch.tensor([1, 1, 1], dtype=torch.int32,
                                   device="cuda")
             cache.seq_len += 1
@@ -69,8 +82,8 @@ def test_engine_parity_short_T():
     ids = torch.randint(0, cfg.vocab_size, (1, 64), device="cuda")
     ref = _decode_steps(q, cfg, ids, 8, 512, use_engine=False)
     got = _decode_steps(q, cfg, ids, 8, 512, use_engine=True)
-    rel = (got - ref).abs().max() / ref.abs().max()
-    assert float(rel) < 2e-2, f"engine parity rel={float(rel)}"
+    rel = _per_step_rel(got, ref)
+    assert rel < 2e-2, f"engine parity rel={rel}"
 
 
 @needs_gpu
@@ -81,8 +94,8 @@ def test_engine_parity_across_combine_boundary():
     ids = torch.randint(0, cfg.vocab_size, (1, 250), device="cuda")
     ref = _decode_steps(q, cfg, ids, 12, 512, use_engine=False)
     got = _decode_steps(q, cfg, ids, 12, 512, use_engine=True)
-    rel = (got - ref).abs().max() / ref.abs().max()
-    assert float(rel) < 2e-2, f"combine parity rel={float(rel)}"
+    rel = _per_step_rel(got, ref)
+    assert rel < 2e-2, f"combine parity rel={rel}"
 
 
 @needs_gpu
@@ -92,17 +105,17 @@ def test_engine_parity_long_T():
     ids = torch.randint(0, cfg.vocab_size, (1, 590), device="cuda")
     ref = _decode_steps(q, cfg, ids, 4, 1024, use_engine=False)
     got = _decode_steps(q, cfg, ids, 4, 1024, use_engine=True)
-    rel = (got - ref).abs().max() / ref.abs().max()
-    assert float(rel) < 2e-2, f"long-T parity rel={float(rel)}"
+    rel = _per_step_rel(got, ref)
+    assert rel < 2e-2, f"long-T parity rel={rel}"
 
 
 @needs_gpu
-def test_engine_greedy_tokens_match():
-    """Token-level agreement over a longer greedy run (argmax is the
-    bench-visible output; logit noise must not flip it often)."""
+def test_engine_parity_many_steps():
+    """24 teacher-forced steps: per-step error must stay bounded (no
+    drift or state corruption across repeated engine launches)."""
     q, cfg = _model8b2()
     ids = torch.randint(0, cfg.vocab_size, (1, 32), device="cuda")
     ref = _decode_steps(q, cfg, ids, 24, 512, use_engine=False)
     got = _decode_steps(q, cfg, ids, 24, 512, use_engine=True)
-    agree = (ref.argmax(-1) == got.argmax(-1)).float().mean()
-    assert float(agree) >= 0.9, f"greedy agreement {float(agree)}"
+    rel = _per_step_rel(got, ref)
+    assert rel < 2e-2, f"per-step drift rel={rel}"
