@@ -152,9 +152,10 @@ __device__ __forceinline__ float record_dot(const char* lds_rec,
 
 // wait until the ring holds stream bytes [0, end_off): filled*SLOT >= end
 __device__ __forceinline__ bool wait_fill(Ctl* ctl, u64 end_off, gu32* tmo,
-                                          unsigned code, int lane) {
+                                          unsigned code, int lane,
+                                          unsigned* acc = nullptr) {
   const unsigned need = (unsigned)((end_off + SLOT - 1) / SLOT);
-  return lds_wait_ge(&ctl->filled, need, tmo, code, lane);
+  return lds_wait_ge(&ctl->filled, need, tmo, code, lane, acc);
 }
 
 // ---- 3-way gather helpers --------------------------------------------------
@@ -163,16 +164,18 @@ __device__ __forceinline__ bool wait_fill(Ctl* ctl, u64 end_off, gu32* tmo,
 __device__ __forceinline__ bool gather_share(gu64* g, int cnt, unsigned epoch,
                                              u32* lds_dst, int wave, Ctl* ctl,
                                              gu32* tmo, unsigned code,
-                                             int lane) {
+                                             int lane,
+                                             unsigned* passes = nullptr) {
   const int b = cnt * wave / 3, e = cnt * (wave + 1) / 3;
-  if (!sweep_stage(g + b, e - b, epoch, lds_dst + b, tmo, code, lane))
+  if (!sweep_stage(g + b, e - b, epoch, lds_dst + b, tmo, code, lane, passes))
     return false;
   __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
   return true;
 }
 __device__ __forceinline__ bool wait_gath(Ctl* ctl, unsigned gseq, gu32* tmo,
-                                          unsigned code, int lane) {
-  return lds_wait_ge(&ctl->gath, 3 * gseq, tmo, code, lane);
+                                          unsigned code, int lane,
+                                          unsigned* acc = nullptr) {
+  return lds_wait_ge(&ctl->gath, 3 * gseq, tmo, code, lane, acc);
 }
 
 // single-granule bounded wait (combine path)
@@ -232,18 +235,31 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     const unsigned c2 = lds_off(&ctl->consumed[2]);
     const unsigned b_off = lds_off(&ctl->busy);
     const int total = a.n_slots_total;
+    unsigned last_pub = 0;
+    unsigned poll_spins_tot = 0;
+#define ENG_PUB(VML, val)                                  \
+  do {                                                     \
+    const unsigned v_ = (val);                             \
+    if (v_ > last_pub) {                                   \
+      loader_publish<VML>(f_off, v_);                      \
+      last_pub = v_;                                       \
+    }                                                      \
+  } while (0)
     for (int s = 0; s < total; ++s) {
       if (s >= a.nslot_ring) {
         // about to stall for ring space: ladder-publish the issued slots
         // as they land (otherwise up to LAG-1 landed slots stay invisible
         // and consumers lock-step with the loader across every stall);
-        // progressive vmcnt steps hand the oldest slots over immediately
-        // instead of waiting for the full in-flight burst to land.
+        // progressive vmcnt steps hand the oldest slots over immediately.
+        // Publishes are MONOTONIC (ENG_PUB): a later lagged in-loop
+        // publish must never overwrite a ladder publish with a smaller
+        // value — consumers poll `filled >= need` and a regressed value
+        // re-blocks them on slots that landed long ago.
         if (s >= 3) {
-          loader_publish<32>(f_off, (unsigned)(s - 2));
-          loader_publish<16>(f_off, (unsigned)(s - 1));
+          ENG_PUB(32, (unsigned)(s - 2));
+          ENG_PUB(16, (unsigned)(s - 1));
         }
-        loader_publish<0>(f_off, (unsigned)s);
+        ENG_PUB(0, (unsigned)s);
         unsigned spins = 0;
         for (;;) {
           unsigned d0, d1, d2;
@@ -251,21 +267,28 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
           if ((int)min(d0, min(d1, d2)) >= s - a.nslot_ring + 1) break;
           if (spin_fail(spins, tmo, 0x10, lane)) return;
         }
+        poll_spins_tot += spins;
       }
       char* lbase = ring + (s % a.nslot_ring) * SLOT;
       const char* g = wg_src + (u64)s * SLOT + lane * 16;
 #pragma unroll
       for (int i = 0; i < 16; ++i) glds_fill_1k(g + i * 1024, lbase + i * 1024);
+      // loader timeline (debug): issue timestamp per slot, CU 0
+      if (a.prof && cu == 0 && lane == 0)
+        a.prof[(u64)a.n_layers * 8 + s] = __builtin_amdgcn_s_memtime();
       {  // gather-pass thinning: one outstanding fill while the CU sweeps
         unsigned busy;
         asm volatile("ds_read_b32 %0, %1\n\ts_waitcnt lgkmcnt(0)"
                      : "=&v"(busy) : "v"(b_off) : "memory");
         if (busy) asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
       }
-      if (s >= LAG - 1)
-        loader_publish<(LAG - 1) * 16>(f_off, (unsigned)(s - LAG + 2));
+      if (s >= LAG - 1) ENG_PUB((LAG - 1) * 16, (unsigned)(s - LAG + 2));
     }
-    loader_publish<0>(f_off, (unsigned)total);
+    ENG_PUB(0, (unsigned)total);
+#undef ENG_PUB
+    if (a.prof && lane == 0)  // loader space-poll spins (all CUs summed)
+      __hip_atomic_fetch_add((gu32*)a.tmo + 2, poll_spins_tot,
+                             LFMQ_RLX_AGENT);
     return;
   }
 
@@ -290,6 +313,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
   unsigned gseq = 0;
   unsigned phase_want = 0;
   bool ok = true;
+  unsigned dbg_fill = 0, dbg_sweep = 0, dbg_sync = 0;
 
 #define ENG_CHECK(x) do { if (!(x)) return; } while (0)
 #define ENG_PROF(slot_)                                                   \
@@ -315,7 +339,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     if (lane == 0) ctl->ss_part[wave] = ss;
     __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
     ++gseq;
-    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x20, lane));
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x20, lane, &dbg_sync));
     if (wave == 0) {
       const float sst = ctl->ss_part[0] + ctl->ss_part[1] + ctl->ss_part[2];
       ctl->inv_cur = rsqrtf(sst / (float)a.hidden + a.rms_eps);
@@ -338,7 +362,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       for (int r = 0; r < npc_qkv; ++r) {
         const u64 at = cursor_next(cur, rb_qkv, ring_bytes);
         if ((r >> 1) % 3 != wave) continue;
-        ENG_CHECK(wait_fill(ctl, at + rb_qkv, tmo, 0x30, lane));
+        ENG_CHECK(wait_fill(ctl, at + rb_qkv, tmo, 0x30, lane, &dbg_fill));
         const float v = inv * record_dot(ring + (u32)(at % ring_bytes),
                                          (const char*)xbuf, a.hidden, G1,
                                          a.log2g, lane);
@@ -358,9 +382,9 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     // ---------------- P2: gather qkv into union ------------------------
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
     ENG_CHECK(gather_share(g_qkv, a.nqkv / 2, e_qkv, (u32*)unionb, wave, ctl,
-                           tmo, 0x31, lane));
+                           tmo, 0x31, lane, &dbg_sweep));
     ++gseq;
-    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x32, lane));
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x32, lane, &dbg_sync));
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
     ENG_PROF(2);
 
@@ -521,9 +545,9 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     // ---------------- P4: gather attention output ----------------------
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
     ENG_CHECK(gather_share(g_attn, a.hq * D / 2, e_attn, (u32*)unionb, wave,
-                           ctl, tmo, 0x50, lane));
+                           ctl, tmo, 0x50, lane, &dbg_sweep));
     ++gseq;
-    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x51, lane));
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x51, lane, &dbg_sync));
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
     ENG_PROF(4);
 
@@ -533,7 +557,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       for (int r = 0; r < npc_o; ++r) {
         const u64 at = cursor_next(cur, rb_o, ring_bytes);
         if ((r >> 1) % 3 != wave) continue;
-        ENG_CHECK(wait_fill(ctl, at + rb_o, tmo, 0x52, lane));
+        ENG_CHECK(wait_fill(ctl, at + rb_o, tmo, 0x52, lane, &dbg_fill));
         const float v = record_dot(ring + (u32)(at % ring_bytes), unionb,
                                    a.hq * D, Go, a.log2g, lane);
         if (r & 1) {
@@ -552,7 +576,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     ENG_CHECK(gather_share(g_o, a.hidden / 2, e_o, (u32*)ascr, wave, ctl, tmo,
                            0x53, lane));
     ++gseq;
-    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x54, lane));
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x54, lane, &dbg_sync));
     {
       u32* x32 = (u32*)xbuf;
       const u32* add32 = (const u32*)ascr;
@@ -573,7 +597,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       if (lane == 0) ctl->ss_part[wave] = ss;
       __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
       ++gseq;
-      ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x55, lane));
+      ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x55, lane, &dbg_sync));
       if (wave == 0) {
         lds_flag_store_rlx(&ctl->busy, 0u);
         const float sst = ctl->ss_part[0] + ctl->ss_part[1] + ctl->ss_part[2];
@@ -593,7 +617,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       for (int r = 0; r < npc_gu; ++r) {
         const u64 at = cursor_next(cur, rb_gu, ring_bytes);
         if ((r >> 2) % 3 != wave) continue;
-        ENG_CHECK(wait_fill(ctl, at + rb_gu, tmo, 0x60, lane));
+        ENG_CHECK(wait_fill(ctl, at + rb_gu, tmo, 0x60, lane, &dbg_fill));
         const float v = inv * record_dot(ring + (u32)(at % ring_bytes),
                                          (const char*)xbuf, a.hidden, G1,
                                          a.log2g, lane);
@@ -617,9 +641,9 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     // ---------------- P8: gather h ------------------------------------
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
     ENG_CHECK(gather_share(g_h, a.inter / 2, e_h, (u32*)unionb, wave, ctl,
-                           tmo, 0x61, lane));
+                           tmo, 0x61, lane, &dbg_sweep));
     ++gseq;
-    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x62, lane));
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x62, lane, &dbg_sync));
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
 
     // ---------------- P9: down records (src = hbuf) --------------------
@@ -628,7 +652,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       for (int r = 0; r < npc_dn; ++r) {
         const u64 at = cursor_next(cur, rb_dn, ring_bytes);
         if ((r >> 1) % 3 != wave) continue;
-        ENG_CHECK(wait_fill(ctl, at + rb_dn, tmo, 0x63, lane));
+        ENG_CHECK(wait_fill(ctl, at + rb_dn, tmo, 0x63, lane, &dbg_fill));
         const float v = record_dot(ring + (u32)(at % ring_bytes), unionb,
                                    a.inter, G2, a.log2g, lane);
         if (r & 1) {
@@ -647,9 +671,9 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     // ---------------- P10: gather down, resid-add, inv1 of next layer --
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
     ENG_CHECK(gather_share(g_down, a.hidden / 2, e_down, (u32*)ascr, wave,
-                           ctl, tmo, 0x64, lane));
+                           ctl, tmo, 0x64, lane, &dbg_sweep));
     ++gseq;
-    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x65, lane));
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x65, lane, &dbg_sync));
     {
       u32* x32 = (u32*)xbuf;
       const u32* add32 = (const u32*)ascr;
@@ -670,7 +694,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       if (lane == 0) ctl->ss_part[wave] = ss;
       __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
       ++gseq;
-      ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x66, lane));
+      ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x66, lane, &dbg_sync));
       if (wave == 0) {
         lds_flag_store_rlx(&ctl->busy, 0u);
         const float sst = ctl->ss_part[0] + ctl->ss_part[1] + ctl->ss_part[2];
@@ -681,6 +705,12 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
         ENG_CHECK(lds_wait_ge(&ctl->phase, phase_want, tmo, 0x67, lane));
       }
     }
+  }
+
+  if (a.prof && lane == 0) {
+    __hip_atomic_fetch_add((gu32*)a.tmo + 1, dbg_fill, LFMQ_RLX_AGENT);
+    __hip_atomic_fetch_add((gu32*)a.tmo + 3, dbg_sweep, LFMQ_RLX_AGENT);
+    __hip_atomic_fetch_add((gu32*)a.tmo + 4, dbg_sync, LFMQ_RLX_AGENT);
   }
 
   // ---- epilogue: CU 0 wave 0 writes the final-norm output -------------

@@ -55,7 +55,8 @@ __device__ __forceinline__ void store_granule(gu64* g, unsigned epoch,
 // false on spin give-up (outputs poisoned by the caller's memset protocol).
 __device__ __forceinline__ bool sweep_stage(gu64* g, int cnt, unsigned epoch,
                                             u32* lds_dst, gu32* tmo,
-                                            unsigned code, int lane) {
+                                            unsigned code, int lane,
+                                            unsigned* passes = nullptr) {
   for (unsigned spins = 0;;) {
     bool ok = true;
     for (int idx = lane; idx < cnt; idx += LFMQ_WAVE) {
@@ -63,6 +64,7 @@ __device__ __forceinline__ bool sweep_stage(gu64* g, int cnt, unsigned epoch,
       lds_dst[idx] = (unsigned)x;
       ok &= (unsigned)(x >> 32) == epoch;
     }
+    if (passes) ++*passes;
     if (__all(ok)) return true;
     if (spin_fail(spins, tmo, code, lane)) return false;
   }
@@ -108,9 +110,13 @@ __device__ __forceinline__ unsigned lds_flag_load(const unsigned* f) {
 // bounded poll until *f >= v (monotonic counters)
 __device__ __forceinline__ bool lds_wait_ge(const unsigned* f, unsigned v,
                                             gu32* tmo, unsigned code,
-                                            int lane) {
+                                            int lane,
+                                            unsigned* acc = nullptr) {
   for (unsigned spins = 0;;) {
-    if (lds_flag_load(f) >= v) return true;
+    if (lds_flag_load(f) >= v) {
+      if (acc) *acc += spins;
+      return true;
+    }
     if (spin_fail(spins, tmo, code, lane)) return false;
   }
 }

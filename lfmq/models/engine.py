@@ -118,14 +118,15 @@ class DecodeEngine:
         self.g_h = torch.zeros(inter // 2, **i64)
         self.g_down = torch.zeros(hid // 2, **i64)
         self.g_scr = torch.zeros(hq * nch_max * (d + 2), **i64)
-        self.tmo = torch.zeros(1, dtype=torch.int32, device=dev)
+        self.tmo = torch.zeros(8, dtype=torch.int32, device=dev)
         self.xout = torch.zeros(1, 1, hid, dtype=torch.bfloat16, device=dev)
         self.inv_freq = F._inv_freq(d, cfg.rope_theta, dev)
         self.attn_scale = 1.0 / (d ** 0.5)
         self.prof: Optional[torch.Tensor] = None
         if os.environ.get("LFMQ_ENGINE_PROF"):
-            self.prof = torch.zeros(cfg.n_layers * 8, dtype=torch.int64,
-                                    device=dev)
+            n_slots = (self.per_cu + SLOT - 1) // SLOT
+            self.prof = torch.zeros(cfg.n_layers * 8 + n_slots,
+                                    dtype=torch.int64, device=dev)
 
     # ------------------------------------------------------------ build
     @torch.no_grad()

@@ -25,9 +25,14 @@ def build(n_layers, seed=7):
 
 
 @torch.no_grad()
-def steps(model, cfg, ids, nsteps, use_engine, seed=11):
+def steps(model, cfg, ids, nsteps, variant, seed=11):
+    """variant: 'engine' | 'fused' (launched, norm-fused GEMVs) |
+    'nonfused' (launched, separate norm kernels — the numerical-noise
+    yardstick between two valid launched implementations)."""
+    use_engine = variant == "engine"
     model._engine = None
     model._engine_unsupported = False
+    model._fused_decode = None if variant != "nonfused" else False
     os.environ["LFMQ_ENGINE"] = "1" if use_engine else "0"
     g = torch.Generator().manual_seed(seed)
     forced = torch.randint(0, cfg.vocab_size, (nsteps,), generator=g)
@@ -46,9 +51,14 @@ def steps(model, cfg, ids, nsteps, use_engine, seed=11):
     if use_engine:
         assert model._engine is not None
         model._engine.check_health()
+        t = model._engine.tmo.cpu().tolist()
+        print(f"   spins(last step): consumer_fill={t[1]} loader_space={t[2]} "
+              f"sweep_passes={t[3]} sync={t[4]}")
+        model._engine.tmo.zero_()
         if model._engine.prof is not None:
             prof = model._engine.prof.clone().cpu()
     os.environ.pop("LFMQ_ENGINE", None)
+    model._fused_decode = None
     return torch.cat(outs), prof
 
 
@@ -68,17 +78,18 @@ def main():
         q, cfg = build(nl)
         for prompt, ns in ((64, 8), (300, 4)):
             ids = torch.randint(0, cfg.vocab_size, (1, prompt), device="cuda")
-            ref, _ = steps(q, cfg, ids, ns, False)
-            got, prof = steps(q, cfg, ids, ns, True)
-            mx, mn = rel(got, ref)
-            per = [float((got[i] - ref[i]).abs().max() / ref[i].abs().max())
-                   for i in range(ns)]
-            print(f"layers={nl} prompt={prompt} steps={ns}: "
-                  f"max_rel={mx:.5f} mean_rel={mn:.5f} "
-                  f"per_step_rel={[f'{p:.4f}' for p in per]} "
-                  f"ref_scale={float(ref.abs().max()):.3f}")
+            ref, _ = steps(q, cfg, ids, ns, "fused")
+            ref2, _ = steps(q, cfg, ids, ns, "nonfused")
+            got, prof = steps(q, cfg, ids, ns, "engine")
+            eng_abs = [float((got[i] - ref[i]).abs().max()) for i in range(ns)]
+            yard = [float((ref2[i] - ref[i]).abs().max()) for i in range(ns)]
+            scale = [float(ref[i].abs().max()) for i in range(ns)]
+            print(f"layers={nl} prompt={prompt} steps={ns}:")
+            print(f"   eng_abs={[f'{v:.4f}' for v in eng_abs]}")
+            print(f"   yardstick(fused-vs-nonfused)={[f'{v:.4f}' for v in yard]}")
+            print(f"   ref_step_scale={[f'{v:.3f}' for v in scale]}")
             if prof is not None:
-                p = prof.reshape(nl, 8).double()
+                p = prof[:nl * 8].reshape(nl, 8).double()
                 # deltas within layer (stamps 0..7) + layer-to-layer
                 intra = (p[:, 1:] - p[:, :-1])
                 l2l = p[1:, 0] - p[:-1, 0] if nl > 1 else None
@@ -88,6 +99,16 @@ def main():
                 tot = float((p[-1, 7] - p[0, 0]) / 2.0e3)
                 print(f"   phase medians (assuming 2GHz): {names}; "
                       f"stamp-span/step={tot:.1f}us")
+                lts = prof[nl * 8:].double()
+                lts = lts[lts > 0]
+                if lts.numel() > 2:
+                    d = (lts[1:] - lts[:-1]) / 2.0e3  # us between slot issues
+                    dd = d.sort().values
+                    print(f"   loader: {lts.numel()} slots, slot-issue deltas "
+                          f"med={float(dd[len(dd)//2]):.2f}us "
+                          f"p90={float(dd[int(len(dd)*0.9)]):.2f}us "
+                          f"max={float(dd[-1]):.2f}us "
+                          f"span={float((lts[-1]-lts[0])/2.0e3):.1f}us")
         del q
         torch.cuda.empty_cache()
 
