@@ -105,20 +105,22 @@ __device__ __forceinline__ u64 cursor_next(Cursor& c, int rec, u32 ring_bytes) {
 __device__ __forceinline__ float record_dot(const char* lds_rec,
                                             const char* lds_src, int k, int G,
                                             int log2g, int lane) {
-  constexpr int CH = 8;
+  // CH must divide k/512 for every op (host asserts k % 2048 == 0): the
+  // chunk body is then BRANCH-FREE — a runtime `if (j >= n) break` inside
+  // #pragma unroll defeats the unroll and serializes every LDS load on
+  // its own latency (measured ~600 cycles/iteration before this fix).
+  constexpr int CH = 4;
   const int iters = k >> 9;
   const u32* wbase = (const u32*)lds_rec;
   const unsigned short* sbase = (const unsigned short*)(lds_rec + k / 2);
   const unsigned char* zbase = (const unsigned char*)(lds_rec + k / 2 + G * 2);
   float acc = 0.f;
   for (int c0 = 0; c0 < iters; c0 += CH) {
-    const int n = min(CH, iters - c0);
     u32 w[CH];
     u32x4 xq[CH];
     float s[CH], sz[CH];
 #pragma unroll
     for (int j = 0; j < CH; ++j) {
-      if (j >= n) break;
       const int i = c0 + j;
       const int k0 = i * 512 + lane * 8;
       w[j] = wbase[i * 64 + lane];
@@ -132,7 +134,6 @@ __device__ __forceinline__ float record_dot(const char* lds_rec,
     }
 #pragma unroll
     for (int j = 0; j < CH; ++j) {
-      if (j >= n) break;
       const DequantPairs4 dq = dequant4_pairs(w[j]);
       float cx = 0.f, xs = 0.f;
       cx = dot2_bf16(dq.p01, xq[j][0], cx);
@@ -466,15 +467,17 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
         // partial pairs + unroll keep ~8 V-row loads in flight (a single
         // accumulate chain serializes on global-load latency — the same
         // structure as csrc/attn_decode.hip's PV pass).
+        // branch-free main blocks of 4 + a short tail (a break inside
+        // #pragma unroll kills the unroll — same lesson as record_dot)
         float oa[4][2] = {{0.f, 0.f}, {0.f, 0.f}, {0.f, 0.f}, {0.f, 0.f}};
         const int nt = min(AT, T - c * AT);
+        const int nt4 = nt & ~3;
         const u32 v_last = qkv32[((u64)a.hq + a.hkv + hk) * (D / 2) + lane];
 #pragma unroll 2
-        for (int tb = 0; tb < nt; tb += 4) {
+        for (int tb = 0; tb < nt4; tb += 4) {
 #pragma unroll
           for (int q = 0; q < 4; ++q) {
             const int tt = tb + q;
-            if (tt >= nt) break;
             const int t = c * AT + tt;
             const u32 vd =
                 (t == pos) ? v_last : ((const u32*)(vcache + (u64)t * D))[lane];
@@ -484,6 +487,16 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
             oa[q][0] = fmaf(p, v0, oa[q][0]);
             oa[q][1] = fmaf(p, v1, oa[q][1]);
           }
+        }
+        for (int tt = nt4; tt < nt; ++tt) {
+          const int t = c * AT + tt;
+          const u32 vd =
+              (t == pos) ? v_last : ((const u32*)(vcache + (u64)t * D))[lane];
+          float v0, v1;
+          bf16x2_to_f32(vd, v0, v1);
+          const float p = ps[tt];
+          oa[tt & 3][0] = fmaf(p, v0, oa[tt & 3][0]);
+          oa[tt & 3][1] = fmaf(p, v1, oa[tt & 3][1]);
         }
         const float o0 = (oa[0][0] + oa[1][0]) + (oa[2][0] + oa[3][0]);
         const float o1 = (oa[0][1] + oa[1][1]) + (oa[2][1] + oa[3][1]);

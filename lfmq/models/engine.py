@@ -59,8 +59,9 @@ def engine_supported(model, batch: int) -> bool:
     if cfg.head_dim != 128:
         return False
     nqkv = (cfg.n_heads + 2 * cfg.n_kv_heads) * cfg.head_dim
-    if (cfg.hidden_size % 512 or cfg.intermediate_size % 512 or nqkv % 512
-            or (cfg.n_heads * cfg.head_dim) % 512):
+    # k dims % 2048: record_dot runs branch-free 4x512-value chunks
+    if (cfg.hidden_size % 2048 or cfg.intermediate_size % 2048 or nqkv % 512
+            or (cfg.n_heads * cfg.head_dim) % 2048):
         return False
     g = None
     for layer in model.layers:
@@ -221,6 +222,6 @@ class DecodeEngine:
 
     def check_health(self):
         """Host-side give-up check (cheap sync; call between batches)."""
-        code = int(self.tmo.item())
+        code = int(self.tmo[0].item())
         if code != 0:
             raise RuntimeError(f"decode engine spin give-up code 0x{code:x}")

@@ -14,10 +14,10 @@ from lfmq.models.engine import (DecodeEngine, engine_supported, _rec_bytes,
 
 
 def _engine_model():
-    # smallest config the engine supports: d=128, all dims % 512 == 0
-    cfg = LlamaConfig(vocab_size=1024, hidden_size=512, intermediate_size=512,
-                      n_layers=2, n_heads=4, n_kv_heads=2, head_dim=128,
-                      max_seq=512)
+    # smallest config the engine supports: d=128, k dims % 2048 == 0
+    cfg = LlamaConfig(vocab_size=1024, hidden_size=2048,
+                      intermediate_size=2048, n_layers=2, n_heads=16,
+                      n_kv_heads=4, head_dim=128, max_seq=512)
     m = lfmq.build_model(cfg, seed=3)
     q = lfmq.quantize(m, lfmq.QuantConfig(scheme="rtn", bits=4, group_size=128))
     return q
@@ -28,9 +28,9 @@ def test_engine_supported_gates():
     assert engine_supported(q, batch=1)
     assert not engine_supported(q, batch=2)
     # unsupported: group too small for the lane-chunk layout
-    cfg = LlamaConfig(vocab_size=256, hidden_size=512, intermediate_size=512,
-                      n_layers=1, n_heads=4, n_kv_heads=2, head_dim=128,
-                      max_seq=256)
+    cfg = LlamaConfig(vocab_size=256, hidden_size=2048,
+                      intermediate_size=2048, n_layers=1, n_heads=16,
+                      n_kv_heads=4, head_dim=128, max_seq=256)
     q2 = lfmq.quantize(lfmq.build_model(cfg, seed=1),
                        lfmq.QuantConfig(scheme="rtn", bits=4, group_size=16))
     assert not engine_supported(q2, batch=1)
