@@ -80,6 +80,28 @@ __device__ __forceinline__ float wave_reduce_sum(float v) {
   for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
   return v;
 }
+
+// DPP-based wave sum: ~6 VALU ops instead of 6 LDS-unit bpermute round
+// trips (~20 vs ~500 cycles — decisive at 1 wave/SIMD where nothing hides
+// the bpermute latency).  rocPRIM's classic CDNA ladder: row_shr 1/2/4/8
+// (bound_ctrl: shifted-in lanes read 0), then row_bcast:15 into rows 1|3,
+// row_bcast:31 into rows 2|3; lane 63 holds the total, broadcast via
+// readlane.  The update_dpp builtin is compiler-modeled (hazards padded).
+__device__ __forceinline__ float wave_reduce_sum_dpp(float v) {
+#define LFMQ_DPP_ADD(ctrl, rmask)                                         \
+  v += __builtin_bit_cast(float,                                          \
+      __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, v), (ctrl),  \
+                                  (rmask), 0xf, true))
+  LFMQ_DPP_ADD(0x111, 0xf);  // row_shr:1
+  LFMQ_DPP_ADD(0x112, 0xf);  // row_shr:2
+  LFMQ_DPP_ADD(0x114, 0xf);  // row_shr:4
+  LFMQ_DPP_ADD(0x118, 0xf);  // row_shr:8
+  LFMQ_DPP_ADD(0x142, 0xa);  // row_bcast:15 -> rows 1,3
+  LFMQ_DPP_ADD(0x143, 0xc);  // row_bcast:31 -> rows 2,3
+#undef LFMQ_DPP_ADD
+  return __builtin_bit_cast(
+      float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, v), 63));
+}
 __device__ __forceinline__ float wave_reduce_max(float v) {
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));

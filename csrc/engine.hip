@@ -102,55 +102,6 @@ __device__ __forceinline__ u64 cursor_next(Cursor& c, int rec, u32 ring_bytes) {
 // KI-unroll): per-iteration loads make the compiler reuse registers and
 // serialize the loop on LDS latency.  CHUNK=8 covers k=4096 ops in one
 // chunk; k=14336 (down) runs 3.5 chunks.
-__device__ __forceinline__ float record_dot(const char* lds_rec,
-                                            const char* lds_src, int k, int G,
-                                            int log2g, int lane) {
-  // CH must divide k/512 for every op (host asserts k % 2048 == 0): the
-  // chunk body is then BRANCH-FREE — a runtime `if (j >= n) break` inside
-  // #pragma unroll defeats the unroll and serializes every LDS load on
-  // its own latency (measured ~600 cycles/iteration before this fix).
-  constexpr int CH = 4;
-  const int iters = k >> 9;
-  const u32* wbase = (const u32*)lds_rec;
-  const unsigned short* sbase = (const unsigned short*)(lds_rec + k / 2);
-  const unsigned char* zbase = (const unsigned char*)(lds_rec + k / 2 + G * 2);
-  float acc = 0.f;
-  for (int c0 = 0; c0 < iters; c0 += CH) {
-    u32 w[CH];
-    u32x4 xq[CH];
-    float s[CH], sz[CH];
-#pragma unroll
-    for (int j = 0; j < CH; ++j) {
-      const int i = c0 + j;
-      const int k0 = i * 512 + lane * 8;
-      w[j] = wbase[i * 64 + lane];
-      xq[j] = *(const u32x4*)(lds_src + k0 * 2);
-      const int gi = k0 >> log2g;
-      const u32 sraw = sbase[gi];
-      const u32 zb = zbase[gi >> 1];
-      const u32 z = (gi & 1) ? (zb >> 4) : (zb & 0xFu);
-      s[j] = bf16_bits_to_f32(sraw);
-      sz[j] = s[j] * (float)(z + 128u);
-    }
-#pragma unroll
-    for (int j = 0; j < CH; ++j) {
-      const DequantPairs4 dq = dequant4_pairs(w[j]);
-      float cx = 0.f, xs = 0.f;
-      cx = dot2_bf16(dq.p01, xq[j][0], cx);
-      cx = dot2_bf16(dq.p23, xq[j][1], cx);
-      cx = dot2_bf16(dq.p45, xq[j][2], cx);
-      cx = dot2_bf16(dq.p67, xq[j][3], cx);
-      xs = dot2_bf16(ONE2_BF16, xq[j][0], xs);
-      xs = dot2_bf16(ONE2_BF16, xq[j][1], xs);
-      xs = dot2_bf16(ONE2_BF16, xq[j][2], xs);
-      xs = dot2_bf16(ONE2_BF16, xq[j][3], xs);
-      acc = fmaf(s[j], cx, acc);
-      acc = fmaf(-sz[j], xs, acc);
-    }
-  }
-  return wave_reduce_sum(acc);
-}
-
 // wait until the ring holds stream bytes [0, end_off): filled*SLOT >= end
 __device__ __forceinline__ bool wait_fill(Ctl* ctl, u64 end_off, gu32* tmo,
                                           unsigned code, int lane,

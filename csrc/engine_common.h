@@ -67,6 +67,13 @@ __device__ __forceinline__ bool sweep_stage(gu64* g, int cnt, unsigned epoch,
     if (passes) ++*passes;
     if (__all(ok)) return true;
     if (spin_fail(spins, tmo, code, lane)) return false;
+    // back off: each pass re-reads cnt/64 sc1 granules per lane — real
+    // fabric traffic from every CU; widen the interval once the edge is
+    // clearly not ready yet (keeps gather latency, cuts poll bandwidth).
+    if (spins > 8)
+      __builtin_amdgcn_s_sleep(64);
+    else if (spins > 2)
+      __builtin_amdgcn_s_sleep(16);
   }
 }
 
@@ -175,5 +182,91 @@ __device__ __forceinline__ void glds_fill_1k(const char* gsrc_lane,
   __builtin_amdgcn_global_load_lds((lfmq_gptr)gsrc_lane, (lfmq_lptr)lds_base,
                                    16, 0, /*aux nt*/ 2);
 }
+
+// A chunk is CH=4 k-iterations (2048 values); CH must divide k/512 for
+// every op (host asserts k % 2048 == 0) so the bodies are BRANCH-FREE —
+// a runtime `if (j >= n) break` inside #pragma unroll defeats the unroll
+// and serializes every LDS load on its own latency.  Chunks are DOUBLE-
+// BUFFERED (load c+1 while computing c): at 1 wave/SIMD there is no other
+// wave to hide the ~64-cycle LDS latency, so the wave's own ILP must —
+// the isolated microbench measured 0.78 us/record chunk-serial.
+struct RecChunk {
+  u32 w[4];
+  u32x4 xq[4];
+  float s[4], sz[4];
+};
+
+__device__ __forceinline__ void chunk_load(RecChunk& ch, int c0,
+                                           const u32* wbase,
+                                           const unsigned short* sbase,
+                                           const unsigned char* zbase,
+                                           const char* lds_src, int log2g,
+                                           int lane) {
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int i = c0 + j;
+    const int k0 = i * 512 + lane * 8;
+    ch.w[j] = wbase[i * 64 + lane];
+    ch.xq[j] = *(const u32x4*)(lds_src + k0 * 2);
+    const int gi = k0 >> log2g;
+    const u32 sraw = sbase[gi];
+    const u32 zb = zbase[gi >> 1];
+    const u32 z = (gi & 1) ? (zb >> 4) : (zb & 0xFu);
+    ch.s[j] = bf16_bits_to_f32(sraw);
+    ch.sz[j] = ch.s[j] * (float)(z + 128u);
+  }
+}
+
+__device__ __forceinline__ void chunk_dot(const RecChunk& ch, float& acc) {
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const DequantPairs4 dq = dequant4_pairs(ch.w[j]);
+    float cx = 0.f, xs = 0.f;
+    cx = dot2_bf16(dq.p01, ch.xq[j][0], cx);
+    cx = dot2_bf16(dq.p23, ch.xq[j][1], cx);
+    cx = dot2_bf16(dq.p45, ch.xq[j][2], cx);
+    cx = dot2_bf16(dq.p67, ch.xq[j][3], cx);
+    xs = dot2_bf16(0x3F803F80u, ch.xq[j][0], xs);
+    xs = dot2_bf16(0x3F803F80u, ch.xq[j][1], xs);
+    xs = dot2_bf16(0x3F803F80u, ch.xq[j][2], xs);
+    xs = dot2_bf16(0x3F803F80u, ch.xq[j][3], xs);
+    acc = fmaf(ch.s[j], cx, acc);
+    acc = fmaf(-ch.sz[j], xs, acc);
+  }
+}
+
+__device__ __forceinline__ float record_dot_nored(const char* lds_rec,
+                                                  const char* lds_src, int k,
+                                                  int G, int log2g, int lane) {
+  const int iters = k >> 9;
+  const u32* wbase = (const u32*)lds_rec;
+  const unsigned short* sbase = (const unsigned short*)(lds_rec + k / 2);
+  const unsigned char* zbase = (const unsigned char*)(lds_rec + k / 2 + G * 2);
+  float acc = 0.f;
+  RecChunk A, B;
+  chunk_load(A, 0, wbase, sbase, zbase, lds_src, log2g, lane);
+  int c0 = 0;
+  for (;;) {  // ping-pong; iters may be an odd number of chunks (down: 7)
+    if (c0 + 4 < iters)
+      chunk_load(B, c0 + 4, wbase, sbase, zbase, lds_src, log2g, lane);
+    chunk_dot(A, acc);
+    c0 += 4;
+    if (c0 >= iters) break;
+    if (c0 + 4 < iters)
+      chunk_load(A, c0 + 4, wbase, sbase, zbase, lds_src, log2g, lane);
+    chunk_dot(B, acc);
+    c0 += 4;
+    if (c0 >= iters) break;
+  }
+  return acc;
+}
+
+__device__ __forceinline__ float record_dot(const char* lds_rec,
+                                            const char* lds_src, int k, int G,
+                                            int log2g, int lane) {
+  return wave_reduce_sum_dpp(
+      record_dot_nored(lds_rec, lds_src, k, G, log2g, lane));
+}
+
 
 }  // namespace lfmq_engine

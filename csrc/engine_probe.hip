@@ -144,6 +144,19 @@ __global__ __launch_bounds__(TPB) void allgather_kernel(
     }                                                             \
   } while (0)
 
+namespace eng_ubench {
+void run_record_ubench();
+}
+
+// DPP reduce must match the shfl reduce bit-for-bit-ish (different add
+// order: compare with tight fp32 tolerance)
+__global__ __launch_bounds__(64) void reduce_check_kernel(
+    const float* __restrict__ in, float* __restrict__ out) {
+  const float v = in[blockIdx.x * 64 + threadIdx.x];
+  out[blockIdx.x * 2] = wave_reduce_sum(v);
+  out[blockIdx.x * 2 + 1] = wave_reduce_sum_dpp(v);
+}
+
 int main() {
   int fails = 0;
 
@@ -263,6 +276,138 @@ int main() {
     hipFree((void*)g0); hipFree((void*)g1); hipFree(out); hipFree((void*)tmo);
   }
 
+  // ---- DPP reduce equivalence ----
+  {
+    const int n = 128;
+    float *in, *out;
+    CHK(hipMalloc(&in, n * 64 * 4));
+    CHK(hipMalloc(&out, n * 2 * 4));
+    std::vector<float> h(n * 64);
+    for (size_t i = 0; i < h.size(); ++i)
+      h[i] = (float)((i * 2654435761u) % 1000) / 37.f - 13.f;
+    CHK(hipMemcpy(in, h.data(), h.size() * 4, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(reduce_check_kernel, dim3(n), dim3(64), 0, 0, in, out);
+    CHK(hipDeviceSynchronize());
+    std::vector<float> o(n * 2);
+    CHK(hipMemcpy(o.data(), out, n * 2 * 4, hipMemcpyDeviceToHost));
+    int bad = 0;
+    for (int i = 0; i < n; ++i) {
+      const float a = o[2 * i], b = o[2 * i + 1];
+      if (fabsf(a - b) > 1e-3f * fmaxf(fabsf(a), 1.f)) bad++;
+    }
+    printf("dpp reduce check: %s (%d bad)\n", bad ? "FAIL" : "ok", bad);
+    if (bad) fails++;
+    hipFree(in); hipFree(out);
+  }
+
+  eng_ubench::run_record_ubench();
+
   printf(fails ? "ENGINE_PROBE FAIL (%d)\n" : "ENGINE_PROBE PASS\n", fails);
   return fails ? 1 : 0;
 }
+
+// ---------------------------------------------------- record microbench
+// The engine's consumer record loop in ISOLATION (no loader, no cross-CU
+// waits): LDS pre-filled once, then REPS passes over NREC records of
+// k=4096 exactly as csrc/engine.hip P7 (gate_up) processes them.
+// ABLATE: 0 = full (dot+reduce+silu+publish), 1 = no publish,
+//         2 = no reduce (lane-0 partial only), 3 = dot only.
+#include "../csrc/engine_args.h"
+namespace eng_ubench {
+constexpr int KREC = 4096;
+constexpr int GREC = 32;                       // k/g, g=128
+constexpr int RB = (KREC / 2 + GREC * 2 + 16 + 15) & ~15;  // 2128
+constexpr int NREC = 36;
+
+template <int ABLATE>
+__global__ __launch_bounds__(TPB) void record_ubench_kernel(
+    const char* __restrict__ src, unsigned long long* __restrict__ gran,
+    unsigned long long* __restrict__ tstamp, int reps) {
+  __shared__ char recs[NREC * RB];
+  __shared__ char xbuf[8192];
+  const int wave = threadIdx.x / LFMQ_WAVE;
+  const int lane = threadIdx.x % LFMQ_WAVE;
+  // fill LDS once (all 4 waves)
+  for (int i = threadIdx.x * 16; i < NREC * RB; i += TPB * 16)
+    *(u32x4*)(recs + i) = *(const u32x4*)(src + i);
+  for (int i = threadIdx.x * 16; i < 8192; i += TPB * 16)
+    *(u32x4*)(xbuf + i) = *(const u32x4*)(src + i);
+  __syncthreads();
+  if (wave == 3) return;
+  gu64* g = (gu64*)gran + blockIdx.x * NREC;
+  unsigned long long t0 = 0;
+  if (blockIdx.x == 0 && wave == 0 && lane == 0)
+    t0 = __builtin_amdgcn_s_memtime();
+  float sink = 0.f;
+  for (int rep = 0; rep < reps; ++rep) {
+    float vg = 0.f, h0 = 0.f;
+    for (int r = 0; r < NREC; ++r) {
+      if ((r >> 2) % 3 != wave) continue;
+      float v = ABLATE >= 2 ? lfmq_engine::record_dot_nored(recs + r * RB, xbuf, KREC, GREC, 7, lane)
+                      : lfmq_engine::record_dot(recs + r * RB, xbuf, KREC, GREC, 7, lane);
+      if (ABLATE < 3) {
+        switch (r & 3) {
+          case 0: vg = v; break;
+          case 1: h0 = (vg / (1.f + __expf(-vg))) * v; break;
+          case 2: vg = v; break;
+          default: {
+            const float h1 = (vg / (1.f + __expf(-vg))) * v;
+            if (ABLATE < 1 && lane == 0)
+              lfmq_engine::store_granule(g + (r >> 1), rep + 1,
+                                         pack_bf16x2(h0, h1));
+            sink += h1;
+          }
+        }
+      } else {
+        sink += v;
+      }
+    }
+  }
+  if (blockIdx.x == 0 && wave == 0 && lane == 0) {
+    const unsigned long long t1 = __builtin_amdgcn_s_memtime();
+    tstamp[ABLATE] = t1 - t0;
+  }
+  if (sink == 1234.5f) ((float*)gran)[0] = sink;  // keep alive
+}
+
+void run_record_ubench() {
+  const int reps = 200;
+  char* src;
+  unsigned long long *gran, *ts;
+  (void)hipMalloc(&src, NREC * RB);
+  (void)hipMalloc((void**)&gran, (u64)NWG * NREC * 8);
+  (void)hipMalloc((void**)&ts, 8 * 8);
+  std::vector<char> h(NREC * RB);
+  for (size_t i = 0; i < h.size(); ++i) h[i] = (char)(i * 131 + 7);
+  (void)hipMemcpy(src, h.data(), h.size(), hipMemcpyHostToDevice);
+  (void)hipMemset(ts, 0, 64);
+  hipEvent_t e0, e1;
+  (void)hipEventCreate(&e0);
+  (void)hipEventCreate(&e1);
+#define RUN(AB)                                                            \
+  do {                                                                     \
+    (void)hipEventRecord(e0);                                              \
+    hipLaunchKernelGGL(record_ubench_kernel<AB>, dim3(NWG), dim3(TPB), 0,  \
+                       0, src, gran, ts, reps);                            \
+    (void)hipEventRecord(e1);                                              \
+    (void)hipDeviceSynchronize();                                          \
+    float ms;                                                              \
+    (void)hipEventElapsedTime(&ms, e0, e1);                                \
+    unsigned long long st;                                                 \
+    (void)hipMemcpy(&st, ts + AB, 8, hipMemcpyDeviceToHost);               \
+    const double recs_per_wave = (double)reps * 12.0; /* 36/3 */           \
+    printf("record_ubench ablate=%d: %.3f ms wall -> %.3f us/record "      \
+           "(stamp %.3f us/rec @2GHz)\n",                                  \
+           AB, ms, ms * 1e3 / recs_per_wave,                               \
+           (double)st / 2e3 / recs_per_wave);                              \
+  } while (0)
+  RUN(0);
+  RUN(1);
+  RUN(2);
+  RUN(3);
+#undef RUN
+  (void)hipFree(src);
+  (void)hipFree(gran);
+  (void)hipFree(ts);
+}
+}  // namespace eng_ubench
