@@ -70,9 +70,9 @@ struct Ctl {
   float ss_part[3];      // per-wave sum-of-squares partials
 };
 
-// record byte size (padded to 16) — device twin of lfmq_engine_rec_bytes
+// record byte size v2 (padded to 16) — device twin of lfmq_engine_rec_bytes
 __device__ inline int rec_bytes(int k, int G) {
-  return (k / 2 + G * 2 + (G + 1) / 2 + 15) & ~15;
+  return (k / 2 + G * 8 + 15) & ~15;
 }
 
 // ---- record walking (must mirror lfmq/models/engine.py exactly) ----------
@@ -507,6 +507,13 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     ENG_PROF(3);
 
     // ---------------- P4: gather attention output ----------------------
+    // intra-CU barrier FIRST: the gather stages into unionb, overwriting
+    // qkvbuf, while this CU's other waves may still be inside attention
+    // units READING qkvbuf (this race was caught as run-to-run engine
+    // nondeterminism, max|run1-run2| up to 1e-2).
+    __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
+    ++gseq;
+    ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x4f, lane, &dbg_sync));
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
     ENG_CHECK(gather_share(g_attn, a.hq * D / 2, e_attn, (u32*)unionb, wave,
                            ctl, tmo, 0x50, lane, &dbg_sweep));

@@ -27,10 +27,10 @@ struct EngineArgs {
   unsigned long long* prof;          // optional [n_layers*8] s_memtime, CU0
 };
 
-// record byte size in the weight stream (padded to 16 B); must match
-// lfmq/models/engine.py::_rec_bytes
+// record byte size in the weight stream v2 (padded to 16 B); must match
+// lfmq/models/engine.py::_rec_bytes: lane-major nibbles + fp32 (s,sz) pairs
 inline int lfmq_engine_rec_bytes(int k, int G) {
-  return (k / 2 + G * 2 + (G + 1) / 2 + 15) & ~15;
+  return (k / 2 + G * 8 + 15) & ~15;
 }
 
 extern "C" hipError_t lfmq_engine_step(const EngineArgs* args, int d,

@@ -183,37 +183,39 @@ __device__ __forceinline__ void glds_fill_1k(const char* gsrc_lane,
                                    16, 0, /*aux nt*/ 2);
 }
 
-// A chunk is CH=4 k-iterations (2048 values); CH must divide k/512 for
-// every op (host asserts k % 2048 == 0) so the bodies are BRANCH-FREE —
-// a runtime `if (j >= n) break` inside #pragma unroll defeats the unroll
-// and serializes every LDS load on its own latency.  Chunks are DOUBLE-
-// BUFFERED (load c+1 while computing c): at 1 wave/SIMD there is no other
-// wave to hide the ~64-cycle LDS latency, so the wave's own ILP must —
-// the isolated microbench measured 0.78 us/record chunk-serial.
+// ---- record dequant-dot, stream format v2 -------------------------------
+// Record layout (builder: lfmq/models/engine.py):
+//   [k/2 B nibbles, LANE-MAJOR dwords: dword for (iter i, lane l) at
+//    (l*iters + i)*4 — one b128 per lane fetches a whole 4-iter chunk]
+//   [G x 8 B fp32 (s, s*(z+128)) pairs — one broadcast b64 per iter
+//    replaces the v1 u16+u8 scalar reads and their convert math]
+//   [pad to 16 B]
+// A chunk is 4 k-iterations (2048 values); 4 must divide k/512 (host
+// asserts k % 2048 == 0) so chunk bodies are BRANCH-FREE (a runtime break
+// inside #pragma unroll defeats the unroll).  Chunks are DOUBLE-BUFFERED:
+// at 1 wave/SIMD nothing else hides the ~64-cycle LDS latency, so the
+// wave's own ILP must (isolated microbench: 0.78 us/record chunk-serial,
+// dominated by per-LDS-instruction cost — hence format v2's fewer, wider
+// reads).
+typedef float f32x2 __attribute__((ext_vector_type(2)));
+
 struct RecChunk {
-  u32 w[4];
+  u32x4 w;        // 4 iters of nibble dwords for this lane
   u32x4 xq[4];
-  float s[4], sz[4];
+  f32x2 ssz[4];   // (s, s*(z+128)) per iter
 };
 
 __device__ __forceinline__ void chunk_load(RecChunk& ch, int c0,
-                                           const u32* wbase,
-                                           const unsigned short* sbase,
-                                           const unsigned char* zbase,
+                                           const char* lds_rec, int iters,
                                            const char* lds_src, int log2g,
                                            int lane) {
+  ch.w = *(const u32x4*)(lds_rec + (lane * iters + c0) * 4);
+  const f32x2* sbase = (const f32x2*)(lds_rec + iters * 256);
 #pragma unroll
   for (int j = 0; j < 4; ++j) {
-    const int i = c0 + j;
-    const int k0 = i * 512 + lane * 8;
-    ch.w[j] = wbase[i * 64 + lane];
+    const int k0 = (c0 + j) * 512 + lane * 8;
     ch.xq[j] = *(const u32x4*)(lds_src + k0 * 2);
-    const int gi = k0 >> log2g;
-    const u32 sraw = sbase[gi];
-    const u32 zb = zbase[gi >> 1];
-    const u32 z = (gi & 1) ? (zb >> 4) : (zb & 0xFu);
-    ch.s[j] = bf16_bits_to_f32(sraw);
-    ch.sz[j] = ch.s[j] * (float)(z + 128u);
+    ch.ssz[j] = sbase[k0 >> log2g];
   }
 }
 
@@ -230,30 +232,28 @@ __device__ __forceinline__ void chunk_dot(const RecChunk& ch, float& acc) {
     xs = dot2_bf16(0x3F803F80u, ch.xq[j][1], xs);
     xs = dot2_bf16(0x3F803F80u, ch.xq[j][2], xs);
     xs = dot2_bf16(0x3F803F80u, ch.xq[j][3], xs);
-    acc = fmaf(ch.s[j], cx, acc);
-    acc = fmaf(-ch.sz[j], xs, acc);
+    acc = fmaf(ch.ssz[j][0], cx, acc);
+    acc = fmaf(-ch.ssz[j][1], xs, acc);
   }
 }
 
 __device__ __forceinline__ float record_dot_nored(const char* lds_rec,
                                                   const char* lds_src, int k,
                                                   int G, int log2g, int lane) {
+  (void)G;
   const int iters = k >> 9;
-  const u32* wbase = (const u32*)lds_rec;
-  const unsigned short* sbase = (const unsigned short*)(lds_rec + k / 2);
-  const unsigned char* zbase = (const unsigned char*)(lds_rec + k / 2 + G * 2);
   float acc = 0.f;
   RecChunk A, B;
-  chunk_load(A, 0, wbase, sbase, zbase, lds_src, log2g, lane);
+  chunk_load(A, 0, lds_rec, iters, lds_src, log2g, lane);
   int c0 = 0;
   for (;;) {  // ping-pong; iters may be an odd number of chunks (down: 7)
     if (c0 + 4 < iters)
-      chunk_load(B, c0 + 4, wbase, sbase, zbase, lds_src, log2g, lane);
+      chunk_load(B, c0 + 4, lds_rec, iters, lds_src, log2g, lane);
     chunk_dot(A, acc);
     c0 += 4;
     if (c0 >= iters) break;
     if (c0 + 4 < iters)
-      chunk_load(A, c0 + 4, wbase, sbase, zbase, lds_src, log2g, lane);
+      chunk_load(A, c0 + 4, lds_rec, iters, lds_src, log2g, lane);
     chunk_dot(B, acc);
     c0 += 4;
     if (c0 >= iters) break;
@@ -267,6 +267,5 @@ __device__ __forceinline__ float record_dot(const char* lds_rec,
   return wave_reduce_sum_dpp(
       record_dot_nored(lds_rec, lds_src, k, G, log2g, lane));
 }
-
 
 }  // namespace lfmq_engine

@@ -316,7 +316,7 @@ int main() {
 namespace eng_ubench {
 constexpr int KREC = 4096;
 constexpr int GREC = 32;                       // k/g, g=128
-constexpr int RB = (KREC / 2 + GREC * 2 + 16 + 15) & ~15;  // 2128
+constexpr int RB = (KREC / 2 + GREC * 8 + 15) & ~15;  // format v2
 constexpr int NREC = 36;
 
 template <int ABLATE>
@@ -343,8 +343,12 @@ __global__ __launch_bounds__(TPB) void record_ubench_kernel(
     float vg = 0.f, h0 = 0.f;
     for (int r = 0; r < NREC; ++r) {
       if ((r >> 2) % 3 != wave) continue;
-      float v = ABLATE >= 2 ? lfmq_engine::record_dot_nored(recs + r * RB, xbuf, KREC, GREC, 7, lane)
-                      : lfmq_engine::record_dot(recs + r * RB, xbuf, KREC, GREC, 7, lane);
+      float v;
+      if (ABLATE >= 2)
+        v = lfmq_engine::record_dot_nored(recs + r * RB, xbuf, KREC, GREC, 7,
+                                          lane);
+      else
+        v = lfmq_engine::record_dot(recs + r * RB, xbuf, KREC, GREC, 7, lane);
       if (ABLATE < 3) {
         switch (r & 3) {
           case 0: vg = v; break;

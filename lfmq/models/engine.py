@@ -37,7 +37,9 @@ AT = 256  # attention chunk (engine kernel constant)
 
 
 def _rec_bytes(k: int, G: int) -> int:
-    return (k // 2 + G * 2 + (G + 1) // 2 + 15) & ~15
+    """Stream-record bytes, format v2: lane-major nibble dwords + fp32
+    (s, s*(z+128)) pairs (csrc/engine_common.h record_dot)."""
+    return (k // 2 + G * 8 + 15) & ~15
 
 
 def _is_pow2(x: int) -> bool:
@@ -136,17 +138,27 @@ class DecodeEngine:
         ring = self.nslot * SLOT
 
         def op_records(q, interleave_half: bool):
-            """[NCU, npc, rec_b] uint8 record block for one QLinear."""
+            """[NCU, npc, rec_b] uint8 record block for one QLinear
+            (format v2: lane-major nibbles, fp32 (s, sz) pairs)."""
             n, k = q.out_features, q.in_features
             G = k // self.g
             rb = _rec_bytes(k, G)
-            qw = q.qweight.view(torch.uint8).reshape(n, k // 2)
-            sc = q.scales.view(torch.uint8).reshape(n, G * 2)
-            zz = q.qzeros.view(torch.uint8).reshape(n, -1)[:, :(G + 1) // 2]
+            iters = k // 512
+            # nibbles: dword for (iter i, lane l) at (l*iters + i):
+            # transpose the row's [iters, 64] dword grid to [64, iters]
+            qw = q.qweight.reshape(n, iters, 64).transpose(1, 2).contiguous()
+            nib = qw.view(torch.uint8).reshape(n, k // 2)
+            # fp32 (s, s*(z+128)) pairs — sz computed exactly as the v1
+            # kernel did (bf16 s -> fp32, times the int zero bias)
+            s = q.scales.float()
+            gidx = torch.arange(G, device=q.qzeros.device)
+            z = (q.qzeros[:, gidx // 8] >> (4 * (gidx % 8))) & 0xF
+            sz = s * (z.float() + 128.0)
+            pairs = torch.stack([s, sz], dim=-1).reshape(n, 2 * G)
+            pb = pairs.view(torch.uint8).reshape(n, G * 8)
             rec = torch.zeros(n, rb, dtype=torch.uint8, device=dev)
-            rec[:, :k // 2] = qw
-            rec[:, k // 2:k // 2 + G * 2] = sc
-            rec[:, k // 2 + G * 2:k // 2 + G * 2 + (G + 1) // 2] = zz
+            rec[:, :k // 2] = nib
+            rec[:, k // 2:k // 2 + G * 8] = pb
             if interleave_half:
                 # gate_up: rows [g_0..g_{I-1}, u_0..u_{I-1}] -> per-CU
                 # interleaved [g_j, u_j, ...]

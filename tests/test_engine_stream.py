@@ -74,6 +74,7 @@ def test_stream_layout_roundtrip():
         # never straddles the ring wrap
         assert off % ring + rb <= ring
         assert off % 16 == 0
+        iters = k // 512
         # spot-check CU 0 and CU 17
         for cu in (0, 17):
             if inter_leaved:
@@ -82,13 +83,17 @@ def test_stream_layout_roundtrip():
             else:
                 row = cu * (op.out_features // NCU) + r
             got = stream[cu, off:off + rb]
-            qw = op.qweight.view(torch.uint8).reshape(op.out_features, k // 2)
-            sc = op.scales.view(torch.uint8).reshape(op.out_features, 2 * G)
-            zz = op.qzeros.view(torch.uint8).reshape(op.out_features, -1)
-            assert torch.equal(got[:k // 2], qw[row])
-            assert torch.equal(got[k // 2:k // 2 + 2 * G], sc[row])
-            assert torch.equal(got[k // 2 + 2 * G:k // 2 + 2 * G + (G + 1) // 2],
-                               zz[row][:(G + 1) // 2])
+            # v2 nibbles: lane-major dwords = transposed [iters, 64] grid
+            qw_t = op.qweight[row].reshape(iters, 64).t().contiguous()
+            assert torch.equal(got[:k // 2],
+                               qw_t.view(torch.uint8).reshape(k // 2))
+            # v2 fp32 (s, s*(z+128)) pairs
+            s = op.scales[row].float()
+            gidx = torch.arange(G)
+            z = (op.qzeros[row][gidx // 8] >> (4 * (gidx % 8))) & 0xF
+            pairs = torch.stack([s, s * (z.float() + 128.0)], -1).reshape(-1)
+            assert torch.equal(
+                got[k // 2:k // 2 + 8 * G].view(torch.float32), pairs)
 
 
 def test_kvcache_stacked_views():

@@ -81,6 +81,9 @@ def main():
             ref, _ = steps(q, cfg, ids, ns, "fused")
             ref2, _ = steps(q, cfg, ids, ns, "nonfused")
             got, prof = steps(q, cfg, ids, ns, "engine")
+            got2, _ = steps(q, cfg, ids, ns, "engine")
+            det = float((got - got2).abs().max())
+            print(f"   engine determinism: max|run1-run2|={det:.6f}")
             eng_abs = [float((got[i] - ref[i]).abs().max()) for i in range(ns)]
             yard = [float((ref2[i] - ref[i]).abs().max()) for i in range(ns)]
             scale = [float(ref[i].abs().max()) for i in range(ns)]
