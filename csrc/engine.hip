@@ -234,7 +234,12 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
                      : "=&v"(busy) : "v"(b_off) : "memory");
         if (busy) asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
       }
-      if (s >= LAG - 1) ENG_PUB((LAG - 1) * 16, (unsigned)(s - LAG + 2));
+      if (a.debug & 1) {
+        loader_publish<0>(f_off, (unsigned)(s + 1));
+        last_pub = (unsigned)(s + 1);
+      } else if (s >= LAG - 1) {
+        ENG_PUB((LAG - 1) * 16, (unsigned)(s - LAG + 2));
+      }
     }
     ENG_PUB(0, (unsigned)total);
 #undef ENG_PUB
@@ -266,6 +271,20 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
   unsigned phase_want = 0;
   bool ok = true;
   unsigned dbg_fill = 0, dbg_sweep = 0, dbg_sync = 0;
+#define ENG_DBG_BARRIER()                                                  \
+  do {                                                                     \
+    if (a.debug & 2) {                                                     \
+      __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL,             \
+                             LFMQ_WG_SCOPE);                               \
+      ++gseq;                                                              \
+      ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x7f, lane, &dbg_sync));         \
+    }                                                                      \
+  } while (0)
+#define ENG_DBG_FENCE()                                                    \
+  do {                                                                     \
+    if (a.debug & 4)                                                       \
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");                   \
+  } while (0)
 
 #define ENG_CHECK(x) do { if (!(x)) return; } while (0)
 #define ENG_PROF(slot_)                                                   \
@@ -329,6 +348,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       }
       lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
     }
+    ENG_DBG_BARRIER();
     ENG_PROF(1);
 
     // ---------------- P2: gather qkv into union ------------------------
@@ -338,6 +358,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x32, lane, &dbg_sync));
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
+    ENG_DBG_FENCE();
     ENG_PROF(2);
 
     // ---------------- P3: attention units ------------------------------
@@ -504,6 +525,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
         }
       }
     }
+    ENG_DBG_BARRIER();
     ENG_PROF(3);
 
     // ---------------- P4: gather attention output ----------------------
@@ -520,6 +542,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x51, lane, &dbg_sync));
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
+    ENG_DBG_FENCE();
     ENG_PROF(4);
 
     // ---------------- P5: o records (src = attnbuf) --------------------
@@ -548,6 +571,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
                            0x53, lane));
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x54, lane, &dbg_sync));
+    ENG_DBG_FENCE();
     {
       u32* x32 = (u32*)xbuf;
       const u32* add32 = (const u32*)ascr;
@@ -607,6 +631,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       }
       lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
     }
+    ENG_DBG_BARRIER();
     ENG_PROF(6);
 
     // ---------------- P8: gather h ------------------------------------
@@ -615,6 +640,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
                            tmo, 0x61, lane, &dbg_sweep));
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x62, lane, &dbg_sync));
+    ENG_DBG_FENCE();
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
 
     // ---------------- P9: down records (src = hbuf) --------------------
@@ -637,6 +663,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       }
       lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
     }
+    ENG_DBG_BARRIER();
     ENG_PROF(7);
 
     // ---------------- P10: gather down, resid-add, inv1 of next layer --
@@ -645,6 +672,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
                            ctl, tmo, 0x64, lane, &dbg_sweep));
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x65, lane, &dbg_sync));
+    ENG_DBG_FENCE();
     {
       u32* x32 = (u32*)xbuf;
       const u32* add32 = (const u32*)ascr;
@@ -699,6 +727,8 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
   }
 #undef ENG_CHECK
 #undef ENG_PROF
+#undef ENG_DBG_BARRIER
+#undef ENG_DBG_FENCE
 }
 
 }  // namespace

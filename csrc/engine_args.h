@@ -25,6 +25,11 @@ struct EngineArgs {
   void *g_qkv, *g_attn, *g_o, *g_h, *g_down, *g_scr;
   void* tmo;                         // u32 give-up code (0 = healthy)
   unsigned long long* prof;          // optional [n_layers*8] s_memtime, CU0
+  unsigned debug;                    // race-bisect flags (LFMQ_ENGINE_DEBUG):
+                                     // 1 = conservative loader (vmcnt(0)
+                                     //     publish per slot), 2 = arrival
+                                     //     barrier around every phase,
+                                     // 4 = agent-acquire fence after sweeps
 };
 
 // record byte size in the weight stream v2 (padded to 16 B); must match

@@ -412,7 +412,8 @@ void engine_step(torch::Tensor stream_t, long stream_per_cu, long nslot_ring,
                  torch::Tensor final_w, torch::Tensor xout,
                  torch::Tensor g_qkv, torch::Tensor g_attn, torch::Tensor g_o,
                  torch::Tensor g_h, torch::Tensor g_down, torch::Tensor g_scr,
-                 torch::Tensor tmo, c10::optional<torch::Tensor> prof) {
+                 torch::Tensor tmo, c10::optional<torch::Tensor> prof,
+                 long debug) {
   check_t(kc, torch::kBFloat16, "k_cache");
   check_t(vc, torch::kBFloat16, "v_cache");
   check_t(embed, torch::kBFloat16, "embed");
@@ -460,6 +461,7 @@ void engine_step(torch::Tensor stream_t, long stream_per_cu, long nslot_ring,
   a.prof = prof.has_value()
                ? (unsigned long long*)prof->data_ptr()
                : nullptr;
+  a.debug = (unsigned)debug;
   CHECK_HIP(lfmq_engine_step(&a, d, stream()));
 }
 

@@ -125,6 +125,7 @@ class DecodeEngine:
         self.xout = torch.zeros(1, 1, hid, dtype=torch.bfloat16, device=dev)
         self.inv_freq = F._inv_freq(d, cfg.rope_theta, dev)
         self.attn_scale = 1.0 / (d ** 0.5)
+        self.debug = int(os.environ.get("LFMQ_ENGINE_DEBUG", "0"))
         self.prof: Optional[torch.Tensor] = None
         if os.environ.get("LFMQ_ENGINE_PROF"):
             n_slots = (self.per_cu + SLOT - 1) // SLOT
@@ -229,7 +230,8 @@ class DecodeEngine:
             self.inv_freq, self.cache.k_all, self.cache.v_all, state,
             input_ids.reshape(-1), self.model.embed_tokens.weight,
             self.model.norm.weight, self.xout, self.g_qkv, self.g_attn,
-            self.g_o, self.g_h, self.g_down, self.g_scr, self.tmo, self.prof)
+            self.g_o, self.g_h, self.g_down, self.g_scr, self.tmo, self.prof,
+            self.debug)
         return self.xout
 
     def check_health(self):
