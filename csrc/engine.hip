@@ -117,10 +117,19 @@ __device__ __forceinline__ bool gather_share(gu64* g, int cnt, unsigned epoch,
                                              u32* lds_dst, int wave, Ctl* ctl,
                                              gu32* tmo, unsigned code,
                                              int lane,
-                                             unsigned* passes = nullptr) {
+                                             unsigned* passes = nullptr,
+                                             unsigned dbg = 0) {
   const int b = cnt * wave / 3, e = cnt * (wave + 1) / 3;
   if (!sweep_stage(g + b, e - b, epoch, lds_dst + b, tmo, code, lane, passes))
     return false;
+  if (dbg & 16) {  // re-read and compare the staged third (stale staging?)
+    for (int idx = lane; idx < e - b; idx += LFMQ_WAVE) {
+      const unsigned long long x = __hip_atomic_load(g + b + idx,
+                                                     LFMQ_RLX_AGENT);
+      if ((u32)x != lds_dst[b + idx])
+        __hip_atomic_store(tmo + 6, 0xE8u, LFMQ_RLX_AGENT);
+    }
+  }
   __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
   return true;
 }
@@ -128,6 +137,19 @@ __device__ __forceinline__ bool wait_gath(Ctl* ctl, unsigned gseq, gu32* tmo,
                                           unsigned code, int lane,
                                           unsigned* acc = nullptr) {
   return lds_wait_ge(&ctl->gath, 3 * gseq, tmo, code, lane, acc);
+}
+
+// debug bit 8: read each record twice; a mismatch means the ring bytes
+// changed under the read (overwrite race) or were not yet filled.
+__device__ __forceinline__ float record_dot_checked(
+    const char* lds_rec, const char* lds_src, int k, int G, int log2g,
+    int lane, unsigned dbg, gu32* tmo, unsigned code) {
+  const float v = record_dot(lds_rec, lds_src, k, G, log2g, lane);
+  if (dbg & 8) {
+    const float v2 = record_dot(lds_rec, lds_src, k, G, log2g, lane);
+    if (lane == 0 && v != v2) __hip_atomic_store(tmo + 5, code, LFMQ_RLX_AGENT);
+  }
+  return v;
 }
 
 // single-granule bounded wait (combine path)
@@ -334,9 +356,9 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
         const u64 at = cursor_next(cur, rb_qkv, ring_bytes);
         if ((r >> 1) % 3 != wave) continue;
         ENG_CHECK(wait_fill(ctl, at + rb_qkv, tmo, 0x30, lane, &dbg_fill));
-        const float v = inv * record_dot(ring + (u32)(at % ring_bytes),
-                                         (const char*)xbuf, a.hidden, G1,
-                                         a.log2g, lane);
+        const float v = inv * record_dot_checked(
+            ring + (u32)(at % ring_bytes), (const char*)xbuf, a.hidden, G1,
+            a.log2g, lane, a.debug, tmo, 0xE1);
         if (r & 1) {
           if (lane == 0)
             store_granule(g_qkv + (cu * npc_qkv + r) / 2, e_qkv,
@@ -354,7 +376,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     // ---------------- P2: gather qkv into union ------------------------
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
     ENG_CHECK(gather_share(g_qkv, a.nqkv / 2, e_qkv, (u32*)unionb, wave, ctl,
-                           tmo, 0x31, lane, &dbg_sweep));
+                           tmo, 0x31, lane, &dbg_sweep, a.debug));
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x32, lane, &dbg_sync));
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
@@ -538,7 +560,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x4f, lane, &dbg_sync));
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
     ENG_CHECK(gather_share(g_attn, a.hq * D / 2, e_attn, (u32*)unionb, wave,
-                           ctl, tmo, 0x50, lane, &dbg_sweep));
+                           ctl, tmo, 0x50, lane, &dbg_sweep, a.debug));
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x51, lane, &dbg_sync));
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 0u);
@@ -552,8 +574,9 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
         const u64 at = cursor_next(cur, rb_o, ring_bytes);
         if ((r >> 1) % 3 != wave) continue;
         ENG_CHECK(wait_fill(ctl, at + rb_o, tmo, 0x52, lane, &dbg_fill));
-        const float v = record_dot(ring + (u32)(at % ring_bytes), unionb,
-                                   a.hq * D, Go, a.log2g, lane);
+        const float v = record_dot_checked(ring + (u32)(at % ring_bytes),
+                                           unionb, a.hq * D, Go, a.log2g,
+                                           lane, a.debug, tmo, 0xE2);
         if (r & 1) {
           if (lane == 0)
             store_granule(g_o + (cu * npc_o + r) / 2, e_o,
@@ -568,7 +591,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     // ---------------- P6: gather o, residual-add into xbuf, inv2 -------
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
     ENG_CHECK(gather_share(g_o, a.hidden / 2, e_o, (u32*)ascr, wave, ctl, tmo,
-                           0x53, lane));
+                           0x53, lane, &dbg_sweep, a.debug));
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x54, lane, &dbg_sync));
     ENG_DBG_FENCE();
@@ -613,9 +636,9 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
         const u64 at = cursor_next(cur, rb_gu, ring_bytes);
         if ((r >> 2) % 3 != wave) continue;
         ENG_CHECK(wait_fill(ctl, at + rb_gu, tmo, 0x60, lane, &dbg_fill));
-        const float v = inv * record_dot(ring + (u32)(at % ring_bytes),
-                                         (const char*)xbuf, a.hidden, G1,
-                                         a.log2g, lane);
+        const float v = inv * record_dot_checked(
+            ring + (u32)(at % ring_bytes), (const char*)xbuf, a.hidden, G1,
+            a.log2g, lane, a.debug, tmo, 0xE3);
         switch (r & 3) {
           case 0: vg = v; break;
           case 1: h0 = (vg / (1.f + __expf(-vg))) * v; break;
@@ -637,7 +660,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     // ---------------- P8: gather h ------------------------------------
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
     ENG_CHECK(gather_share(g_h, a.inter / 2, e_h, (u32*)unionb, wave, ctl,
-                           tmo, 0x61, lane, &dbg_sweep));
+                           tmo, 0x61, lane, &dbg_sweep, a.debug));
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x62, lane, &dbg_sync));
     ENG_DBG_FENCE();
@@ -650,8 +673,9 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
         const u64 at = cursor_next(cur, rb_dn, ring_bytes);
         if ((r >> 1) % 3 != wave) continue;
         ENG_CHECK(wait_fill(ctl, at + rb_dn, tmo, 0x63, lane, &dbg_fill));
-        const float v = record_dot(ring + (u32)(at % ring_bytes), unionb,
-                                   a.inter, G2, a.log2g, lane);
+        const float v = record_dot_checked(ring + (u32)(at % ring_bytes),
+                                           unionb, a.inter, G2, a.log2g,
+                                           lane, a.debug, tmo, 0xE4);
         if (r & 1) {
           if (lane == 0)
             store_granule(g_down + (cu * npc_dn + r) / 2, e_down,
@@ -669,7 +693,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     // ---------------- P10: gather down, resid-add, inv1 of next layer --
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
     ENG_CHECK(gather_share(g_down, a.hidden / 2, e_down, (u32*)ascr, wave,
-                           ctl, tmo, 0x64, lane, &dbg_sweep));
+                           ctl, tmo, 0x64, lane, &dbg_sweep, a.debug));
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x65, lane, &dbg_sync));
     ENG_DBG_FENCE();
