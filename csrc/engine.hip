@@ -122,12 +122,16 @@ __device__ __forceinline__ bool gather_share(gu64* g, int cnt, unsigned epoch,
   const int b = cnt * wave / 3, e = cnt * (wave + 1) / 3;
   if (!sweep_stage(g + b, e - b, epoch, lds_dst + b, tmo, code, lane, passes))
     return false;
-  if (dbg & 16) {  // re-read and compare the staged third (stale staging?)
+  if (dbg & 16) {  // re-read twice: active writer vs stale sweep read
     for (int idx = lane; idx < e - b; idx += LFMQ_WAVE) {
-      const unsigned long long x = __hip_atomic_load(g + b + idx,
-                                                     LFMQ_RLX_AGENT);
-      if ((u32)x != lds_dst[b + idx])
-        __hip_atomic_store(tmo + 6, 0xE8u, LFMQ_RLX_AGENT);
+      const unsigned long long x1 = __hip_atomic_load(g + b + idx,
+                                                      LFMQ_RLX_AGENT);
+      const unsigned long long x2 = __hip_atomic_load(g + b + idx,
+                                                      LFMQ_RLX_AGENT);
+      if (x1 != x2)
+        __hip_atomic_store(tmo + 6, 0xE9u, LFMQ_RLX_AGENT);  // 2nd writer!
+      else if ((u32)x1 != lds_dst[b + idx])
+        __hip_atomic_store(tmo + 6, 0xEAu, LFMQ_RLX_AGENT);  // stale sweep
     }
   }
   __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
@@ -139,15 +143,34 @@ __device__ __forceinline__ bool wait_gath(Ctl* ctl, unsigned gseq, gu32* tmo,
   return lds_wait_ge(&ctl->gath, 3 * gseq, tmo, code, lane, acc);
 }
 
-// debug bit 8: read each record twice; a mismatch means the ring bytes
-// changed under the read (overwrite race) or were not yet filled.
+// debug bit 8: read each record twice and byte-checksum it twice.
+// checksum mismatch  -> ring bytes changed under us (overwrite race),
+//   code 0xB0 | phase in tmo[5]
+// checksums equal but dots differ -> compute nondeterminism (hazard in
+//   the dot/reduce), code 0xC0 | phase in tmo[5]
+__device__ __forceinline__ u32 rec_cksum(const char* lds_rec, int bytes,
+                                         int lane) {
+  const u32* p = (const u32*)lds_rec;
+  u32 s = 0;
+  for (int i = lane; i < bytes / 4; i += LFMQ_WAVE) s += p[i];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) s += __shfl_xor(s, off, 64);
+  return s;
+}
 __device__ __forceinline__ float record_dot_checked(
     const char* lds_rec, const char* lds_src, int k, int G, int log2g,
     int lane, unsigned dbg, gu32* tmo, unsigned code) {
   const float v = record_dot(lds_rec, lds_src, k, G, log2g, lane);
   if (dbg & 8) {
+    const u32 ck1 = rec_cksum(lds_rec, k / 2 + G * 8, lane);
     const float v2 = record_dot(lds_rec, lds_src, k, G, log2g, lane);
-    if (lane == 0 && v != v2) __hip_atomic_store(tmo + 5, code, LFMQ_RLX_AGENT);
+    const u32 ck2 = rec_cksum(lds_rec, k / 2 + G * 8, lane);
+    if (lane == 0) {
+      if (ck1 != ck2)
+        __hip_atomic_store(tmo + 5, 0xB00u | (code & 0xF), LFMQ_RLX_AGENT);
+      else if (v != v2)
+        __hip_atomic_store(tmo + 5, 0xC00u | (code & 0xF), LFMQ_RLX_AGENT);
+    }
   }
   return v;
 }
