@@ -134,7 +134,8 @@ __device__ __forceinline__ bool gather_share(gu64* g, int cnt, unsigned epoch,
         __hip_atomic_store(tmo + 6, 0xEAu, LFMQ_RLX_AGENT);  // stale sweep
     }
   }
-  __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
+  if (lane == 0)
+      __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
   return true;
 }
 __device__ __forceinline__ bool wait_gath(Ctl* ctl, unsigned gseq, gu32* tmo,
@@ -160,16 +161,33 @@ __device__ __forceinline__ u32 rec_cksum(const char* lds_rec, int bytes,
 __device__ __forceinline__ float record_dot_checked(
     const char* lds_rec, const char* lds_src, int k, int G, int log2g,
     int lane, unsigned dbg, gu32* tmo, unsigned code) {
+  if (!(dbg & 8)) return record_dot(lds_rec, lds_src, k, G, log2g, lane);
+  // bracket BOTH dots with ring-record AND x-source checksums:
+  //   0xB.. ring bytes changed, 0xA.. x bytes changed, 0xC.. checksums
+  //   stable but dots differ (true compute nondeterminism)
+  // 16-block partial checksums over x locate WHERE instability lands
+  u32 xb1[16], xb2[16];
+  const int xblk = (k * 2) / 16;
+  const u32 r1 = rec_cksum(lds_rec, k / 2 + G * 8, lane);
+#pragma unroll
+  for (int b = 0; b < 16; ++b) xb1[b] = rec_cksum(lds_src + b * xblk, xblk, lane);
   const float v = record_dot(lds_rec, lds_src, k, G, log2g, lane);
-  if (dbg & 8) {
-    const u32 ck1 = rec_cksum(lds_rec, k / 2 + G * 8, lane);
-    const float v2 = record_dot(lds_rec, lds_src, k, G, log2g, lane);
-    const u32 ck2 = rec_cksum(lds_rec, k / 2 + G * 8, lane);
-    if (lane == 0) {
-      if (ck1 != ck2)
-        __hip_atomic_store(tmo + 5, 0xB00u | (code & 0xF), LFMQ_RLX_AGENT);
-      else if (v != v2)
-        __hip_atomic_store(tmo + 5, 0xC00u | (code & 0xF), LFMQ_RLX_AGENT);
+  const u32 r2 = rec_cksum(lds_rec, k / 2 + G * 8, lane);
+  const float v2 = record_dot(lds_rec, lds_src, k, G, log2g, lane);
+  const u32 r3 = rec_cksum(lds_rec, k / 2 + G * 8, lane);
+#pragma unroll
+  for (int b = 0; b < 16; ++b) xb2[b] = rec_cksum(lds_src + b * xblk, xblk, lane);
+  u32 bm = 0;
+#pragma unroll
+  for (int b = 0; b < 16; ++b) bm |= (xb1[b] != xb2[b]) ? (1u << b) : 0u;
+  if (lane == 0) {
+    if (r1 != r2 || r2 != r3)
+      __hip_atomic_store(tmo + 5, 0xB00u | (code & 0xF), LFMQ_RLX_AGENT);
+    else if (bm) {
+      __hip_atomic_store(tmo + 5, 0xA00u | (code & 0xF), LFMQ_RLX_AGENT);
+      __hip_atomic_fetch_or(tmo + 7, bm, LFMQ_RLX_AGENT);
+    } else if (v != v2) {
+      __hip_atomic_store(tmo + 5, 0xC00u | (code & 0xF), LFMQ_RLX_AGENT);
     }
   }
   return v;
@@ -319,8 +337,9 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
 #define ENG_DBG_BARRIER()                                                  \
   do {                                                                     \
     if (a.debug & 2) {                                                     \
-      __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL,             \
-                             LFMQ_WG_SCOPE);                               \
+      if (lane == 0)                                                       \
+        __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL,           \
+                               LFMQ_WG_SCOPE);                             \
       ++gseq;                                                              \
       ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x7f, lane, &dbg_sync));         \
     }                                                                      \
@@ -353,7 +372,8 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     }
     ss = wave_reduce_sum(ss);
     if (lane == 0) ctl->ss_part[wave] = ss;
-    __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
+    if (lane == 0)
+      __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x20, lane, &dbg_sync));
     if (wave == 0) {
@@ -578,7 +598,8 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     // qkvbuf, while this CU's other waves may still be inside attention
     // units READING qkvbuf (this race was caught as run-to-run engine
     // nondeterminism, max|run1-run2| up to 1e-2).
-    __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
+    if (lane == 0)
+      __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x4f, lane, &dbg_sync));
     if (wave == 0) lds_flag_store_rlx(&ctl->busy, 1u);
@@ -636,6 +657,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       }
       ss = wave_reduce_sum(ss);
       if (lane == 0) ctl->ss_part[wave] = ss;
+      if (lane == 0)
       __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
       ++gseq;
       ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x55, lane, &dbg_sync));
@@ -738,6 +760,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       }
       ss = wave_reduce_sum(ss);
       if (lane == 0) ctl->ss_part[wave] = ss;
+      if (lane == 0)
       __hip_atomic_fetch_add(&ctl->gath, 1u, __ATOMIC_ACQ_REL, LFMQ_WG_SCOPE);
       ++gseq;
       ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x66, lane, &dbg_sync));

@@ -264,7 +264,11 @@ __device__ __forceinline__ float record_dot_nored(const char* lds_rec,
 __device__ __forceinline__ float record_dot(const char* lds_rec,
                                             const char* lds_src, int k, int G,
                                             int log2g, int lane) {
-  return wave_reduce_sum_dpp(
+  // shfl (ds_bpermute) reduce: the DPP ladder measured faster in isolation
+  // but produced timing-dependent WRONG sums under full engine load
+  // (race_hunt detector 0xC0: identical input, different dot) — a DPP
+  // read-after-VALU hazard the compiler does not pad reliably here.
+  return wave_reduce_sum(
       record_dot_nored(lds_rec, lds_src, k, G, log2g, lane));
 }
 

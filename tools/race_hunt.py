@@ -42,7 +42,7 @@ def main():
             state += torch.tensor([1, 1, 1], dtype=torch.int32, device="cuda")
             cache.seq_len += 1
         t = q._engine.tmo.cpu().tolist()
-        print(f"  tmo: giveup={t[0]:#x} dblread={t[5]:#x} restage={t[6]:#x}")
+        print(f"  tmo: giveup={t[0]:#x} dblread={t[5]:#x} restage={t[6]:#x} xblocks={t[7]:#06x}")
         q._engine.tmo.zero_()
         return torch.cat(xo)
 
