@@ -55,19 +55,23 @@ using namespace lfmq_engine;
 namespace {
 
 constexpr int SLOT = 16384;
-constexpr int LAG = 4;   // publish slot s-LAG+1 after issuing slot s
-constexpr int AT = 256;  // attention tokens per unit (one wave per unit)
+constexpr int LAG = 4;     // publish slot s-LAG+1 after issuing slot s
+constexpr int AT = 256;    // attention tokens per unit (one wave per unit)
 constexpr int NCU = 256;
+// 8 waves per CU: 1 loader + 7 consumers (2 waves/SIMD).  At 1 wave/SIMD
+// nothing hides a consumer's LDS/VALU latency chains; the second wave per
+// SIMD roughly doubles record throughput and halves gather sweeps.
+constexpr int NCONS = 7;
 constexpr u32 ONE2_BF16 = 0x3F803F80u;
 
 struct Ctl {
   unsigned filled;       // loader: slots published
-  unsigned consumed[3];  // consumer waves: slots fully passed
+  unsigned consumed[NCONS];  // consumer waves: slots fully passed
   unsigned gath;         // gather arrivals (monotonic, +3 per event)
   unsigned phase;        // lead-published phase (monotonic)
   unsigned busy;         // gather window hint for loader thinning
   float inv_cur;         // norm scalar of the current phase
-  float ss_part[3];      // per-wave sum-of-squares partials
+  float ss_part[NCONS];  // per-wave sum-of-squares partials
 };
 
 // record byte size v2 (padded to 16) — device twin of lfmq_engine_rec_bytes
@@ -119,7 +123,7 @@ __device__ __forceinline__ bool gather_share(gu64* g, int cnt, unsigned epoch,
                                              int lane,
                                              unsigned* passes = nullptr,
                                              unsigned dbg = 0) {
-  const int b = cnt * wave / 3, e = cnt * (wave + 1) / 3;
+  const int b = cnt * wave / NCONS, e = cnt * (wave + 1) / NCONS;
   if (!sweep_stage(g + b, e - b, epoch, lds_dst + b, tmo, code, lane, passes))
     return false;
   if (dbg & 16) {  // re-read twice: active writer vs stale sweep read
@@ -141,7 +145,7 @@ __device__ __forceinline__ bool gather_share(gu64* g, int cnt, unsigned epoch,
 __device__ __forceinline__ bool wait_gath(Ctl* ctl, unsigned gseq, gu32* tmo,
                                           unsigned code, int lane,
                                           unsigned* acc = nullptr) {
-  return lds_wait_ge(&ctl->gath, 3 * gseq, tmo, code, lane, acc);
+  return lds_wait_ge(&ctl->gath, NCONS * gseq, tmo, code, lane, acc);
 }
 
 // debug bit 8: read each record twice and byte-checksum it twice.
@@ -205,7 +209,7 @@ __device__ __forceinline__ u32 gran_wait(gu64* g, unsigned epoch, gu32* tmo,
 
 // ============================================================ main kernel
 template <int D>
-__global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
+__global__ __launch_bounds__((NCONS + 1) * 64, 1) void lfmq_engine_kernel(EngineArgs a) {
   extern __shared__ char smem[];
   const int cu = blockIdx.x;
   const int wave = threadIdx.x / LFMQ_WAVE;
@@ -218,10 +222,11 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
   const int union_bytes =
       max(a.inter * 2, max(a.nqkv * 2, a.hq * D * 2));
   char* ascr = unionb + union_bytes;   // attention scratch / resid staging
-  unsigned short* q_scr = (unsigned short*)ascr;          // [3][D]
-  unsigned short* k_scr = q_scr + 3 * D;                  // [3][D]
-  float* p_scr = (float*)(k_scr + 3 * D);                 // [3][AT]
-  const int ascr_bytes = max(3 * D * 4 + 3 * AT * 4, a.hidden * 2);
+  unsigned short* q_scr = (unsigned short*)ascr;          // [NCONS][D]
+  unsigned short* k_scr = q_scr + NCONS * D;              // [NCONS][D]
+  float* p_scr = (float*)(k_scr + NCONS * D);             // [NCONS][AT]
+  const int ascr_bytes =
+      max(NCONS * D * 4 + NCONS * AT * 4, a.hidden * 2);
   Ctl* ctl = (Ctl*)(ascr + ((ascr_bytes + 63) & ~63));
 
   gu64* g_qkv = (gu64*)a.g_qkv;
@@ -234,7 +239,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
 
   if (threadIdx.x == 0) {
     ctl->filled = 0;
-    ctl->consumed[0] = ctl->consumed[1] = ctl->consumed[2] = 0;
+    for (int w = 0; w < NCONS; ++w) ctl->consumed[w] = 0;
     ctl->gath = 0;
     ctl->phase = 0;
     ctl->busy = 0;
@@ -242,12 +247,11 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
   __syncthreads();  // last convergent barrier; waves diverge below
 
   // ======================================================== loader wave
-  if (wave == 3) {
+  if (wave == NCONS) {
     const char* wg_src = a.stream + (u64)cu * a.stream_per_cu;
     const unsigned f_off = lds_off(&ctl->filled);
-    const unsigned c0 = lds_off(&ctl->consumed[0]);
-    const unsigned c1 = lds_off(&ctl->consumed[1]);
-    const unsigned c2 = lds_off(&ctl->consumed[2]);
+    unsigned c_off[NCONS];
+    for (int w = 0; w < NCONS; ++w) c_off[w] = lds_off(&ctl->consumed[w]);
     const unsigned b_off = lds_off(&ctl->busy);
     const int total = a.n_slots_total;
     unsigned last_pub = 0;
@@ -277,9 +281,13 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
         ENG_PUB(0, (unsigned)s);
         unsigned spins = 0;
         for (;;) {
-          unsigned d0, d1, d2;
-          loader_read3(c0, c1, c2, d0, d1, d2);
-          if ((int)min(d0, min(d1, d2)) >= s - a.nslot_ring + 1) break;
+          unsigned d0, d1, d2, d3, d4, d5, d6;
+          loader_read3(c_off[0], c_off[1], c_off[2], d0, d1, d2);
+          loader_read3(c_off[3], c_off[4], c_off[5], d3, d4, d5);
+          d6 = loader_read1(c_off[6]);
+          const unsigned m = min(min(min(d0, d1), min(d2, d3)),
+                                 min(min(d4, d5), d6));
+          if ((int)m >= s - a.nslot_ring + 1) break;
           if (spin_fail(spins, tmo, 0x10, lane)) return;
         }
         poll_spins_tot += spins;
@@ -363,7 +371,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     const u32* erow = (const u32*)(a.embed + (u64)tk * a.hidden);
     u32* x32 = (u32*)xbuf;
     const int nd = a.hidden / 2;
-    const int b = nd * wave / 3, e = nd * (wave + 1) / 3;
+    const int b = nd * wave / NCONS, e = nd * (wave + 1) / NCONS;
     float ss = 0.f;
     for (int j = b + lane; j < e; j += LFMQ_WAVE) {
       const u32 v = erow[j];
@@ -377,7 +385,8 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
     ++gseq;
     ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x20, lane, &dbg_sync));
     if (wave == 0) {
-      const float sst = ctl->ss_part[0] + ctl->ss_part[1] + ctl->ss_part[2];
+      float sst = 0.f;
+        for (int w = 0; w < NCONS; ++w) sst += ctl->ss_part[w];
       ctl->inv_cur = rsqrtf(sst / (float)a.hidden + a.rms_eps);
       lds_flag_store(&ctl->phase, ++phase_want);
     } else {
@@ -397,7 +406,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       float v0 = 0.f;
       for (int r = 0; r < npc_qkv; ++r) {
         const u64 at = cursor_next(cur, rb_qkv, ring_bytes);
-        if ((r >> 1) % 3 != wave) continue;
+        if ((r >> 1) % NCONS != wave) continue;
         ENG_CHECK(wait_fill(ctl, at + rb_qkv, tmo, 0x30, lane, &dbg_fill));
         const float v = inv * record_dot_checked(
             ring + (u32)(at % ring_bytes), (const char*)xbuf, a.hidden, G1,
@@ -434,7 +443,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       unsigned short* ks = k_scr + wave * D;
       float* ps = p_scr + wave * AT;
       const int total_units = a.hq * nch;
-      for (int u = cu + NCU * wave; u < total_units; u += NCU * 3) {
+      for (int u = cu + NCU * wave; u < total_units; u += NCU * NCONS) {
         const int h = u / nch, c = u % nch;
         const int hk = h / rep;
         const unsigned short* kcache =
@@ -554,7 +563,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       }
       // combines: this wave merges every head whose chunk-0 unit it ran
       if (nch > 1) {
-        for (int u = cu + NCU * wave; u < total_units; u += NCU * 3) {
+        for (int u = cu + NCU * wave; u < total_units; u += NCU * NCONS) {
           if (u % nch != 0) continue;
           const int h = u / nch;
           gu64* hb = g_scr + (u64)h * a.nch_max * (D + 2);
@@ -616,7 +625,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       float v0 = 0.f;
       for (int r = 0; r < npc_o; ++r) {
         const u64 at = cursor_next(cur, rb_o, ring_bytes);
-        if ((r >> 1) % 3 != wave) continue;
+        if ((r >> 1) % NCONS != wave) continue;
         ENG_CHECK(wait_fill(ctl, at + rb_o, tmo, 0x52, lane, &dbg_fill));
         const float v = record_dot_checked(ring + (u32)(at % ring_bytes),
                                            unionb, a.hq * D, Go, a.log2g,
@@ -643,7 +652,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       u32* x32 = (u32*)xbuf;
       const u32* add32 = (const u32*)ascr;
       const int nd = a.hidden / 2;
-      const int b = nd * wave / 3, e = nd * (wave + 1) / 3;
+      const int b = nd * wave / NCONS, e = nd * (wave + 1) / NCONS;
       float ss = 0.f;
       for (int j = b + lane; j < e; j += LFMQ_WAVE) {
         float x0, x1, a0, a1;
@@ -663,7 +672,8 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x55, lane, &dbg_sync));
       if (wave == 0) {
         lds_flag_store_rlx(&ctl->busy, 0u);
-        const float sst = ctl->ss_part[0] + ctl->ss_part[1] + ctl->ss_part[2];
+        float sst = 0.f;
+        for (int w = 0; w < NCONS; ++w) sst += ctl->ss_part[w];
         ctl->inv_cur = rsqrtf(sst / (float)a.hidden + a.rms_eps);
         lds_flag_store(&ctl->phase, ++phase_want);
       } else {
@@ -679,7 +689,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       float vg = 0.f, h0 = 0.f;
       for (int r = 0; r < npc_gu; ++r) {
         const u64 at = cursor_next(cur, rb_gu, ring_bytes);
-        if ((r >> 2) % 3 != wave) continue;
+        if ((r >> 2) % NCONS != wave) continue;
         ENG_CHECK(wait_fill(ctl, at + rb_gu, tmo, 0x60, lane, &dbg_fill));
         const float v = inv * record_dot_checked(
             ring + (u32)(at % ring_bytes), (const char*)xbuf, a.hidden, G1,
@@ -716,7 +726,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       float v0 = 0.f;
       for (int r = 0; r < npc_dn; ++r) {
         const u64 at = cursor_next(cur, rb_dn, ring_bytes);
-        if ((r >> 1) % 3 != wave) continue;
+        if ((r >> 1) % NCONS != wave) continue;
         ENG_CHECK(wait_fill(ctl, at + rb_dn, tmo, 0x63, lane, &dbg_fill));
         const float v = record_dot_checked(ring + (u32)(at % ring_bytes),
                                            unionb, a.inter, G2, a.log2g,
@@ -746,7 +756,7 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       u32* x32 = (u32*)xbuf;
       const u32* add32 = (const u32*)ascr;
       const int nd = a.hidden / 2;
-      const int b = nd * wave / 3, e = nd * (wave + 1) / 3;
+      const int b = nd * wave / NCONS, e = nd * (wave + 1) / NCONS;
       float ss = 0.f;
       for (int j = b + lane; j < e; j += LFMQ_WAVE) {
         float x0, x1, a0, a1;
@@ -766,7 +776,8 @@ __global__ __launch_bounds__(256, 1) void lfmq_engine_kernel(EngineArgs a) {
       ENG_CHECK(wait_gath(ctl, gseq, tmo, 0x66, lane, &dbg_sync));
       if (wave == 0) {
         lds_flag_store_rlx(&ctl->busy, 0u);
-        const float sst = ctl->ss_part[0] + ctl->ss_part[1] + ctl->ss_part[2];
+        float sst = 0.f;
+        for (int w = 0; w < NCONS; ++w) sst += ctl->ss_part[w];
         ctl->inv_cur = rsqrtf(sst / (float)a.hidden + a.rms_eps);
         lds_flag_store(&ctl->phase, ++phase_want);
       } else {
@@ -811,7 +822,8 @@ extern "C" hipError_t lfmq_engine_step(const EngineArgs* args, int d,
   const u32 ring_bytes = (u32)args->nslot_ring * SLOT;
   const int union_bytes =
       std::max(args->inter * 2, std::max(args->nqkv * 2, args->hq * d * 2));
-  const int ascr_bytes = std::max(3 * d * 4 + 3 * AT * 4, args->hidden * 2);
+  const int ascr_bytes =
+      std::max(NCONS * d * 4 + NCONS * AT * 4, args->hidden * 2);
   const size_t lds = ring_bytes + args->hidden * 2 + union_bytes +
                      ((ascr_bytes + 63) & ~63) + sizeof(Ctl) + 64;
   static size_t lds_set = 0;
@@ -822,7 +834,7 @@ extern "C" hipError_t lfmq_engine_step(const EngineArgs* args, int d,
     if (e != hipSuccess) return e;
     lds_set = lds;
   }
-  hipLaunchKernelGGL(lfmq_engine_kernel<128>, dim3(NCU), dim3(256), lds,
-                     stream, *args);
+  hipLaunchKernelGGL(lfmq_engine_kernel<128>, dim3(NCU),
+                     dim3((NCONS + 1) * 64), lds, stream, *args);
   return hipGetLastError();
 }

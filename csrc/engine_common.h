@@ -157,6 +157,13 @@ __device__ __forceinline__ void loader_read3(unsigned off0, unsigned off1,
       : "memory");
 }
 
+__device__ __forceinline__ unsigned loader_read1(unsigned off) {
+  unsigned v;
+  asm volatile("ds_read_b32 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=&v"(v) : "v"(off) : "memory");
+  return v;
+}
+
 // publish a fill-progress value: wait until at most VMLAG fill instructions
 // remain outstanding (i.e. everything older has LANDED in LDS), then
 // ds_write the counter — all inside one asm so hipcc adds no drain.

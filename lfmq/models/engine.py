@@ -33,7 +33,8 @@ from ..ops import hip_ext
 SLOT = 16384
 NCU = 256
 LDS_BYTES = 160 * 1024
-AT = 256  # attention chunk (engine kernel constant)
+AT = 256     # attention chunk (engine kernel constant)
+NCONS = 7    # consumer waves per CU (engine kernel constant)
 
 
 def _rec_bytes(k: int, G: int) -> int:
@@ -88,7 +89,8 @@ def _ring_slots(cfg) -> int:
     union = max(cfg.intermediate_size * 2,
                 (cfg.n_heads + 2 * cfg.n_kv_heads) * cfg.head_dim * 2,
                 cfg.n_heads * cfg.head_dim * 2)
-    ascr = max(3 * cfg.head_dim * 4 + 3 * AT * 4, cfg.hidden_size * 2)
+    ascr = max(NCONS * cfg.head_dim * 4 + NCONS * AT * 4,
+               cfg.hidden_size * 2)
     fixed = cfg.hidden_size * 2 + union + ((ascr + 63) & ~63) + 256
     return min(7, (LDS_BYTES - fixed) // SLOT)
 
