@@ -27,9 +27,11 @@ namespace lfmq_engine {
 typedef __attribute__((address_space(1))) unsigned int gu32;
 typedef __attribute__((address_space(1))) unsigned long long gu64;
 
-// bounded-spin budget: ~2^24 polls x s_sleep(2) ~= seconds of wall —
-// far beyond any healthy step; a trip means a lost producer.
-constexpr unsigned SPIN_MAX = 1u << 24;
+// bounded-spin budget: ~2^22 polls x s_sleep(2) ~= 0.25 s of wall per
+// site — far beyond any healthy step; a trip means a lost producer and
+// must surface FAST (cascading give-ups across layers must not look like
+// a hung GPU to the harness).
+constexpr unsigned SPIN_MAX = 1u << 22;
 
 __device__ __forceinline__ bool spin_fail(unsigned& spins, gu32* tmo,
                                           unsigned code, int lane) {
@@ -70,10 +72,7 @@ __device__ __forceinline__ bool sweep_stage(gu64* g, int cnt, unsigned epoch,
     // back off: each pass re-reads cnt/64 sc1 granules per lane — real
     // fabric traffic from every CU; widen the interval once the edge is
     // clearly not ready yet (keeps gather latency, cuts poll bandwidth).
-    if (spins > 8)
-      __builtin_amdgcn_s_sleep(64);
-    else if (spins > 2)
-      __builtin_amdgcn_s_sleep(16);
+    if (spins > 2) __builtin_amdgcn_s_sleep(16);
   }
 }
 
