@@ -58,10 +58,10 @@ constexpr int SLOT = 16384;
 constexpr int LAG = 4;     // publish slot s-LAG+1 after issuing slot s
 constexpr int AT = 256;    // attention tokens per unit (one wave per unit)
 constexpr int NCU = 256;
-// 8 waves per CU: 1 loader + 7 consumers (2 waves/SIMD).  At 1 wave/SIMD
-// nothing hides a consumer's LDS/VALU latency chains; the second wave per
-// SIMD roughly doubles record throughput and halves gather sweeps.
-constexpr int NCONS = 7;
+// 1 loader + NCONS consumer waves per CU.  NCONS=7 (8 waves, 2/SIMD) was
+// tried for latency hiding but stalls in the gate_up/down ring window —
+// parked pending a fill-flow redesign; NCONS=3 is the validated geometry.
+constexpr int NCONS = 3;
 constexpr u32 ONE2_BF16 = 0x3F803F80u;
 
 struct Ctl {
@@ -281,12 +281,14 @@ __global__ __launch_bounds__((NCONS + 1) * 64, 1) void lfmq_engine_kernel(Engine
         ENG_PUB(0, (unsigned)s);
         unsigned spins = 0;
         for (;;) {
-          unsigned d0, d1, d2, d3, d4, d5, d6;
-          loader_read3(c_off[0], c_off[1], c_off[2], d0, d1, d2);
-          loader_read3(c_off[3], c_off[4], c_off[5], d3, d4, d5);
-          d6 = loader_read1(c_off[6]);
-          const unsigned m = min(min(min(d0, d1), min(d2, d3)),
-                                 min(min(d4, d5), d6));
+          unsigned m = 0xFFFFFFFFu;
+          static_assert(NCONS % 3 == 0 || NCONS == 7, "poll reads in 3s");
+          for (int w = 0; w + 3 <= NCONS; w += 3) {
+            unsigned d0, d1, d2;
+            loader_read3(c_off[w], c_off[w + 1], c_off[w + 2], d0, d1, d2);
+            m = min(m, min(d0, min(d1, d2)));
+          }
+          if (NCONS % 3 == 1) m = min(m, loader_read1(c_off[NCONS - 1]));
           if ((int)m >= s - a.nslot_ring + 1) break;
           if (spin_fail(spins, tmo, 0x10, lane)) return;
         }

@@ -39,7 +39,16 @@ __device__ __forceinline__ bool spin_fail(unsigned& spins, gu32* tmo,
     __builtin_amdgcn_s_sleep(2);
     return false;
   }
-  if (lane == 0 && tmo != nullptr) __hip_atomic_store(tmo, code, LFMQ_RLX_AGENT);
+  // write-once: the FIRST give-up code is the diagnostic one — later
+  // give-ups are cascade effects of the first missing producer.
+  if (lane == 0 && tmo != nullptr) {
+    unsigned expected = 0;
+    __hip_atomic_compare_exchange_strong(tmo, &expected, code,
+                                         __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+                                         __HIP_MEMORY_SCOPE_AGENT);
+    // class bitmask of every give-up site (tmo[7]): bit = code high nibble
+    __hip_atomic_fetch_or(tmo + 7, 1u << ((code >> 4) & 15), LFMQ_RLX_AGENT);
+  }
   return true;
 }
 

@@ -5,9 +5,13 @@ re-derives the walk independently from the spec in lfmq/models/engine.py's
 docstring and checks the built bytes.  GPU execution parity lives in
 tests/test_gpu_engine.py.
 """
+import os
+
 import torch
 
 import lfmq
+
+os.environ["LFMQ_ENGINE"] = "1"  # the engine is opt-in; tests opt in
 from lfmq.models.llama import LlamaConfig
 from lfmq.models.engine import (DecodeEngine, engine_supported, _rec_bytes,
                                 SLOT, NCU)

@@ -2,6 +2,7 @@
 """Minimal engine repro: 1 layer, 2 steps; prints tmo codes."""
 import os, sys, torch
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+os.environ["LFMQ_ENGINE_PROF"] = "1"
 import lfmq
 from lfmq.models.llama import LlamaConfig
 
@@ -24,6 +25,8 @@ for i in range(2):
     torch.cuda.synchronize()
     t = q._engine.tmo.cpu().tolist()
     print(f"step {i}: tmo={[hex(v) for v in t]} logit_max={float(lg.float().abs().max()):.4f}")
+    pr = q._engine.prof[:8].cpu().tolist()
+    print(f"   phase stamps nonzero: {[j for j, v in enumerate(pr) if v]}")
     state += torch.tensor([1, 1, 1], dtype=torch.int32, device="cuda")
     cache.seq_len += 1
 print("DONE")
