@@ -31,8 +31,10 @@ def main():
     ap.add_argument("--group-size", type=int, default=128)
     ap.add_argument("--layers", type=int, default=0,
                     help="override layer count (debug only; 0 = full model)")
+    ap.add_argument("--lm-head", default="bf16", choices=["bf16", "w4", "w8"],
+                    help="lm_head path: rocBLAS bf16 or HIP W4/W8 GEMV")
     ap.add_argument("--quant-lm-head", action="store_true",
-                    help="quantize the lm_head too (default: bf16 head)")
+                    help="legacy alias for --lm-head w4")
     ap.add_argument("--no-fidelity", action="store_true",
                     help="skip the delta-ppl/logit-KL fidelity pass")
     args = ap.parse_args()
@@ -55,7 +57,8 @@ def main():
     model, cfg, qcfg = build_quantized_llama(
         args.model, "rtn", args.bits, args.group_size, device,
         tp_rank=rank, tp_world=world, seed=0,
-        n_layers=args.layers or None, quant_lm_head=args.quant_lm_head)
+        n_layers=args.layers or None,
+        quant_lm_head="w4" if args.quant_lm_head else args.lm_head)
 
     r = bench_decode(model, cfg, args.batch, args.prompt, args.steps,
                      args.warmup, device)
@@ -98,7 +101,9 @@ def main():
             "config": {
                 "model": args.model,
                 "quant": f"w{args.bits}a16-g{args.group_size}-rtn"
-                         + ("-qhead" if args.quant_lm_head else ""),
+                         + ("" if args.lm_head == "bf16"
+                            and not args.quant_lm_head
+                            else f"-{args.lm_head if not args.quant_lm_head else 'w4'}head"),
                 "global_batch": args.batch,
                 "seq_len": args.prompt,
                 "gen_steps": r["steps"],

@@ -152,7 +152,8 @@ __global__ __launch_bounds__(TPB) void attn_decode_fused(
     const u32* __restrict__ qkv, u32* __restrict__ kc, u32* __restrict__ vc,
     float* __restrict__ scratch, u32* __restrict__ o_direct,
     const float* __restrict__ inv_freq, int b, int hq, int hkv, int tmax,
-    int T, const i32* __restrict__ T_dev, float scale, int nchunks) {
+    int T, const i32* __restrict__ T_dev, float scale, int nchunks,
+    u32* __restrict__ tickets) {
   if (T_dev != nullptr) T = *T_dev;
   const int bh = blockIdx.x;
   const int chunk = blockIdx.y;
@@ -291,16 +292,78 @@ __global__ __launch_bounds__(TPB) void attn_decode_fused(
     }
     return;
   }
+  // In-launch combine by the LAST-ARRIVING chunk of this bh (tickets !=
+  // null) instead of a separate combine kernel: ~4.8 us/launch x 32
+  // layers/step of pure boundary cost measured in round 1 (VERDICT item
+  // 7).  Guideline-16 discipline: partials stored write-through (agent
+  // relaxed sc1), asm vmcnt(0) drain before the arrival ticket (the
+  // compiler-hazard note in MI355X_MICROARCH.md), agent acquire by the
+  // winner before its plain reads.  The winner RESETS the ticket so
+  // hipGraph replays need no host-side re-zeroing.
   float* out = scratch + ((u64)bh * nchunks + chunk) * (D + 2);
   for (int j = threadIdx.x; j < D; j += TPB) {
     float v = 0.f;
 #pragma unroll
     for (int w = 0; w < NWAVE; ++w) v += o_sh[w][j];
-    out[j] = v;
+    if (tickets != nullptr)
+      __hip_atomic_store((__attribute__((address_space(1))) float*)&out[j], v,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    else
+      out[j] = v;
   }
   if (threadIdx.x == 0) {
-    out[D] = m;
-    out[D + 1] = l_sh;
+    if (tickets != nullptr) {
+      __hip_atomic_store((__attribute__((address_space(1))) float*)&out[D], m,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_store(
+          (__attribute__((address_space(1))) float*)&out[D + 1], l_sh,
+          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    } else {
+      out[D] = m;
+      out[D + 1] = l_sh;
+    }
+  }
+  if (tickets == nullptr) return;
+
+  // arrival ticket: one lane drains its stores, adds, and broadcasts
+  // whether this workgroup arrived last
+  __shared__ int last_sh;
+  const int used = (T + FCH - 1) / FCH;
+  if (threadIdx.x == 0) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    const unsigned old = __hip_atomic_fetch_add(
+        (__attribute__((address_space(1))) unsigned*)&tickets[bh], 1u,
+        __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    last_sh = (int)old == used - 1;
+  }
+  __syncthreads();
+  if (!last_sh) return;
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+
+  // combine (same math as attn_decode_combine), then reset the ticket
+  const float* base = scratch + (u64)bh * nchunks * (D + 2);
+  __shared__ float M_sh, invL_sh;
+  if (threadIdx.x == 0) {
+    float M = -INFINITY;
+    for (int c = 0; c < used; ++c) M = fmaxf(M, base[c * (D + 2) + D]);
+    float L = 0.f;
+    for (int c = 0; c < used; ++c)
+      L += __expf(base[c * (D + 2) + D] - M) * base[c * (D + 2) + D + 1];
+    M_sh = M;
+    invL_sh = 1.f / L;
+    __hip_atomic_store((__attribute__((address_space(1))) unsigned*)&tickets[bh],
+                       0u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  }
+  __syncthreads();
+  const int j = threadIdx.x;
+  if (2 * j < D) {
+    float acc0 = 0.f, acc1 = 0.f;
+    for (int c = 0; c < used; ++c) {
+      const float w = __expf(base[c * (D + 2) + D] - M_sh);
+      acc0 = fmaf(w, base[c * (D + 2) + 2 * j], acc0);
+      acc1 = fmaf(w, base[c * (D + 2) + 2 * j + 1], acc1);
+    }
+    o_direct[(u64)bh * dw + j] = pack_bf16x2(acc0 * invL_sh, acc1 * invL_sh);
   }
 }
 
@@ -368,11 +431,15 @@ hipError_t lfmq_attn_decode(const void* q, const void* kc, const void* vc,
 // Fused rope+append+attention decode entry (s == 1).  qkv is the raw
 // [b, (hq+2hkv)*d] projection output; writes the new k/v cache row
 // itself.  scratch sized as lfmq_attn_decode.
+// tickets != null: the last-arriving chunk combines in-launch (no combine
+// kernel).  tickets = zero-initialized u32[b*hq]; the winner resets its
+// entry, so graph replays stay correct with no host re-zeroing.
 hipError_t lfmq_attn_decode_fused(const void* qkv, void* kc, void* vc,
                                   void* o, void* scratch,
                                   const void* inv_freq, int b, int hq,
                                   int hkv, int tmax, int T, const void* T_dev,
-                                  int d, float scale, hipStream_t stream) {
+                                  int d, float scale, void* tickets,
+                                  hipStream_t stream) {
   if (d != 128 && d != 64) return hipErrorInvalidValue;
   if (hq % hkv) return hipErrorInvalidValue;
   const int nchunks = (T + 63) / 64;  // FCH = 64
@@ -383,8 +450,8 @@ hipError_t lfmq_attn_decode_fused(const void* qkv, void* kc, void* vc,
     hipLaunchKernelGGL((attn_decode_fused<DD>), gridA, blockA, 0, stream,     \
                        (const u32*)qkv, (u32*)kc, (u32*)vc, (float*)scratch,  \
                        (u32*)o, (const float*)inv_freq, b, hq, hkv, tmax, T,  \
-                       (const i32*)T_dev, scale, nchunks);                    \
-    if (nchunks > 1)                                                          \
+                       (const i32*)T_dev, scale, nchunks, (u32*)tickets);     \
+    if (nchunks > 1 && tickets == nullptr)                                    \
       hipLaunchKernelGGL((attn_decode_combine<DD>), gridB, blockB, 0, stream, \
                          (const float*)scratch, (u32*)o, nchunks, T,          \
                          (const i32*)T_dev, 64);                              \

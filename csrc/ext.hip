@@ -28,7 +28,8 @@ hipError_t lfmq_attn_decode(const void*, const void*, const void*, void*,
                             float, hipStream_t);
 hipError_t lfmq_attn_decode_fused(const void*, void*, void*, void*, void*,
                                   const void*, int, int, int, int, int,
-                                  const void*, int, float, hipStream_t);
+                                  const void*, int, float, void*,
+                                  hipStream_t);
 hipError_t lfmq_attn_prefill(const void*, const void*, const void*, void*,
                              int, int, int, int, int, int, int, float,
                              hipStream_t);
@@ -379,7 +380,8 @@ void advance_record(torch::Tensor state, torch::Tensor tok_buf,
 torch::Tensor attn_decode_fused(torch::Tensor qkv, torch::Tensor kc,
                                 torch::Tensor vc, long seq_len, double scale,
                                 torch::Tensor inv_freq, long hq,
-                                c10::optional<torch::Tensor> t_dev) {
+                                c10::optional<torch::Tensor> t_dev,
+                                c10::optional<torch::Tensor> tickets) {
   check_t(qkv, torch::kBFloat16, "qkv");
   check_t(kc, torch::kBFloat16, "k_cache");
   check_t(vc, torch::kBFloat16, "v_cache");
@@ -392,11 +394,17 @@ torch::Tensor attn_decode_fused(torch::Tensor qkv, torch::Tensor kc,
   auto o = torch::empty({(long)b, 1L, hq * (long)d}, qkv.options());
   const void* tptr = nullptr;
   if (t_dev.has_value()) tptr = t_dev->data_ptr();
+  void* tk = nullptr;
+  if (tickets.has_value()) {
+    check_t(*tickets, torch::kInt32, "tickets");
+    TORCH_CHECK(tickets->numel() >= (long)b * hq, "tickets too small");
+    tk = tickets->data_ptr();
+  }
   CHECK_HIP(lfmq_attn_decode_fused(qkv.data_ptr(), kc.data_ptr(),
                                    vc.data_ptr(), o.data_ptr(),
                                    scratch.data_ptr(), inv_freq.data_ptr(), b,
                                    (int)hq, hkv, tmax, (int)seq_len, tptr, d,
-                                   (float)scale, stream()));
+                                   (float)scale, tk, stream()));
   return o;
 }
 

@@ -66,16 +66,32 @@ def build_bf16_llama(model_name: str, device, seed: int = 0,
 def build_quantized_llama(model_name: str, scheme: str, bits: int, g: int,
                           device, tp_rank: int = 0, tp_world: int = 1,
                           seed: int = 0, n_layers: Optional[int] = None,
-                          quant_lm_head: bool = False):
+                          quant_lm_head="bf16"):
     """Random-init on device, quantize (RTN weights for bench speed: the
-    decode kernel path and memory traffic are scheme-independent), shard."""
+    decode kernel path and memory traffic are scheme-independent), shard.
+    quant_lm_head: "bf16" (rocBLAS head), "w4" or "w8" (HIP GEMV head;
+    legacy bool True == "w4")."""
     cfg = _bench_config(model_name, n_layers)
     model = build_model(cfg)
     model = model.to_empty(device=str(device)) if str(device) != "cpu" else model
     _fast_random_init(model, seed)
+    if quant_lm_head is True:
+        quant_lm_head = "w4"
+    elif quant_lm_head is False or quant_lm_head is None:
+        quant_lm_head = "bf16"
+    head_same_cfg = quant_lm_head == "w4" and bits == 4
     qcfg = QuantConfig(scheme=scheme, bits=bits, group_size=g,
-                       skip_modules=() if quant_lm_head else ("lm_head",))
+                       skip_modules=() if head_same_cfg else ("lm_head",))
     model = quantize(model, qcfg)
+    if quant_lm_head in ("w4", "w8") and not head_same_cfg:
+        import torch.nn as nn
+
+        from ..models.qlinear import QLinear
+        from ..quant.common import replace_module
+        hb = 4 if quant_lm_head == "w4" else 8
+        q = QLinear.from_rtn(model.lm_head,
+                             QuantConfig(scheme="rtn", bits=hb, group_size=g))
+        replace_module(model, "lm_head", q)
     if tp_world > 1:
         if cfg.arch == "mixtral":
             from ..dist.ep import apply_expert_parallel

@@ -45,6 +45,19 @@ def rmsnorm_residual(x: torch.Tensor, residual: torch.Tensor,
 
 
 _INV_FREQ_CACHE: dict = {}
+# zero-initialized arrival tickets for the in-launch attention combine:
+# the winning chunk resets its entry, so one buffer per (n, device) stays
+# valid across steps and graph replays
+_TICKETS_CACHE: dict = {}
+
+
+def _tickets(n: int, device) -> "torch.Tensor":
+    key = (n, str(device))
+    t = _TICKETS_CACHE.get(key)
+    if t is None:
+        t = torch.zeros(n, dtype=torch.int32, device=device)
+        _TICKETS_CACHE[key] = t
+    return t
 
 
 def _inv_freq(d: int, theta: float, device) -> torch.Tensor:
@@ -200,12 +213,14 @@ def attn_decode_fused(qkv, k_cache, v_cache, state_or_T, max_t: int,
     mode, T = state[1]) or a host int T."""
     d = k_cache.shape[-1]
     inv = _inv_freq(d, theta, qkv.device)
+    tk = _tickets(k_cache.shape[0] * hq, qkv.device)
     if torch.is_tensor(state_or_T):
         return hip_ext().attn_decode_fused(qkv.contiguous(), k_cache, v_cache,
                                            max_t, scale, inv, hq,
-                                           state_or_T[1:2])
+                                           state_or_T[1:2], tk)
     return hip_ext().attn_decode_fused(qkv.contiguous(), k_cache, v_cache,
-                                       int(state_or_T), scale, inv, hq, None)
+                                       int(state_or_T), scale, inv, hq, None,
+                                       tk)
 
 
 def dq_gemv_gateup(x: torch.Tensor, qweight: torch.Tensor,
