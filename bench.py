@@ -31,8 +31,12 @@ def main():
     ap.add_argument("--group-size", type=int, default=128)
     ap.add_argument("--layers", type=int, default=0,
                     help="override layer count (debug only; 0 = full model)")
-    ap.add_argument("--lm-head", default="bf16", choices=["bf16", "w4", "w8"],
-                    help="lm_head path: rocBLAS bf16 or HIP W4/W8 GEMV")
+    # default W8 head: logit KL identical to the bf16 head (5.1912e-5 vs
+    # 5.1909e-5 on the standard fidelity probe) at +7% tok/s — see
+    # profiles/lm_head_decision_r02.md
+    ap.add_argument("--lm-head", default="w8", choices=["bf16", "w4", "w8"],
+                    help="lm_head path: HIP W8 GEMV (default; KL-neutral), "
+                         "rocBLAS bf16, or HIP W4 (fastest, +6% logit KL)")
     ap.add_argument("--quant-lm-head", action="store_true",
                     help="legacy alias for --lm-head w4")
     ap.add_argument("--no-fidelity", action="store_true",

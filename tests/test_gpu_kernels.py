@@ -350,8 +350,9 @@ def test_attn_decode_fused_kernel():
     # fused kernel
     kcg, vcg = kc.to(DEV), vc.to(DEV)
     inv = ref.rope_freqs(d, theta).to(DEV)
+    tk = torch.zeros(8, dtype=torch.int32, device=DEV)
     o = lfmq_hip.attn_decode_fused(qkv.to(DEV), kcg, vcg, T,
-                                   1.0 / _m.sqrt(d), inv, hq, None)
+                                   1.0 / _m.sqrt(d), inv, hq, None, tk)
     assert rel_err(o.reshape(b, hq, 1, d), o_ref) < 2e-2
     # cache rows written
     assert rel_err(kcg[:, :, T - 1], k_ref[:, 0].transpose(0, 1).transpose(0, 1)) < 2e-2
@@ -376,9 +377,15 @@ def test_attn_decode_fused_multichunk():
     o_ref = ref.attn_decode(q_ref.transpose(1, 2), kc_ref, vc_ref, T,
                             1.0 / _m.sqrt(d))
     inv = ref.rope_freqs(d, 500000.0).to(DEV)
+    tk = torch.zeros(b * hq, dtype=torch.int32, device=DEV)
     o = lfmq_hip.attn_decode_fused(qkv.to(DEV), kc.to(DEV), vc.to(DEV), T,
-                                   1.0 / _m.sqrt(d), inv, hq, None)
+                                   1.0 / _m.sqrt(d), inv, hq, None, tk)
     assert rel_err(o.reshape(b, hq, 1, d), o_ref) < 2e-2
+    assert int(tk.abs().sum()) == 0, "winner must reset its ticket"
+    # and the separate-combine path (tickets=None) must still match
+    o2 = lfmq_hip.attn_decode_fused(qkv.to(DEV), kc.to(DEV), vc.to(DEV), T,
+                                    1.0 / _m.sqrt(d), inv, hq, None, None)
+    assert rel_err(o2.reshape(b, hq, 1, d), o_ref) < 2e-2
 
 
 @pytest.mark.parametrize("m", [5, 8, 16])
