@@ -74,7 +74,7 @@ def _manifest(model: nn.Module) -> dict:
             man[name] = {"kind": "qlinear", "in": mod.in_features,
                          "out": mod.out_features, "bits": mod.bits,
                          "g": mod.group_size, "bias": mod.bias is not None,
-                         "g_idx": hasattr(mod, "g_idx")}
+                         "perm": hasattr(mod, "perm")}
         elif isinstance(mod, Int8PerTensorLinear):
             man[name] = {"kind": "int8pt", "in": mod.in_features,
                          "out": mod.out_features, "bias": mod.bias is not None}
@@ -94,8 +94,9 @@ def _build_quantized_skeleton(meta: dict, dtype=None) -> nn.Module:
         if spec["kind"] == "qlinear":
             q = QLinear(spec["in"], spec["out"], spec["bits"], spec["g"],
                         bias=spec["bias"])
-            if spec.get("g_idx"):
-                q.register_buffer("g_idx", torch.zeros(spec["in"], dtype=torch.int32))
+            if spec.get("perm") or spec.get("g_idx"):
+                q.register_buffer("perm", torch.zeros(spec["in"],
+                                                      dtype=torch.int64))
         else:
             q = Int8PerTensorLinear(spec["in"], spec["out"], bias=spec["bias"])
         replace_module(model, name, q)

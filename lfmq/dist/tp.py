@@ -21,7 +21,9 @@ from typing import Dict, List, Optional, Tuple
 import torch
 import torch.nn as nn
 
-from ..models.qlinear import ColumnParallelQLinear, QLinear, RowParallelQLinear
+from ..models.qlinear import (ColumnParallelQLinear,
+                              GatherColumnParallelQLinear, QLinear,
+                              RowParallelQLinear)
 from ..quant.common import get_module, replace_module
 
 
@@ -41,9 +43,6 @@ def _cat_rows(t: torch.Tensor, slices: List[slice]) -> torch.Tensor:
 
 def _shard_col(q: QLinear, parts: List[Tuple[int, int]], rank: int, world: int,
                gather_output: bool = False) -> ColumnParallelQLinear:
-    assert not hasattr(q, "g_idx"), (
-        "act-order (g_idx) layers cannot be TP-sharded in this round — "
-        "re-quantize with act_order=False for TP")
     sl = _row_slices(parts, rank, world)
     out = ColumnParallelQLinear(q.in_features, q.out_features // world, q.bits,
                                 q.group_size, bias=q.bias is not None,
@@ -53,13 +52,38 @@ def _shard_col(q: QLinear, parts: List[Tuple[int, int]], rank: int, world: int,
     out.qzeros.copy_(_cat_rows(q.qzeros, sl))
     if q.bias is not None:
         out.bias.copy_(_cat_rows(q.bias, sl))
+    # act-order: the perm gathers on the (un-sharded) INPUT dim — replicate
+    if hasattr(q, "perm"):
+        out.register_buffer("perm", q.perm.clone())
     return out
 
 
+def _shard_gather_col(q: QLinear, rank: int,
+                      world: int) -> GatherColumnParallelQLinear:
+    """Column-shard an act-order layer that would normally be row-parallel
+    (its input arrives k-sharded and is all-gathered in forward)."""
+    sl = _row_slices([(0, q.out_features)], rank, world)
+    out = GatherColumnParallelQLinear(q.in_features, q.out_features // world,
+                                      q.bits, q.group_size,
+                                      bias=q.bias is not None)
+    out.qweight.copy_(_cat_rows(q.qweight, sl))
+    out.scales.copy_(_cat_rows(q.scales, sl))
+    out.qzeros.copy_(_cat_rows(q.qzeros, sl))
+    if q.bias is not None:
+        out.bias.copy_(_cat_rows(q.bias, sl))
+    out.register_buffer("perm", q.perm.clone())
+    return out
+
+
+def _shard_row_or_gather(q: QLinear, rank: int, world: int):
+    return _shard_gather_col(q, rank, world) if hasattr(q, "perm")         else _shard_row(q, rank, world)
+
+
 def _shard_row(q: QLinear, rank: int, world: int) -> RowParallelQLinear:
-    assert not hasattr(q, "g_idx"), (
-        "act-order (g_idx) layers cannot be TP-sharded in this round — "
-        "re-quantize with act_order=False for TP")
+    assert not hasattr(q, "perm"), (
+        "act-order row-parallel layers use the gather-column plan "
+        "(_shard_row_act_order), not a k-shard: the permutation crosses "
+        "k-shard boundaries")
     k = q.in_features
     assert k % world == 0, f"in_features {k} % tp {world} != 0"
     ks = k // world
@@ -111,14 +135,16 @@ def shard_llama(model: nn.Module, rank: int, world: int) -> nn.Module:
         qkv_parts = [(0, hq * d), (hq * d, hkv * d), ((hq + hkv) * d, hkv * d)]
         replace_module(attn, "qkv_proj",
                        _shard_col(attn.qkv_proj, qkv_parts, rank, world))
-        replace_module(attn, "o_proj", _shard_row(attn.o_proj, rank, world))
+        replace_module(attn, "o_proj",
+                       _shard_row_or_gather(attn.o_proj, rank, world))
         attn.n_heads = hq // world
         attn.n_kv_heads = hkv // world
         mlp = layer.mlp
         gu_parts = [(0, inter), (inter, inter)]
         replace_module(mlp, "gate_up_proj",
                        _shard_col(mlp.gate_up_proj, gu_parts, rank, world))
-        replace_module(mlp, "down_proj", _shard_row(mlp.down_proj, rank, world))
+        replace_module(mlp, "down_proj",
+                       _shard_row_or_gather(mlp.down_proj, rank, world))
         mlp.inter = inter // world
 
     # lm_head: vocab column-shard + all-gather (dense bf16 or quantized)
@@ -155,7 +181,8 @@ def shard_mixtral(model: nn.Module, rank: int, world: int) -> nn.Module:
         qkv_parts = [(0, hq * d), (hq * d, hkv * d), ((hq + hkv) * d, hkv * d)]
         replace_module(attn, "qkv_proj",
                        _shard_col(attn.qkv_proj, qkv_parts, rank, world))
-        replace_module(attn, "o_proj", _shard_row(attn.o_proj, rank, world))
+        replace_module(attn, "o_proj",
+                       _shard_row_or_gather(attn.o_proj, rank, world))
         attn.n_heads = hq // world
         attn.n_kv_heads = hkv // world
     lm = model.lm_head

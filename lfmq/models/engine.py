@@ -80,7 +80,7 @@ def engine_supported(model, batch: int) -> bool:
                      layer.mlp.gate_up_proj, layer.mlp.down_proj):
             if not isinstance(proj, QLinear) or proj.bits != 4:
                 return False
-            if proj.bias is not None or hasattr(proj, "g_idx"):
+            if proj.bias is not None or hasattr(proj, "perm"):
                 return False
             if g is None:
                 g = proj.group_size

@@ -188,7 +188,7 @@ class LlamaMLP(nn.Module):
         gp = self.gate_up_proj
         m = x.numel() // x.shape[-1]
         if (x.is_cuda and m <= 4 and type(gp).forward is QLinear.forward
-                and isinstance(gp, QLinear)):
+                and isinstance(gp, QLinear) and not hasattr(gp, "perm")):
             # fused gate_up GEMV + silu·mul epilogue (no separate pass)
             h = F.dq_gemv_gateup(x, gp.qweight, gp.scales, gp.qzeros,
                                  gp.bits, gp.group_size)
@@ -330,6 +330,13 @@ class LlamaForCausalLM(nn.Module):
                 return False
             if not isinstance(layer.self_attn.qkv_proj, QLinear):
                 return False
+            for proj in (layer.self_attn.qkv_proj, layer.self_attn.o_proj,
+                         layer.mlp.gate_up_proj, layer.mlp.down_proj):
+                if hasattr(proj, "perm"):
+                    # act-order: inputs need the perm gather — QLinear
+                    # .forward handles it; the fused 6-kernel chain bypasses
+                    # QLinear.forward, so fall back to the step path
+                    return False
         return True
 
     def forward_decode_step(self, input_ids, cache, state, max_seq):

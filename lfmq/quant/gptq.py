@@ -111,13 +111,13 @@ def gptq_quantize_layer(
         if i2 < k:
             w[:, i2:] -= err @ U[i1:i2, i2:]
 
-    g_idx = None
+    aperm = None
     if perm is not None:
-        inv = torch.argsort(perm)
-        codes = codes[:, inv]
-        # group of original column c = (its position in permuted order) // g
-        g_idx = (inv // g).to(torch.int32)
-    return codes, scales, zeros, g_idx
+        # codes stay in QUANT (permuted) order: groups are CONTIGUOUS, so
+        # every fused GPU kernel works unchanged; the forward gathers
+        # x[perm] instead (column j of W corresponds to input perm[j]).
+        aperm = perm.to(torch.int64)
+    return codes, scales, zeros, aperm
 
 
 class GPTQQuantizer:
@@ -165,12 +165,12 @@ class GPTQQuantizer:
                         lin = block.get_submodule(lname)
                         q = QLinear(meta["in"], meta["out"], meta["bits"],
                                     meta["g"], bias=meta["bias"]).to(device)
-                        if meta.get("g_idx"):
-                            # act_order checkpoints carry a g_idx buffer;
+                        if meta.get("perm"):
+                            # act_order checkpoints carry a perm buffer;
                             # register it so load_state_dict finds the key
                             # (resume of act_order=True failed).
-                            q.register_buffer("g_idx", torch.zeros(
-                                meta["in"], dtype=torch.int32, device=device))
+                            q.register_buffer("perm", torch.zeros(
+                                meta["in"], dtype=torch.int64, device=device))
                         replace_module(block, lname, q)
                     block.load_state_dict(saved["state"])
                     xs = [run_block(block, x.to(device)).to(xs[0].device)
@@ -193,15 +193,15 @@ class GPTQQuantizer:
                     run_block(block, x.to(device))
 
             for lname, lin in linears.items():
-                codes, scales, zeros, g_idx = gptq_quantize_layer(
+                codes, scales, zeros, perm = gptq_quantize_layer(
                     lin.weight.float(), hess[lname], cfg)
                 q = QLinear(lin.in_features, lin.out_features, cfg.bits,
                             cfg.group_size if cfg.group_size != -1 else lin.in_features,
                             bias=lin.bias is not None)
                 q.load_quantized(codes.cpu(), scales.cpu(), zeros.cpu(),
                                  lin.bias.detach() if lin.bias is not None else None)
-                if g_idx is not None:
-                    q.register_buffer("g_idx", g_idx.cpu())
+                if perm is not None:
+                    q.register_buffer("perm", perm.cpu())
                 q.to(device)
                 replace_module(block, lname, q)
                 del hess[lname]
@@ -214,7 +214,7 @@ class GPTQQuantizer:
                                       "out": mod.out_features,
                                       "bits": mod.bits, "g": mod.group_size,
                                       "bias": mod.bias is not None,
-                                      "g_idx": hasattr(mod, "g_idx")}
+                                      "perm": hasattr(mod, "perm")}
                 fd, tmp = tempfile.mkstemp(dir=os.path.join(resume_dir, "blocks"))
                 os.close(fd)
                 torch.save({"state": {k: v.cpu() for k, v in

@@ -75,19 +75,22 @@ def test_gptq_beats_rtn_on_correlated_inputs():
     assert err_gptq < err_rtn * 0.9  # GPTQ must clearly win on task loss
 
 
-def test_gptq_act_order_gidx():
+def test_gptq_act_order_perm():
     torch.manual_seed(1)
     n, k, g = 8, 64, 16
     w = torch.randn(n, k)
     x = torch.randn(500, k) * torch.linspace(0.1, 3.0, k)
     h = 2 * x.t() @ x
     cfg = QuantConfig(scheme="gptq", bits=4, group_size=g, act_order=True)
-    c, s, z, gi = gptq_quantize_layer(w, h, cfg)
-    assert gi is not None and gi.shape == (k,)
-    assert gi.min() >= 0 and gi.max() < k // g
-    # dequant via g_idx reproduces a weight close to w
-    dq = (c.float() - z.float()[:, gi.long()]) * s[:, gi.long()]
-    assert ((w - dq).abs().mean()) < 0.2
+    c, s, z, perm = gptq_quantize_layer(w, h, cfg)
+    assert perm is not None and perm.shape == (k,)
+    # perm is a permutation of 0..k-1, highest-|H| diagonal first
+    assert torch.equal(perm.sort().values, torch.arange(k))
+    # codes are in QUANT order (contiguous groups): column j of the
+    # dequant corresponds to ORIGINAL input perm[j]
+    gi = torch.arange(k) // g
+    dq = (c.float() - z.float()[:, gi]) * s[:, gi]
+    assert ((w[:, perm] - dq).abs().mean()) < 0.2
 
 
 def test_gptq_dead_column():
