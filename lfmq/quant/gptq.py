@@ -142,6 +142,21 @@ class GPTQQuantizer:
 
         from .common import make_block_runner
         run_block = make_block_runner(model)
+        # Hessian sampling sanity (profiles/gptq_fidelity_r02.md): with
+        # fewer calibration tokens than a layer's in_features the Hessian
+        # is rank-deficient and error propagation against its noisy
+        # inverse makes GPTQ WORSE than plain RTN (measured 2x logit KL).
+        n_tok = cfg.calib_n_seqs * cfg.calib_seq_len
+        max_k = max((m.in_features for _, m in
+                     iter_target_linears(model, cfg)), default=0)
+        if n_tok < max_k:
+            from ..obs import get_logger
+            get_logger().warning(
+                "GPTQ calibration has %d tokens but the widest target "
+                "layer has in_features=%d: the Hessian will be rank-"
+                "deficient and fidelity can drop BELOW plain RTN "
+                "(profiles/gptq_fidelity_r02.md). Use calib_n_seqs * "
+                "calib_seq_len >= %d.", n_tok, max_k, max_k)
         xs = _capture_block0_inputs(model, calib, device)
 
         if resume_dir:
