@@ -141,7 +141,7 @@ class ExpertParallelMoE(nn.Module):
         recv = xl.new_empty(world * C, h + 2)
         dist.all_to_all_single(recv, send.contiguous())
 
-        rows = recv[:, :h]
+        rows = recv[:, :h].contiguous()  # ext kernels need contiguous x
         recv_enc = recv[:, h:].to(torch.bfloat16).contiguous() \
             .view(torch.int32).reshape(world * C)
         # every local expert over the whole slab, one-hot masked: no

@@ -350,7 +350,7 @@ def test_attn_decode_fused_kernel():
     # fused kernel
     kcg, vcg = kc.to(DEV), vc.to(DEV)
     inv = ref.rope_freqs(d, theta).to(DEV)
-    tk = torch.zeros(8, dtype=torch.int32, device=DEV)
+    tk = torch.zeros(b * hq, dtype=torch.int32, device=DEV)
     o = lfmq_hip.attn_decode_fused(qkv.to(DEV), kcg, vcg, T,
                                    1.0 / _m.sqrt(d), inv, hq, None, tk)
     assert rel_err(o.reshape(b, hq, 1, d), o_ref) < 2e-2
