@@ -46,6 +46,12 @@ hipError_t lfmq_grouped_dq_gemv(const void*, const void*, const void*,
                                 const void*, void*, const void*, const void*,
                                 int, int, int, int, int, int, bool,
                                 hipStream_t);
+hipError_t lfmq_grouped_dq_gemv_scatter(const void*, const void*, const void*,
+                                        const void*, void*, const void*,
+                                        const void*, const void*, int, int,
+                                        int, int, int, int, hipStream_t);
+hipError_t lfmq_moe_route(const void*, void*, void*, void*, int, int, int,
+                          hipStream_t);
 hipError_t lfmq_argmax_rows(const void*, void*, void*, void*, int, int,
                             hipStream_t);
 hipError_t lfmq_read_probe(const void*, void*, long, long, int, int, int,
@@ -351,6 +357,43 @@ torch::Tensor grouped_dq_gateup(torch::Tensor x, torch::Tensor src_tok,
   return y;
 }
 
+// one-kernel MoE routing: logits [t, E] bf16 -> (src_tok, slot_w, offsets)
+std::vector<torch::Tensor> moe_route(torch::Tensor logits, long topk) {
+  check_t(logits, torch::kBFloat16, "logits");
+  const int t = (int)logits.size(0), E = (int)logits.size(1);
+  auto src_tok = torch::empty({(long)t * topk},
+                              logits.options().dtype(torch::kInt32));
+  auto slot_w = torch::empty({(long)t * topk},
+                             logits.options().dtype(torch::kFloat32));
+  auto offsets = torch::empty({(long)E + 1},
+                              logits.options().dtype(torch::kInt32));
+  CHECK_HIP(lfmq_moe_route(logits.data_ptr(), src_tok.data_ptr(),
+                           slot_w.data_ptr(), offsets.data_ptr(), t, E,
+                           (int)topk, stream()));
+  return {src_tok, slot_w, offsets};
+}
+
+// w2 grouped GEMV + fused weighted scatter into y_f32 [t, n] (caller zeroed)
+void grouped_dq_linear_scatter(torch::Tensor x, torch::Tensor offsets,
+                               torch::Tensor dst_tok, torch::Tensor slot_w,
+                               torch::Tensor y_f32, torch::Tensor qweight,
+                               torch::Tensor scales, torch::Tensor qzeros,
+                               long bits, long g) {
+  check_t(x, torch::kBFloat16, "x");
+  check_t(y_f32, torch::kFloat32, "y_f32");
+  check_t(dst_tok, torch::kInt32, "dst_tok");
+  check_t(slot_w, torch::kFloat32, "slot_w");
+  check_t(offsets, torch::kInt32, "offsets");
+  const int E = (int)qweight.size(0), n = (int)qweight.size(1);
+  const int k = (int)x.size(1);
+  TORCH_CHECK(y_f32.size(1) == n, "y_f32 width mismatch");
+  CHECK_HIP(lfmq_grouped_dq_gemv_scatter(
+      qweight.data_ptr(), scales.data_ptr(), qzeros.data_ptr(), x.data_ptr(),
+      y_f32.data_ptr(), offsets.data_ptr(), dst_tok.data_ptr(),
+      slot_w.data_ptr(), E, n, k, (int)g, (int)bits, (int)x.size(0),
+      stream()));
+}
+
 torch::Tensor grouped_dq_linear(torch::Tensor x, torch::Tensor offsets,
                                 torch::Tensor qweight, torch::Tensor scales,
                                 torch::Tensor qzeros, long bits, long g) {
@@ -507,5 +550,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused rope+KV-append+attention decode");
   m.def("grouped_dq_gateup", &grouped_dq_gateup, "MoE grouped gate_up+silu");
   m.def("grouped_dq_linear", &grouped_dq_linear, "MoE grouped linear");
+  m.def("moe_route", &moe_route, "one-kernel MoE top-k routing");
+  m.def("grouped_dq_linear_scatter", &grouped_dq_linear_scatter,
+        "MoE w2 grouped GEMV + fused weighted scatter");
   m.def("engine_step", &engine_step, "persistent decode engine step");
 }

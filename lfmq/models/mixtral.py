@@ -124,14 +124,15 @@ class MixtralMoE(nn.Module):
     def forward(self, x):
         b, s, h = x.shape
         x2 = x.reshape(-1, h)
-        weights, ids = self.route(x2)
         t = x2.shape[0]
         if (x.is_cuda and self._grouped is None
                 and isinstance(self.experts[0].w13, QLinear)):
             self.consolidate()  # lazy one-time stacking for the grouped kernel
         if x.is_cuda and self._grouped is not None and t * self.top_k <= 64:
-            y2 = self._forward_grouped(x2, weights, ids)
+            # routing runs inside _forward_grouped (one moe_route kernel)
+            y2 = self._forward_grouped(x2, None, None)
         else:
+            weights, ids = self.route(x2)
             y2 = self._forward_loop(x2, weights, ids)
         return y2.reshape(b, s, h)
 
@@ -149,34 +150,29 @@ class MixtralMoE(nn.Module):
         return y
 
     def _forward_grouped(self, x2, weights, ids):
-        """GPU decode: sort token-slots by expert, grouped GEMV pair,
-        weighted scatter-add back.  All device-side (graph-capturable)."""
+        """GPU decode: ONE routing kernel (moe_route: topk+softmax+counting
+        sort), grouped gateup GEMV, then the w2 grouped GEMV with the
+        weighted token scatter FUSED into its epilogue.  5 launches per MoE
+        layer vs ~16 with the round-1 aten chain; all
+        device-side (graph-capturable).  weights/ids args are unused here —
+        routing runs on the raw router logits."""
+        del weights, ids
         from ..ops import hip_ext
         t = x2.shape[0]
-        flat_ids = ids.reshape(-1)                       # [t*top_k]
-        order = torch.argsort(flat_ids)                  # stable enough: ids small
-        src_tok = (order // self.top_k).to(torch.int32)  # token of each slot
-        # per-slot expert ids in sorted order + counts -> offsets
-        sorted_ids = flat_ids[order]
-        counts = torch.zeros(self.n_experts, dtype=torch.int32, device=x2.device)
-        counts.scatter_add_(0, sorted_ids.long(),
-                            torch.ones_like(sorted_ids, dtype=torch.int32))
-        offsets = torch.zeros(self.n_experts + 1, dtype=torch.int32,
-                              device=x2.device)
-        torch.cumsum(counts, 0, out=offsets[1:].view(self.n_experts))
-        slot_w = weights.reshape(-1)[order].to(x2.dtype)  # [t*top_k]
+        logits16 = self.gate(x2)  # dense bf16 router GEMV
+        src_tok, slot_w, offsets = hip_ext().moe_route(
+            logits16.contiguous(), self.top_k)
 
         gg = self._grouped
         h = hip_ext().grouped_dq_gateup(
             x2, src_tok, offsets, gg["w13"]["qweight"], gg["w13"]["scales"],
             gg["w13"]["qzeros"], gg["w13"]["bits"], gg["w13"]["g"])
-        y_slots = hip_ext().grouped_dq_linear(
-            h, offsets, gg["w2"]["qweight"], gg["w2"]["scales"],
-            gg["w2"]["qzeros"], gg["w2"]["bits"], gg["w2"]["g"])
-        # weighted combine back to tokens
-        y = torch.zeros_like(x2, dtype=torch.float32)
-        y.index_add_(0, src_tok.long(),
-                     y_slots.float() * slot_w.float().unsqueeze(-1))
+        y = torch.zeros(t, x2.shape[1], dtype=torch.float32,
+                        device=x2.device)
+        hip_ext().grouped_dq_linear_scatter(
+            h, offsets, src_tok, slot_w, y, gg["w2"]["qweight"],
+            gg["w2"]["scales"], gg["w2"]["qzeros"], gg["w2"]["bits"],
+            gg["w2"]["g"])
         return y.to(x2.dtype)
 
 

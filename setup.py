@@ -36,6 +36,7 @@ SOURCES = [
     "csrc/probe.hip",
     "csrc/grouped_dq_gemm.hip",
     "csrc/engine.hip",
+    "csrc/moe_route.hip",
 ]
 
 setup(

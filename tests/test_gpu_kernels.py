@@ -441,3 +441,27 @@ def test_attn_prefill_flash(b, s, pos0):
     from lfmq.ops import functional as F
     o = F.attn_prefill_fused(q.to(DEV), kc.to(DEV), vc.to(DEV), pos0, scale)
     assert rel_err(o, o_ref) < 3e-2
+
+
+def test_moe_route_kernel():
+    """One-kernel MoE routing vs the torch reference chain."""
+    import lfmq_hip
+    torch.manual_seed(11)
+    for t, E, k in ((1, 8, 2), (7, 8, 2), (33, 16, 4)):
+        logits = torch.randn(t, E, dtype=torch.bfloat16, device=DEV)
+        src_tok, slot_w, offsets = lfmq_hip.moe_route(logits, k)
+        # reference
+        w_ref, ids_ref = torch.topk(logits.float(), k, dim=-1)
+        w_ref = torch.softmax(w_ref, dim=-1)
+        off = offsets.cpu().tolist()
+        assert off[0] == 0 and off[-1] == t * k
+        counts = torch.bincount(ids_ref.reshape(-1).cpu(), minlength=E)
+        for e in range(E):
+            assert off[e + 1] - off[e] == int(counts[e]), f"expert {e} count"
+            # every slot in the expert's range routes a token that picked e
+            for s in range(off[e], off[e + 1]):
+                tok = int(src_tok[s])
+                j = (ids_ref[tok] == e).nonzero()
+                assert j.numel() == 1
+                assert abs(float(slot_w[s]) -
+                           float(w_ref[tok, j[0, 0]])) < 1e-3
