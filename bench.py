@@ -58,11 +58,16 @@ def main():
     else:
         device = torch.device("cpu")  # contract smoke-run only
 
+    lm_head = "w4" if args.quant_lm_head else args.lm_head
+    if args.batch > 4 and lm_head == "w8":
+        # the W8 GEMV's win is the b<=4 lane-dot path; at larger decode
+        # batches the rocBLAS bf16 head GEMM is faster (the batched MFMA
+        # dequant route is bits=4-only) — keep w8 only where it pays
+        lm_head = "bf16"
     model, cfg, qcfg = build_quantized_llama(
         args.model, "rtn", args.bits, args.group_size, device,
         tp_rank=rank, tp_world=world, seed=0,
-        n_layers=args.layers or None,
-        quant_lm_head="w4" if args.quant_lm_head else args.lm_head)
+        n_layers=args.layers or None, quant_lm_head=lm_head)
 
     r = bench_decode(model, cfg, args.batch, args.prompt, args.steps,
                      args.warmup, device)
@@ -105,9 +110,7 @@ def main():
             "config": {
                 "model": args.model,
                 "quant": f"w{args.bits}a16-g{args.group_size}-rtn"
-                         + ("" if args.lm_head == "bf16"
-                            and not args.quant_lm_head
-                            else f"-{args.lm_head if not args.quant_lm_head else 'w4'}head"),
+                         + ("" if lm_head == "bf16" else f"-{lm_head}head"),
                 "global_batch": args.batch,
                 "seq_len": args.prompt,
                 "gen_steps": r["steps"],

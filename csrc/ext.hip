@@ -44,14 +44,15 @@ hipError_t lfmq_advance_record(void*, void*, const void*, void*, int, int,
                                hipStream_t);
 hipError_t lfmq_grouped_dq_gemv(const void*, const void*, const void*,
                                 const void*, void*, const void*, const void*,
-                                int, int, int, int, int, int, bool,
-                                hipStream_t);
+                                const void*, int, int, int, int, int, int,
+                                int, bool, hipStream_t);
 hipError_t lfmq_grouped_dq_gemv_scatter(const void*, const void*, const void*,
                                         const void*, void*, const void*,
-                                        const void*, const void*, int, int,
-                                        int, int, int, int, hipStream_t);
-hipError_t lfmq_moe_route(const void*, void*, void*, void*, int, int, int,
-                          hipStream_t);
+                                        const void*, const void*, const void*,
+                                        int, int, int, int, int, int,
+                                        hipStream_t);
+hipError_t lfmq_moe_route(const void*, void*, void*, void*, void*, int, int,
+                          int, hipStream_t);
 hipError_t lfmq_argmax_rows(const void*, void*, void*, void*, int, int,
                             hipStream_t);
 hipError_t lfmq_read_probe(const void*, void*, long, long, int, int, int,
@@ -340,7 +341,8 @@ void read_probe(torch::Tensor src, torch::Tensor sink, long row_bytes,
 torch::Tensor grouped_dq_gateup(torch::Tensor x, torch::Tensor src_tok,
                                 torch::Tensor offsets, torch::Tensor qweight,
                                 torch::Tensor scales, torch::Tensor qzeros,
-                                long bits, long g) {
+                                long bits, long g,
+                                c10::optional<torch::Tensor> active) {
   check_t(x, torch::kBFloat16, "x");
   check_t(qweight, torch::kInt32, "qweight");
   check_t(src_tok, torch::kInt32, "src_tok");
@@ -348,12 +350,18 @@ torch::Tensor grouped_dq_gateup(torch::Tensor x, torch::Tensor src_tok,
   const int E = (int)qweight.size(0), n = (int)qweight.size(1);
   const int k = (int)x.size(1);
   const long slots = src_tok.numel();
+  const void* act = nullptr;
+  int grid_e = E;
+  if (active.has_value()) {
+    act = active->data_ptr();
+    grid_e = (int)std::min((long)E, slots);
+  }
   auto y = torch::empty({slots, (long)(n / 2)}, x.options());
   CHECK_HIP(lfmq_grouped_dq_gemv(qweight.data_ptr(), scales.data_ptr(),
                                  qzeros.data_ptr(), x.data_ptr(), y.data_ptr(),
-                                 src_tok.data_ptr(), offsets.data_ptr(), E, n,
-                                 k, (int)g, (int)bits, (int)x.size(0), true,
-                                 stream()));
+                                 src_tok.data_ptr(), offsets.data_ptr(), act,
+                                 grid_e, E, n, k, (int)g, (int)bits,
+                                 (int)x.size(0), true, stream()));
   return y;
 }
 
@@ -367,10 +375,12 @@ std::vector<torch::Tensor> moe_route(torch::Tensor logits, long topk) {
                              logits.options().dtype(torch::kFloat32));
   auto offsets = torch::empty({(long)E + 1},
                               logits.options().dtype(torch::kInt32));
+  auto active = torch::empty({(long)E},
+                             logits.options().dtype(torch::kInt32));
   CHECK_HIP(lfmq_moe_route(logits.data_ptr(), src_tok.data_ptr(),
-                           slot_w.data_ptr(), offsets.data_ptr(), t, E,
-                           (int)topk, stream()));
-  return {src_tok, slot_w, offsets};
+                           slot_w.data_ptr(), offsets.data_ptr(),
+                           active.data_ptr(), t, E, (int)topk, stream()));
+  return {src_tok, slot_w, offsets, active};
 }
 
 // w2 grouped GEMV + fused weighted scatter into y_f32 [t, n] (caller zeroed)
@@ -378,7 +388,8 @@ void grouped_dq_linear_scatter(torch::Tensor x, torch::Tensor offsets,
                                torch::Tensor dst_tok, torch::Tensor slot_w,
                                torch::Tensor y_f32, torch::Tensor qweight,
                                torch::Tensor scales, torch::Tensor qzeros,
-                               long bits, long g) {
+                               long bits, long g,
+                               c10::optional<torch::Tensor> active) {
   check_t(x, torch::kBFloat16, "x");
   check_t(y_f32, torch::kFloat32, "y_f32");
   check_t(dst_tok, torch::kInt32, "dst_tok");
@@ -387,11 +398,17 @@ void grouped_dq_linear_scatter(torch::Tensor x, torch::Tensor offsets,
   const int E = (int)qweight.size(0), n = (int)qweight.size(1);
   const int k = (int)x.size(1);
   TORCH_CHECK(y_f32.size(1) == n, "y_f32 width mismatch");
+  const void* act = nullptr;
+  int grid_e = E;
+  if (active.has_value()) {
+    act = active->data_ptr();
+    grid_e = std::min((long)E, dst_tok.numel());
+  }
   CHECK_HIP(lfmq_grouped_dq_gemv_scatter(
       qweight.data_ptr(), scales.data_ptr(), qzeros.data_ptr(), x.data_ptr(),
       y_f32.data_ptr(), offsets.data_ptr(), dst_tok.data_ptr(),
-      slot_w.data_ptr(), E, n, k, (int)g, (int)bits, (int)x.size(0),
-      stream()));
+      slot_w.data_ptr(), act, grid_e, n, k, (int)g, (int)bits,
+      (int)x.size(0), stream()));
 }
 
 torch::Tensor grouped_dq_linear(torch::Tensor x, torch::Tensor offsets,
@@ -405,8 +422,9 @@ torch::Tensor grouped_dq_linear(torch::Tensor x, torch::Tensor offsets,
   auto y = torch::empty({x.size(0), (long)n}, x.options());
   CHECK_HIP(lfmq_grouped_dq_gemv(qweight.data_ptr(), scales.data_ptr(),
                                  qzeros.data_ptr(), x.data_ptr(), y.data_ptr(),
-                                 nullptr, offsets.data_ptr(), E, n, k, (int)g,
-                                 (int)bits, (int)x.size(0), false, stream()));
+                                 nullptr, offsets.data_ptr(), nullptr, E, E,
+                                 n, k, (int)g, (int)bits, (int)x.size(0),
+                                 false, stream()));
   return y;
 }
 

@@ -32,9 +32,15 @@ __global__ __launch_bounds__(WAVES_PER_WG * LFMQ_WAVE) void grouped_kernel(
     const u32* __restrict__ qzeros, const u32* __restrict__ x,
     u32* __restrict__ y, const i32* __restrict__ src_tok,
     const i32* __restrict__ offsets, const i32* __restrict__ dst_tok,
-    const float* __restrict__ slot_w, int n, int k, int g, int x_rows) {
+    const float* __restrict__ slot_w, const i32* __restrict__ active, int n,
+    int k, int g, int x_rows) {
   static_assert(!GATEUP || COLS % 2 == 0, "gateup pairs");
-  const int e = blockIdx.y;
+  // active-expert indirection (moe_route): grid.y covers only the worst-
+  // case ACTIVE count (min(E, slots)), not all E experts — the dead-
+  // expert workgroup dispatch was measurable at decode (8x7B top-2: 6/8
+  // of the grid exited early).
+  const int e = active ? active[blockIdx.y] : (int)blockIdx.y;
+  if (e < 0) return;
   const int s_begin = offsets[e];
   const int s_end = offsets[e + 1];
   if (s_begin >= s_end) return;
@@ -328,15 +334,16 @@ template <int BITS, bool GATEUP, int COLS, bool SCATTER>
 void launch_ki(dim3 grid, dim3 block, hipStream_t stream, const void* qweight,
                const void* scales, const void* qzeros, const void* x, void* y,
                const void* src_tok, const void* offsets, const void* dst_tok,
-               const void* slot_w, int n, int k, int g, int x_rows) {
+               const void* slot_w, const void* active, int n, int k, int g,
+               int x_rows) {
   const int kiters = (k + 2047) / 2048;
 #define GKI(KIV)                                                             \
   hipLaunchKernelGGL((grouped_kernel<BITS, GATEUP, COLS, KIV, SCATTER>),     \
                      grid, block, 0, stream, (const u32*)qweight,            \
                      (const u32*)scales, (const u32*)qzeros, (const u32*)x,  \
                      (u32*)y, (const i32*)src_tok, (const i32*)offsets,      \
-                     (const i32*)dst_tok, (const float*)slot_w, n, k, g,     \
-                     x_rows)
+                     (const i32*)dst_tok, (const float*)slot_w,              \
+                     (const i32*)active, n, k, g, x_rows)
   if (BITS == 4 && kiters == 2) GKI(2);
   else if (BITS == 4 && kiters == 7 && COLS <= 2) GKI(7);
   else GKI(0);
@@ -348,32 +355,34 @@ void launch_ki(dim3 grid, dim3 block, hipStream_t stream, const void* qweight,
 hipError_t lfmq_grouped_dq_gemv(const void* qweight, const void* scales,
                                 const void* qzeros, const void* x, void* y,
                                 const void* src_tok, const void* offsets,
+                                const void* active, int grid_e,
                                 int n_experts, int n, int k, int g, int bits,
                                 int x_rows, bool gateup, hipStream_t stream) {
   if (k % 8 || g % 32 || g > k || (gateup && n % 2)) return hipErrorInvalidValue;
+  (void)n_experts;
   dim3 block(WAVES_PER_WG * LFMQ_WAVE);
   if (gateup) {
     const int per_wg = WAVES_PER_WG * 2;  // COLS=4 -> 2 output pairs/wave
-    dim3 grid(ceil_div_u32(n / 2, per_wg), n_experts);
+    dim3 grid(ceil_div_u32(n / 2, per_wg), grid_e);
     if (bits == 4)
       launch_ki<4, true, 4, false>(grid, block, stream, qweight, scales,
                                    qzeros, x, y, src_tok, offsets, nullptr,
-                                   nullptr, n, k, g, x_rows);
+                                   nullptr, active, n, k, g, x_rows);
     else
       launch_ki<8, true, 4, false>(grid, block, stream, qweight, scales,
                                    qzeros, x, y, src_tok, offsets, nullptr,
-                                   nullptr, n, k, g, x_rows);
+                                   nullptr, active, n, k, g, x_rows);
   } else {
     const int per_wg = WAVES_PER_WG * 2;  // COLS=2 (deep-k shapes: KI=7)
-    dim3 grid(ceil_div_u32(n, per_wg), n_experts);
+    dim3 grid(ceil_div_u32(n, per_wg), grid_e);
     if (bits == 4)
       launch_ki<4, false, 2, false>(grid, block, stream, qweight, scales,
                                     qzeros, x, y, src_tok, offsets, nullptr,
-                                    nullptr, n, k, g, x_rows);
+                                    nullptr, active, n, k, g, x_rows);
     else
       launch_ki<8, false, 2, false>(grid, block, stream, qweight, scales,
                                     qzeros, x, y, src_tok, offsets, nullptr,
-                                    nullptr, n, k, g, x_rows);
+                                    nullptr, active, n, k, g, x_rows);
   }
   return hipGetLastError();
 }
@@ -383,19 +392,19 @@ hipError_t lfmq_grouped_dq_gemv(const void* qweight, const void* scales,
 hipError_t lfmq_grouped_dq_gemv_scatter(
     const void* qweight, const void* scales, const void* qzeros,
     const void* x, void* y_f32, const void* offsets, const void* dst_tok,
-    const void* slot_w, int n_experts, int n, int k, int g, int bits,
-    int x_rows, hipStream_t stream) {
+    const void* slot_w, const void* active, int grid_e, int n, int k, int g,
+    int bits, int x_rows, hipStream_t stream) {
   if (k % 8 || g % 32 || g > k) return hipErrorInvalidValue;
   dim3 block(WAVES_PER_WG * LFMQ_WAVE);
   const int per_wg = WAVES_PER_WG * 2;
-  dim3 grid(ceil_div_u32(n, per_wg), n_experts);
+  dim3 grid(ceil_div_u32(n, per_wg), grid_e);
   if (bits == 4)
     launch_ki<4, false, 2, true>(grid, block, stream, qweight, scales, qzeros,
                                  x, y_f32, nullptr, offsets, dst_tok, slot_w,
-                                 n, k, g, x_rows);
+                                 active, n, k, g, x_rows);
   else
     launch_ki<8, false, 2, true>(grid, block, stream, qweight, scales, qzeros,
                                  x, y_f32, nullptr, offsets, dst_tok, slot_w,
-                                 n, k, g, x_rows);
+                                 active, n, k, g, x_rows);
   return hipGetLastError();
 }

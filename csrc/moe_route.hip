@@ -18,10 +18,13 @@ namespace {
 constexpr int MAXE = 64;
 constexpr int MAXK = 4;
 
+// active[j] = j-th expert with a nonempty slot range (dense, grid.y
+// indexes it so dead experts never dispatch workgroups); active[j] = -1
+// beyond n_active.
 __global__ __launch_bounds__(64) void moe_route_kernel(
     const unsigned short* __restrict__ logits, i32* __restrict__ src_tok,
-    float* __restrict__ slot_w, i32* __restrict__ offsets, int t, int E,
-    int topk) {
+    float* __restrict__ slot_w, i32* __restrict__ offsets,
+    i32* __restrict__ active, int t, int E, int topk) {
   __shared__ int cnt[MAXE];
   __shared__ int offs[MAXE + 1];
   const int lane = threadIdx.x;
@@ -68,16 +71,19 @@ __global__ __launch_bounds__(64) void moe_route_kernel(
   __syncthreads();
   if (lane == 0) {
     int acc = 0;
+    int na = 0;
     for (int e = 0; e <= E; ++e) {
       offs[e] = acc;
       if (e < E) {
         offsets[e] = acc;
+        if (cnt[e] > 0) active[na++] = e;
         acc += cnt[e];
         cnt[e] = 0;  // becomes the write cursor
       } else {
         offsets[E] = acc;
       }
     }
+    for (int j = na; j < E; ++j) active[j] = -1;
   }
   __syncthreads();
   if (live) {
@@ -96,11 +102,11 @@ __global__ __launch_bounds__(64) void moe_route_kernel(
 }  // namespace
 
 hipError_t lfmq_moe_route(const void* logits, void* src_tok, void* slot_w,
-                          void* offsets, int t, int E, int topk,
+                          void* offsets, void* active, int t, int E, int topk,
                           hipStream_t stream) {
   if (t > 64 || E > MAXE || topk > MAXK) return hipErrorInvalidValue;
   hipLaunchKernelGGL(moe_route_kernel, dim3(1), dim3(64), 0, stream,
                      (const unsigned short*)logits, (i32*)src_tok,
-                     (float*)slot_w, (i32*)offsets, t, E, topk);
+                     (float*)slot_w, (i32*)offsets, (i32*)active, t, E, topk);
   return hipGetLastError();
 }

@@ -160,19 +160,19 @@ class MixtralMoE(nn.Module):
         from ..ops import hip_ext
         t = x2.shape[0]
         logits16 = self.gate(x2)  # dense bf16 router GEMV
-        src_tok, slot_w, offsets = hip_ext().moe_route(
+        src_tok, slot_w, offsets, active = hip_ext().moe_route(
             logits16.contiguous(), self.top_k)
 
         gg = self._grouped
         h = hip_ext().grouped_dq_gateup(
             x2, src_tok, offsets, gg["w13"]["qweight"], gg["w13"]["scales"],
-            gg["w13"]["qzeros"], gg["w13"]["bits"], gg["w13"]["g"])
+            gg["w13"]["qzeros"], gg["w13"]["bits"], gg["w13"]["g"], active)
         y = torch.zeros(t, x2.shape[1], dtype=torch.float32,
                         device=x2.device)
         hip_ext().grouped_dq_linear_scatter(
             h, offsets, src_tok, slot_w, y, gg["w2"]["qweight"],
             gg["w2"]["scales"], gg["w2"]["qzeros"], gg["w2"]["bits"],
-            gg["w2"]["g"])
+            gg["w2"]["g"], active)
         return y.to(x2.dtype)
 
 

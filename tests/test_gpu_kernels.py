@@ -449,7 +449,7 @@ def test_moe_route_kernel():
     torch.manual_seed(11)
     for t, E, k in ((1, 8, 2), (7, 8, 2), (33, 16, 4)):
         logits = torch.randn(t, E, dtype=torch.bfloat16, device=DEV)
-        src_tok, slot_w, offsets = lfmq_hip.moe_route(logits, k)
+        src_tok, slot_w, offsets, active = lfmq_hip.moe_route(logits, k)
         # reference
         w_ref, ids_ref = torch.topk(logits.float(), k, dim=-1)
         w_ref = torch.softmax(w_ref, dim=-1)
