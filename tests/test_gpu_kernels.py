@@ -300,7 +300,7 @@ def test_grouped_dq_kernels():
     offsets[1:] = counts.cumsum(0)
     import lfmq_hip
     h = lfmq_hip.grouped_dq_gateup(x.to(DEV), src_tok.to(DEV), offsets.to(DEV),
-                                   qw, sc, qz, 4, g)
+                                   qw, sc, qz, 4, g, None)
     # reference: per slot, expert gateup + silu
     h_ref = torch.zeros(t * topk, inter)
     for s_ in range(t * topk):
