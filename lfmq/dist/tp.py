@@ -222,6 +222,15 @@ def load_tp(path: str, meta: dict, device: str, rank: int, world: int,
             f"TP load for arch {cfg.arch!r} is not implemented: use "
             "lfmq.load(..., tp_world=1) + lfmq.dist.ep.apply_expert_parallel "
             "(Mixtral) or lfmq.dist.tp.shard_mixtral on the loaded model")
+    if any(spec.get("perm") or spec.get("g_idx")
+           for spec in meta.get("manifest", {}).values()
+           if isinstance(spec, dict)):
+        raise NotImplementedError(
+            "lazy TP load of an act-order checkpoint is not supported (the "
+            "k-permutation crosses the lazy row-shard slices): load with "
+            "tp_world=1 and shard in memory via lfmq.dist.tp.shard_llama, "
+            "which routes act-order row-parallel layers through the "
+            "gather-column plan")
     d, hq, hkv = cfg.head_dim, cfg.n_heads, cfg.n_kv_heads
     inter, vocab = cfg.intermediate_size, cfg.vocab_size
 

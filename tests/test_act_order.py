@@ -60,3 +60,17 @@ def test_act_order_gpu_decode_parity(tmp_path):
         rel = (g - c).abs().max() / c.abs().max()
         assert float(rel) < 2e-2, f"act-order GPU decode rel={float(rel)}"
         tok = c[:, -1:].argmax(dim=-1)
+
+
+def test_act_order_lazy_tp_load_raises(tmp_path):
+    """Lazy TP load slices row-parallel k — invalid under a k-permutation:
+    must refuse loudly and point to the in-memory shard path."""
+    cfg = LlamaConfig.tiny()
+    qcfg = lfmq.QuantConfig(scheme="gptq", bits=4, group_size=64,
+                            act_order=True, calib_n_seqs=2, calib_seq_len=32)
+    calib = CalibStream(cfg.vocab_size, 32, 2, qcfg.calib_seed)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=12), qcfg, calib=calib)
+    p = str(tmp_path / "ao")
+    lfmq.save(qm, p, qcfg)
+    with pytest.raises(NotImplementedError, match="act-order"):
+        lfmq.load(p, tp_rank=0, tp_world=2)
