@@ -8,7 +8,7 @@ all-reduce on RCCL over xGMI.  See SURVEY.md for the blueprint.
 from .api import build_model, load, quantize, save
 from .config import FORMAT_VERSION, GenerationConfig, QuantConfig
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 __all__ = ["quantize", "save", "load", "build_model", "QuantConfig",
            "GenerationConfig", "FORMAT_VERSION", "__version__"]

@@ -9,7 +9,9 @@ Math (the north star's inventory, BASELINE.json:5):
       err = (w_j − dq_j) / U_jj ;  w_{j+1..} −= err ⊗ U_{j, j+1..}
 Scales/zeros are recomputed at every group boundary from the
 error-compensated weights.  act_order=True quantizes columns in
-descending diag(H) order and records g_idx (per-column group index).
+descending diag(H) order; the packed result keeps that (group-contiguous)
+order and records `perm` — the forward gathers x[perm], so every fused
+GPU kernel serves act-order checkpoints unchanged.
 
 Calibration is sequential per decoder block: only one block's Hessians
 (≈1 GB fp32 for 8B, ≈4 GB for 70B) are live at a time, and the next
@@ -43,11 +45,12 @@ def gptq_quantize_layer(
     w: torch.Tensor, h: torch.Tensor, cfg: QuantConfig
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
     """w fp32 [n, k], h fp32 [k, k] -> (codes int32 [n,k], scales fp32 [n,G],
-    zeros int32 [n,G], g_idx int32 [k] or None).
+    zeros int32 [n,G], perm int64 [k] or None).
 
-    codes/scales/zeros are in ORIGINAL column order; with act_order the
-    per-column group assignment is returned as g_idx (scales row-indexed
-    by that group id).
+    Without act_order, codes are in original column order and perm is
+    None.  With act_order, codes stay in QUANT (permuted) order — groups
+    contiguous, scales indexed by position//g — and perm maps quant
+    column j to original input perm[j] (the forward gathers x[perm]).
     """
     n, k = w.shape
     g = k if cfg.group_size == -1 else cfg.group_size

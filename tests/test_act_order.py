@@ -1,4 +1,4 @@
-"""act-order GPTQ end-to-end on the CPU path (g_idx layout)."""
+"""act-order GPTQ end-to-end (group-contiguous + perm layout v2)."""
 import pytest
 import torch
 

@@ -108,3 +108,33 @@ def test_kvcache_stacked_views():
     cache.k[1][0, 0, 3, 5] = 7.0
     assert float(cache.k_all[1, 0, 0, 3, 5]) == 7.0
     assert cache.k[0].is_contiguous()
+
+
+def test_stream_ring_wrap_padding():
+    """A stream larger than the LDS ring must pad records at ring-wrap
+    boundaries (the kernel mirrors this rule; a straddling record would
+    read across the LDS ring seam)."""
+    cfg = LlamaConfig(vocab_size=512, hidden_size=2048,
+                      intermediate_size=2048, n_layers=8, n_heads=16,
+                      n_kv_heads=4, head_dim=128, max_seq=256)
+    q = lfmq.quantize(lfmq.build_model(cfg, seed=4),
+                      lfmq.QuantConfig(scheme="rtn", bits=4, group_size=128))
+    cache = q.make_cache(1, 128)
+    eng = DecodeEngine(q, cache, 128)
+    ring = eng.nslot * SLOT
+    assert eng.per_cu > ring, "test needs a stream larger than the ring"
+    g = eng.g
+    cur = 0
+    pads = 0
+    for layer in q.layers:
+        for op in (layer.self_attn.qkv_proj, layer.self_attn.o_proj,
+                   layer.mlp.gate_up_proj, layer.mlp.down_proj):
+            rb = _rec_bytes(op.in_features, op.in_features // g)
+            for _ in range(op.out_features // NCU):
+                if cur % ring + rb > ring:
+                    pads += 1
+                    cur += ring - cur % ring
+                assert cur % ring + rb <= ring
+                cur += rb
+    assert pads > 0, "no wrap pad exercised — grow the model"
+    assert eng.per_cu == (cur + SLOT - 1) // SLOT * SLOT
