@@ -150,6 +150,8 @@ def bench(ckpt: Optional[str] = typer.Option(None),
           prompt: int = typer.Option(128),
           steps: int = typer.Option(512),
           warmup: int = typer.Option(16),
+          lm_head: str = typer.Option("w8", help="bf16 | w8 | w4 "
+                                      "(profiles/lm_head_decision_r02.md)"),
           no_graph: bool = typer.Option(False)):
     """Decode tok/s (the BASELINE.json headline metric)."""
     import lfmq
@@ -160,7 +162,8 @@ def bench(ckpt: Optional[str] = typer.Option(None),
         qm = lfmq.load(ckpt, device=dev)
         cfg = qm.cfg
     else:
-        qm, cfg, _ = build_quantized_llama(model, scheme, 4, 128, dev)
+        qm, cfg, _ = build_quantized_llama(model, scheme, 4, 128, dev,
+                                           quant_lm_head=lm_head)
     r = bench_decode(qm, cfg, batch, prompt, steps, warmup, dev,
                      use_graph=not no_graph)
     typer.echo(json.dumps(r, indent=2))
