@@ -197,6 +197,22 @@ __device__ __forceinline__ float record_dot_checked(
   return v;
 }
 
+// Pair form: the fast path runs the interleaved two-record dot (twice the
+// independent load chains — the single-record chain is latency-bound at
+// 1 wave/SIMD); the debug-8 detector path falls back to the bracketed
+// single-record version so the corruption checks stay byte-exact.
+__device__ __forceinline__ void record_pair_dot_checked(
+    const char* recA, const char* recB, const char* lds_src, int k, int G,
+    int log2g, int lane, unsigned dbg, gu32* tmo, unsigned code, float& vA,
+    float& vB) {
+  if (dbg & 8) {
+    vA = record_dot_checked(recA, lds_src, k, G, log2g, lane, dbg, tmo, code);
+    vB = record_dot_checked(recB, lds_src, k, G, log2g, lane, dbg, tmo, code);
+    return;
+  }
+  record_pair_dot(recA, recB, lds_src, k, G, log2g, lane, vA, vB);
+}
+
 // single-granule bounded wait (combine path)
 __device__ __forceinline__ u32 gran_wait(gu64* g, unsigned epoch, gu32* tmo,
                                          unsigned code, int lane, bool* okp) {
@@ -405,21 +421,20 @@ __global__ __launch_bounds__((NCONS + 1) * 64, 1) void lfmq_engine_kernel(Engine
     // ---------------- P1: qkv records (norm-folded, scale inv_cur) -----
     {
       const float inv = ctl->inv_cur;
-      float v0 = 0.f;
-      for (int r = 0; r < npc_qkv; ++r) {
-        const u64 at = cursor_next(cur, rb_qkv, ring_bytes);
-        if ((r >> 1) % NCONS != wave) continue;
-        ENG_CHECK(wait_fill(ctl, at + rb_qkv, tmo, 0x30, lane, &dbg_fill));
-        const float v = inv * record_dot_checked(
-            ring + (u32)(at % ring_bytes), (const char*)xbuf, a.hidden, G1,
-            a.log2g, lane, a.debug, tmo, 0xE1);
-        if (r & 1) {
-          if (lane == 0)
-            store_granule(g_qkv + (cu * npc_qkv + r) / 2, e_qkv,
-                          pack_bf16x2(v0, v));
-        } else {
-          v0 = v;
-        }
+      for (int rp = 0; rp < npc_qkv / 2; ++rp) {
+        const u64 atA = cursor_next(cur, rb_qkv, ring_bytes);
+        const u64 atB = cursor_next(cur, rb_qkv, ring_bytes);
+        if (rp % NCONS != wave) continue;
+        // fills are stream-monotonic: waiting for B's end covers A
+        ENG_CHECK(wait_fill(ctl, atB + rb_qkv, tmo, 0x30, lane, &dbg_fill));
+        float vA, vB;
+        record_pair_dot_checked(ring + (u32)(atA % ring_bytes),
+                                ring + (u32)(atB % ring_bytes),
+                                (const char*)xbuf, a.hidden, G1, a.log2g,
+                                lane, a.debug, tmo, 0xE1, vA, vB);
+        if (lane == 0)
+          store_granule(g_qkv + cu * (npc_qkv / 2) + rp, e_qkv,
+                        pack_bf16x2(inv * vA, inv * vB));
         lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
       }
       lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
@@ -624,21 +639,19 @@ __global__ __launch_bounds__((NCONS + 1) * 64, 1) void lfmq_engine_kernel(Engine
 
     // ---------------- P5: o records (src = attnbuf) --------------------
     {
-      float v0 = 0.f;
-      for (int r = 0; r < npc_o; ++r) {
-        const u64 at = cursor_next(cur, rb_o, ring_bytes);
-        if ((r >> 1) % NCONS != wave) continue;
-        ENG_CHECK(wait_fill(ctl, at + rb_o, tmo, 0x52, lane, &dbg_fill));
-        const float v = record_dot_checked(ring + (u32)(at % ring_bytes),
-                                           unionb, a.hq * D, Go, a.log2g,
-                                           lane, a.debug, tmo, 0xE2);
-        if (r & 1) {
-          if (lane == 0)
-            store_granule(g_o + (cu * npc_o + r) / 2, e_o,
-                          pack_bf16x2(v0, v));
-        } else {
-          v0 = v;
-        }
+      for (int rp = 0; rp < npc_o / 2; ++rp) {
+        const u64 atA = cursor_next(cur, rb_o, ring_bytes);
+        const u64 atB = cursor_next(cur, rb_o, ring_bytes);
+        if (rp % NCONS != wave) continue;
+        ENG_CHECK(wait_fill(ctl, atB + rb_o, tmo, 0x52, lane, &dbg_fill));
+        float vA, vB;
+        record_pair_dot_checked(ring + (u32)(atA % ring_bytes),
+                                ring + (u32)(atB % ring_bytes), unionb,
+                                a.hq * D, Go, a.log2g, lane, a.debug, tmo,
+                                0xE2, vA, vB);
+        if (lane == 0)
+          store_granule(g_o + cu * (npc_o / 2) + rp, e_o,
+                        pack_bf16x2(vA, vB));
       }
       lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
     }
@@ -688,25 +701,31 @@ __global__ __launch_bounds__((NCONS + 1) * 64, 1) void lfmq_engine_kernel(Engine
     // ---------------- P7: gate_up records (scale inv2, silu pairs) -----
     {
       const float inv = ctl->inv_cur;
-      float vg = 0.f, h0 = 0.f;
-      for (int r = 0; r < npc_gu; ++r) {
-        const u64 at = cursor_next(cur, rb_gu, ring_bytes);
-        if ((r >> 2) % NCONS != wave) continue;
-        ENG_CHECK(wait_fill(ctl, at + rb_gu, tmo, 0x60, lane, &dbg_fill));
-        const float v = inv * record_dot_checked(
-            ring + (u32)(at % ring_bytes), (const char*)xbuf, a.hidden, G1,
-            a.log2g, lane, a.debug, tmo, 0xE3);
-        switch (r & 3) {
-          case 0: vg = v; break;
-          case 1: h0 = (vg / (1.f + __expf(-vg))) * v; break;
-          case 2: vg = v; break;
-          default: {
-            const float h1 = (vg / (1.f + __expf(-vg))) * v;
-            if (lane == 0)
-              store_granule(g_h + (cu * (npc_gu / 2) + (r >> 1)) / 2, e_h,
-                            pack_bf16x2(h0, h1));
-          }
-        }
+      for (int rq = 0; rq < npc_gu / 4; ++rq) {
+        // quad = (g0, u0, g1, u1): two pair-dots, silu(g)*u per pair
+        const u64 at0 = cursor_next(cur, rb_gu, ring_bytes);
+        const u64 at1 = cursor_next(cur, rb_gu, ring_bytes);
+        const u64 at2 = cursor_next(cur, rb_gu, ring_bytes);
+        const u64 at3 = cursor_next(cur, rb_gu, ring_bytes);
+        if (rq % NCONS != wave) continue;
+        ENG_CHECK(wait_fill(ctl, at1 + rb_gu, tmo, 0x60, lane, &dbg_fill));
+        float g0, u0;
+        record_pair_dot_checked(ring + (u32)(at0 % ring_bytes),
+                                ring + (u32)(at1 % ring_bytes),
+                                (const char*)xbuf, a.hidden, G1, a.log2g,
+                                lane, a.debug, tmo, 0xE3, g0, u0);
+        ENG_CHECK(wait_fill(ctl, at3 + rb_gu, tmo, 0x60, lane, &dbg_fill));
+        float g1, u1;
+        record_pair_dot_checked(ring + (u32)(at2 % ring_bytes),
+                                ring + (u32)(at3 % ring_bytes),
+                                (const char*)xbuf, a.hidden, G1, a.log2g,
+                                lane, a.debug, tmo, 0xE3, g1, u1);
+        g0 *= inv; u0 *= inv; g1 *= inv; u1 *= inv;
+        const float h0 = (g0 / (1.f + __expf(-g0))) * u0;
+        const float h1 = (g1 / (1.f + __expf(-g1))) * u1;
+        if (lane == 0)
+          store_granule(g_h + cu * (npc_gu / 4) + rq, e_h,
+                        pack_bf16x2(h0, h1));
         lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
       }
       lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
@@ -725,21 +744,19 @@ __global__ __launch_bounds__((NCONS + 1) * 64, 1) void lfmq_engine_kernel(Engine
 
     // ---------------- P9: down records (src = hbuf) --------------------
     {
-      float v0 = 0.f;
-      for (int r = 0; r < npc_dn; ++r) {
-        const u64 at = cursor_next(cur, rb_dn, ring_bytes);
-        if ((r >> 1) % NCONS != wave) continue;
-        ENG_CHECK(wait_fill(ctl, at + rb_dn, tmo, 0x63, lane, &dbg_fill));
-        const float v = record_dot_checked(ring + (u32)(at % ring_bytes),
-                                           unionb, a.inter, G2, a.log2g,
-                                           lane, a.debug, tmo, 0xE4);
-        if (r & 1) {
-          if (lane == 0)
-            store_granule(g_down + (cu * npc_dn + r) / 2, e_down,
-                          pack_bf16x2(v0, v));
-        } else {
-          v0 = v;
-        }
+      for (int rp = 0; rp < npc_dn / 2; ++rp) {
+        const u64 atA = cursor_next(cur, rb_dn, ring_bytes);
+        const u64 atB = cursor_next(cur, rb_dn, ring_bytes);
+        if (rp % NCONS != wave) continue;
+        ENG_CHECK(wait_fill(ctl, atB + rb_dn, tmo, 0x63, lane, &dbg_fill));
+        float vA, vB;
+        record_pair_dot_checked(ring + (u32)(atA % ring_bytes),
+                                ring + (u32)(atB % ring_bytes), unionb,
+                                a.inter, G2, a.log2g, lane, a.debug, tmo,
+                                0xE4, vA, vB);
+        if (lane == 0)
+          store_granule(g_down + cu * (npc_dn / 2) + rp, e_down,
+                        pack_bf16x2(vA, vB));
         lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));
       }
       lds_flag_store_rlx(&ctl->consumed[wave], (unsigned)(cur.off / SLOT));

@@ -276,6 +276,42 @@ __device__ __forceinline__ float record_dot_nored(const char* lds_rec,
   return acc;
 }
 
+// TWO records interleaved: a single record's chunk chain is latency-bound
+// at 1 wave/SIMD (load -> dot -> load ...); a pair gives two independent
+// chains and roughly doubles the load-level parallelism.  Records are
+// published in pairs anyway (one granule per two outputs).
+__device__ __forceinline__ void record_pair_dot(
+    const char* lds_recA, const char* lds_recB, const char* lds_src, int k,
+    int G, int log2g, int lane, float& vA, float& vB) {
+  (void)G;
+  const int iters = k >> 9;
+  float accA = 0.f, accB = 0.f;
+  RecChunk A0, A1, B0, B1;
+  chunk_load(A0, 0, lds_recA, iters, lds_src, log2g, lane);
+  chunk_load(B0, 0, lds_recB, iters, lds_src, log2g, lane);
+  int c0 = 0;
+  for (;;) {
+    if (c0 + 4 < iters) {
+      chunk_load(A1, c0 + 4, lds_recA, iters, lds_src, log2g, lane);
+      chunk_load(B1, c0 + 4, lds_recB, iters, lds_src, log2g, lane);
+    }
+    chunk_dot(A0, accA);
+    chunk_dot(B0, accB);
+    c0 += 4;
+    if (c0 >= iters) break;
+    if (c0 + 4 < iters) {
+      chunk_load(A0, c0 + 4, lds_recA, iters, lds_src, log2g, lane);
+      chunk_load(B0, c0 + 4, lds_recB, iters, lds_src, log2g, lane);
+    }
+    chunk_dot(A1, accA);
+    chunk_dot(B1, accB);
+    c0 += 4;
+    if (c0 >= iters) break;
+  }
+  vA = wave_reduce_sum(accA);
+  vB = wave_reduce_sum(accB);
+}
+
 __device__ __forceinline__ float record_dot(const char* lds_rec,
                                             const char* lds_src, int k, int G,
                                             int log2g, int lane) {

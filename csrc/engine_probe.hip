@@ -311,7 +311,8 @@ int main() {
 // waits): LDS pre-filled once, then REPS passes over NREC records of
 // k=4096 exactly as csrc/engine.hip P7 (gate_up) processes them.
 // ABLATE: 0 = full (dot+reduce+silu+publish), 1 = no publish,
-//         2 = no reduce (lane-0 partial only), 3 = dot only.
+//         2 = no reduce (lane-0 partial only), 3 = dot only,
+//         4 = record_pair_dot (the interleaved two-chain fast path).
 #include "../csrc/engine_args.h"
 namespace eng_ubench {
 constexpr int KREC = 4096;
@@ -340,6 +341,25 @@ __global__ __launch_bounds__(TPB) void record_ubench_kernel(
     t0 = __builtin_amdgcn_s_memtime();
   float sink = 0.f;
   for (int rep = 0; rep < reps; ++rep) {
+    if (ABLATE == 4) {
+      // mirrors engine P7: quad = two pair-dots + silu + one publish
+      for (int rq = 0; rq < NREC / 4; ++rq) {
+        if (rq % 3 != wave) continue;
+        float g0, u0, g1, u1;
+        lfmq_engine::record_pair_dot(recs + (4 * rq + 0) * RB,
+                                     recs + (4 * rq + 1) * RB, xbuf, KREC,
+                                     GREC, 7, lane, g0, u0);
+        lfmq_engine::record_pair_dot(recs + (4 * rq + 2) * RB,
+                                     recs + (4 * rq + 3) * RB, xbuf, KREC,
+                                     GREC, 7, lane, g1, u1);
+        const float h0 = (g0 / (1.f + __expf(-g0))) * u0;
+        const float h1 = (g1 / (1.f + __expf(-g1))) * u1;
+        if (lane == 0)
+          lfmq_engine::store_granule(g + rq, rep + 1, pack_bf16x2(h0, h1));
+        sink += h1;
+      }
+      continue;
+    }
     float vg = 0.f, h0 = 0.f;
     for (int r = 0; r < NREC; ++r) {
       if ((r >> 2) % 3 != wave) continue;
@@ -409,6 +429,7 @@ void run_record_ubench() {
   RUN(1);
   RUN(2);
   RUN(3);
+  RUN(4);
 #undef RUN
   (void)hipFree(src);
   (void)hipFree(gran);

@@ -309,7 +309,13 @@ class LlamaForCausalLM(nn.Module):
 
     def _engine_for(self, cache, batch: int) -> bool:
         """Build/bind the persistent decode engine for this cache (lazy;
-        one stream build per cache).  False => launched-kernel path."""
+        one stream build per cache).  False => launched-kernel path.
+        The env opt-in is re-read every step so LFMQ_ENGINE can be
+        toggled within one process; only STRUCTURAL unsupport (arch /
+        shape) latches."""
+        import os
+        if os.environ.get("LFMQ_ENGINE", "0") != "1":
+            return False
         eng = getattr(self, "_engine", None)
         if eng is not None and eng.cache is cache:
             return True
