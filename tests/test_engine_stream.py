@@ -138,3 +138,31 @@ def test_stream_ring_wrap_padding():
                 cur += rb
     assert pads > 0, "no wrap pad exercised — grow the model"
     assert eng.per_cu == (cur + SLOT - 1) // SLOT * SLOT
+
+
+def test_engine_env_gate_not_sticky():
+    """LFMQ_ENGINE=0 must not latch _engine_unsupported: the env opt-in
+    is re-read every step (toggling within one process is how the parity
+    tools compare engine vs launched paths)."""
+    q = _engine_model()
+    cache = q.make_cache(1, 64)
+    old = os.environ.get("LFMQ_ENGINE")
+    try:
+        os.environ["LFMQ_ENGINE"] = "0"
+        assert q._engine_for(cache, 1) is False
+        assert not getattr(q, "_engine_unsupported", False)
+        assert getattr(q, "_engine", None) is None
+    finally:
+        os.environ["LFMQ_ENGINE"] = old if old is not None else "1"
+
+
+def test_stream_pair_divisibility():
+    """The record phases consume PAIRS (quads for gate_up) — the
+    engine_supported shape gates must imply npc_qkv/npc_o/npc_dn even
+    and npc_gu % 4 == 0 for every supported config."""
+    q = _engine_model()
+    cfg = q.cfg
+    nqkv = (cfg.n_heads + 2 * cfg.n_kv_heads) * cfg.head_dim
+    assert (nqkv // NCU) % 2 == 0
+    assert (cfg.hidden_size // NCU) % 2 == 0          # o / down outputs
+    assert (2 * cfg.intermediate_size // NCU) % 4 == 0  # gate_up quads
