@@ -42,6 +42,13 @@ def _bench_config(model_name: str, n_layers: Optional[int] = None):
         cfg = LlamaConfig(vocab_size=512, hidden_size=256,
                           intermediate_size=512, n_layers=2, n_heads=4,
                           n_kv_heads=2, head_dim=64, max_seq=256)
+    elif model_name == "tiny-debug-tp8":
+        # contract-test shape that shards at TP 4 and 8 (8 q/kv heads,
+        # row-shard k dims divisible by 8 ranks x group 32) — exercises
+        # the N=4/8 rank math of the driver's scale run on CPU/gloo
+        cfg = LlamaConfig(vocab_size=512, hidden_size=512,
+                          intermediate_size=1024, n_layers=2, n_heads=8,
+                          n_kv_heads=8, head_dim=64, max_seq=256)
     else:
         raise ValueError(model_name)
     if n_layers:

@@ -7,6 +7,8 @@ import os
 import subprocess
 import sys
 
+import pytest
+
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
@@ -48,3 +50,46 @@ def test_scaling_runner_single_point():
     r = run_point(1, steps=2, warmup=1,
                   extra=["--model", "tiny-debug", "--prompt", "8"])
     assert r["metric"] == "decode tok/s" and r["n_gpus"] == 1
+
+
+def _run_bench_torchrun(world: int, model: str, extra=()):
+    """Launch bench.py exactly as the driver's SCALE_rNN runner does
+    (torch.distributed.run, one rank per GPU, rendezvous on 127.0.0.1)
+    and return rank 0's JSON line."""
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
+         "--gpus", str(world), "--steps", "2", "--warmup", "1",
+         "--prompt", "8", "--model", model, "--no-fidelity", *extra],
+        capture_output=True, text=True, cwd=REPO, timeout=600)
+    assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-2000:])
+    lines = [ln for ln in out.stdout.splitlines()
+             if ln.strip().startswith("{") and '"metric"' in ln]
+    assert len(lines) == 1, out.stdout  # rank 0 only
+    return json.loads(lines[0])
+
+
+def test_bench_torchrun_world2_gloo():
+    """TP-sharded tiny model at world 2 on CPU (gloo): all-reduce /
+    all-gather in the decode loop, one JSON line with n_gpus == 2."""
+    r = _run_bench_torchrun(2, "tiny-debug")
+    assert r["n_gpus"] == 2
+    assert r["config"]["parallelism"] == "tp2"
+    assert r["value"] > 0
+
+
+@pytest.mark.parametrize("world", [4, 8])
+def test_bench_torchrun_world48_gloo(world):
+    """The driver's 8-GPU scale run is the FIRST time TP 4/8 executes on
+    real hardware — de-risk the rank math (head/kv/group splits, lm_head
+    shard gather) at those worlds here on CPU."""
+    r = _run_bench_torchrun(world, "tiny-debug-tp8",
+                            extra=("--group-size", "32"))
+    assert r["n_gpus"] == world
+    assert r["config"]["parallelism"] == f"tp{world}"
+    assert r["value"] > 0
