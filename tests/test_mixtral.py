@@ -81,7 +81,7 @@ def test_ep_gloo_world4(tmp_path):
     """One expert per rank at world 4 (tiny config has 4 experts) — the
     all-to-all splits are maximally uneven at this world; pre-validates
     the EP rank math beyond the usual world-2 double."""
-    mp.spawn(_ep_worker, args=(4, 29741, str(tmp_path)), nprocs=4, join=True)
+    mp.spawn(_ep_worker, args=(4, 29761, str(tmp_path)), nprocs=4, join=True)
 
 
 def _tp_worker(rank, world, port, tmpdir):

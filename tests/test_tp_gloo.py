@@ -107,6 +107,27 @@ def test_tp_gloo_world4(tmp_path, world):
     assert rel < 0.05, f"TP4 logits rel err {float(rel)}"
 
 
+def test_tp_gloo_world8(tmp_path):
+    """8-way TP numerics — the driver's largest scale point.  One q/kv
+    head per rank, row-shard k slices of exactly one group (the
+    tightest legal group/shard alignment)."""
+    world = 8
+    cfg = LlamaConfig(vocab_size=256, hidden_size=256, intermediate_size=512,
+                      n_layers=2, n_heads=8, n_kv_heads=8, head_dim=32,
+                      max_seq=64)
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=43), qcfg)
+    path = str(tmp_path / "ckpt8")
+    lfmq.save(qm, path, qcfg)
+    ids = torch.arange(12).remainder(256).reshape(1, 12)
+    ref_logits = qm(ids, qm.make_cache(1, 32))
+    mp.spawn(_worker, args=(world, path, 29658, str(tmp_path)), nprocs=world,
+             join=True)
+    out = torch.load(str(tmp_path / "tp_out.pt"), weights_only=False)
+    rel = (out["logits"].float() - ref_logits.float()).abs().max() /         ref_logits.float().abs().max()
+    assert rel < 0.05, f"TP8 logits rel err {float(rel)}"
+
+
 def test_tp_gloo_quantized_lm_head(tmp_path):
     """TP with a QUANTIZED lm_head (skip_modules=()): column-shard +
     all-gather over the packed head."""
