@@ -17,7 +17,7 @@ from .config import QuantConfig
 from .models.llama import LlamaConfig, LlamaForCausalLM
 from .models.opt import OPTConfig, OPTForCausalLM
 from .models.qlinear import Int8PerTensorLinear, QLinear
-from .quant.common import iter_target_linears, replace_module
+from .quant.common import replace_module
 
 
 def build_model(model_cfg: Union[dict, LlamaConfig, OPTConfig], dtype=None,

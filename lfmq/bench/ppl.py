@@ -6,7 +6,6 @@ alongside as fidelity checks."""
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 

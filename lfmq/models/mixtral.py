@@ -14,7 +14,6 @@ combine pair instead.
 """
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass, asdict
 from typing import Optional
 

@@ -9,7 +9,6 @@ CPU decode config, which is spec'd as a CPU path.
 """
 from __future__ import annotations
 
-import os
 from typing import Optional
 
 import torch

@@ -16,7 +16,7 @@ expressible through v rows (the standard llm-awq simplification).
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import torch
 import torch.nn as nn

@@ -1,7 +1,7 @@
 """Shared quantizer helpers: module walking/replacement, block discovery."""
 from __future__ import annotations
 
-from typing import Iterator, List, Tuple
+from typing import List, Tuple
 
 import torch.nn as nn
 
