@@ -119,6 +119,12 @@ def eval(ckpt: str = typer.Option(...),
          seq_len: int = typer.Option(2048),
          n_seqs: int = typer.Option(4),
          seed: int = typer.Option(777),
+         seeds: int = typer.Option(1, help="evaluation-stream seeds to "
+                                   "average over.  Single-seed delta-ppl "
+                                   "is noise at random-init resolution "
+                                   "(std 3-8 measured, profiles/"
+                                   "gptq_fidelity_r02.md); >=3 reports "
+                                   "mean/std per fidelity metric"),
          model_seed: int = typer.Option(0, help="seed the checkpointed "
                                         "model was built with (quantize "
                                         "--seed); the bf16 reference must "
@@ -135,8 +141,18 @@ def eval(ckpt: str = typer.Option(...),
     qm = lfmq.load(ckpt, device=dev)
     fp = lfmq.build_model(meta["model"], seed=model_seed)
     fp = fp.to(dev)
-    r = bench_ppl(fp, qm, qm.cfg.vocab_size, seq_len=seq_len, n_seqs=n_seqs,
-                  seed=seed, device=dev)
+    runs = [bench_ppl(fp, qm, qm.cfg.vocab_size, seq_len=seq_len,
+                      n_seqs=n_seqs, seed=seed + i, device=dev)
+            for i in range(max(seeds, 1))]
+    r = dict(runs[0])
+    if len(runs) > 1:
+        import statistics
+        for key in ("delta_ppl", "logit_kl", "ppl_fp16", "ppl_quant"):
+            vals = [x[key] for x in runs if x.get(key) is not None]
+            if vals:
+                r[key] = statistics.fmean(vals)
+                r[key + "_std"] = statistics.pstdev(vals)
+        r["n_eval_seeds"] = len(runs)
     mse = weight_mse(fp.cpu(), qm.cpu())
     r["weight_mse_mean"] = sum(mse.values()) / max(len(mse), 1)
     typer.echo(json.dumps(r, indent=2))

@@ -40,3 +40,21 @@ def test_cli_info(tmp_path):
     out = json.loads(r.stdout[r.stdout.index("{"):])
     assert out["scheme"] == "rtn" and out["bits"] == 4
     assert out["quantized_linears"] > 0 and out["file_bytes"] > 0
+
+
+def test_cli_eval_multiseed(tmp_path):
+    """--seeds N averages the fidelity metrics over N evaluation streams
+    and reports the spread (single-seed delta-ppl is noise at random-init
+    resolution — profiles/gptq_fidelity_r02.md)."""
+    ck = str(tmp_path / "ck")
+    r = _run(["quantize", "--model", "tiny-llama", "--scheme", "rtn",
+              "--group-size", "32", "--out", ck, "--seed", "3"])
+    assert r.returncode == 0, r.stderr[-1500:]
+    r = _run(["eval", "--ckpt", ck, "--seq-len", "32", "--n-seqs", "1",
+              "--model-seed", "3", "--seeds", "3"])
+    assert r.returncode == 0, r.stderr[-1500:]
+    out = json.loads(r.stdout[r.stdout.index("{"):])
+    assert out["n_eval_seeds"] == 3
+    for key in ("delta_ppl", "logit_kl"):
+        assert key in out and key + "_std" in out
+        assert out[key + "_std"] >= 0.0
