@@ -108,13 +108,23 @@ class QLinear(nn.Module):
                            self.bits, self.group_size, self.bias)
 
     def forward_rms(self, x: torch.Tensor, eps: float) -> torch.Tensor:
-        """Decode fast path: rmsnorm(x, 1) fused into the GEMV."""
+        """Decode fast path: rmsnorm(x, 1) fused into the GEMV.
+
+        Act-order: rmsnorm is permutation-invariant (mean(x²) and the
+        elementwise scale commute with the gather), so gathering x FIRST
+        is exact — the fused kernel then sees quant-order columns."""
+        perm = getattr(self, "perm", None)
+        if perm is not None:
+            x = x.index_select(-1, perm)
         return F.dq_linear(x, self.qweight, self.scales, self.qzeros,
                            self.bits, self.group_size, self.bias,
                            rms_eps=eps)
 
     def forward_resid(self, x: torch.Tensor, resid: torch.Tensor) -> torch.Tensor:
         """Decode fast path: y = xW^T + resid fused in the epilogue."""
+        perm = getattr(self, "perm", None)
+        if perm is not None:
+            x = x.index_select(-1, perm)
         return F.dq_linear(x, self.qweight, self.scales, self.qzeros,
                            self.bits, self.group_size, self.bias, resid=resid)
 

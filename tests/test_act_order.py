@@ -74,3 +74,38 @@ def test_act_order_lazy_tp_load_raises(tmp_path):
     lfmq.save(qm, p, qcfg)
     with pytest.raises(NotImplementedError, match="act-order"):
         lfmq.load(p, tp_rank=0, tp_world=2)
+
+
+def test_perm_fused_primitives_cpu(monkeypatch):
+    """forward_resid / forward_rms on an act-order QLinear must apply the
+    perm gather before the fused GEMV (rmsnorm commutes with the
+    permutation, so gather-first is exact).  The fused epilogues are
+    GPU-only, so pin the contract with a capture stub: the x that reaches
+    dq_linear must be x[perm]."""
+    cfg = LlamaConfig.tiny()
+    qcfg = lfmq.QuantConfig(scheme="gptq", bits=4, group_size=64,
+                            act_order=True, calib_n_seqs=2, calib_seq_len=32)
+    calib = CalibStream(cfg.vocab_size, 32, 2, qcfg.calib_seed)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=12), qcfg, calib=calib)
+    ql = qm.layers[0].self_attn.qkv_proj
+    assert hasattr(ql, "perm")
+    torch.manual_seed(5)
+    x = (torch.randn(1, ql.in_features) * 0.1).to(torch.bfloat16)
+    want = x.index_select(-1, ql.perm)
+
+    seen = {}
+
+    def stub(xx, qw, sc, zz, bits, g, bias, rms_eps=-1.0, resid=None):
+        seen["x"] = xx
+        return torch.zeros(xx.shape[0], ql.out_features,
+                           dtype=xx.dtype)
+
+    import lfmq.models.qlinear as qlin_mod
+    monkeypatch.setattr(qlin_mod.F, "dq_linear", stub)
+
+    ql.forward_resid(x, torch.zeros(1, ql.out_features).to(torch.bfloat16))
+    assert torch.equal(seen["x"], want), "forward_resid skipped the gather"
+    ql.forward_rms(x, 1e-5)
+    assert torch.equal(seen["x"], want), "forward_rms skipped the gather"
+    ql.forward(x)
+    assert torch.equal(seen["x"], want)
