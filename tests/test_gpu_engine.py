@@ -40,7 +40,8 @@ def _per_step_rel(got, ref):
     r = ref.abs().amax(dim=-1)
     return float((d / r).max())
 
-def _decode_steps(model, cfg, ids, nsteps, max_seq, use_engine, seed=11):
+def _decode_steps(model, cfg, ids, nsteps, max_seq, use_engine, seed=11,
+                  debug=0):
     """Prefill eagerly, then nsteps of forward_decode_step with a FIXED
     (teacher-forced) token sequence — random-init logits are nearly flat,
     so free-running greedy would diverge on sub-tolerance noise and
@@ -48,6 +49,8 @@ def _decode_steps(model, cfg, ids, nsteps, max_seq, use_engine, seed=11):
     model._engine = None
     model._engine_unsupported = False
     os.environ["LFMQ_ENGINE"] = "1" if use_engine else "0"
+    if debug:
+        os.environ["LFMQ_ENGINE_DEBUG"] = str(debug)
     g = torch.Generator().manual_seed(seed)
     forced = torch.randint(0, cfg.vocab_size, (nsteps,), generator=g)
     try:
@@ -71,6 +74,7 @@ def _decode_steps(model, cfg, ids, nsteps, max_seq, use_engine, seed=11):
         return torch.cat(outs)
     finally:
         os.environ.pop("LFMQ_ENGINE", None)
+        os.environ.pop("LFMQ_ENGINE_DEBUG", None)
         model._engine = None
         model._engine_unsupported = False
 
@@ -119,3 +123,56 @@ def test_engine_parity_many_steps():
     got = _decode_steps(q, cfg, ids, 24, 512, use_engine=True)
     rel = _per_step_rel(got, ref)
     assert rel < 2e-2, f"per-step drift rel={rel}"
+
+
+@needs_gpu
+def test_engine_bit_determinism():
+    """Two identical engine runs from identical state must be BIT-equal:
+    the step has no cross-run nondeterminism (fixed reduction orders,
+    exactly-once granule publishes)."""
+    q, cfg = _model8b2()
+    ids = torch.randint(0, cfg.vocab_size, (1, 64), device="cuda")
+    a = _decode_steps(q, cfg, ids, 8, 512, use_engine=True)
+    b = _decode_steps(q, cfg, ids, 8, 512, use_engine=True)
+    assert torch.equal(a, b), "engine output differs across identical runs"
+
+
+@needs_gpu
+def test_engine_detectors_clean():
+    """Run with the in-kernel corruption detectors armed (debug 8: records
+    double-read + checksummed around each dot, which also exercises the
+    single-record dot path the pair fast path bypasses; debug 16: granule
+    staging re-verified after every sweep).  Any ring overwrite, double
+    writer, or stale sweep shows up in tmo[5]/tmo[6]."""
+    q, cfg = _model8b2()
+    ids = torch.randint(0, cfg.vocab_size, (1, 64), device="cuda")
+    ref = _decode_steps(q, cfg, ids, 4, 512, use_engine=False)
+    got = _decode_steps(q, cfg, ids, 4, 512, use_engine=True, debug=8 | 16)
+    rel = _per_step_rel(got, ref)
+    assert rel < 2e-2, f"detector-mode parity rel={rel}"
+    # _decode_steps clears model._engine; detectors write tmo BEFORE that
+    # teardown, so re-run one step keeping the engine to read tmo.
+    os.environ["LFMQ_ENGINE"] = "1"
+    os.environ["LFMQ_ENGINE_DEBUG"] = str(8 | 16)
+    try:
+        q._engine = None
+        q._engine_unsupported = False
+        cache = q.make_cache(1, max_seq=512, device="cuda")
+        q(ids, cache)
+        state = torch.tensor([64, 65, 0], dtype=torch.int32, device="cuda")
+        tok = torch.tensor([[5]], device="cuda")
+        for _ in range(3):
+            q.forward_decode_step(tok, cache, state, 512)
+            state += torch.tensor([1, 1, 1], dtype=torch.int32,
+                                  device="cuda")
+            cache.seq_len += 1
+        torch.cuda.synchronize()
+        tmo = q._engine.tmo.cpu().tolist()
+        assert tmo[0] == 0, f"give-up 0x{tmo[0]:x}"
+        assert tmo[5] == 0, f"value/ring corruption 0x{tmo[5]:x}"
+        assert tmo[6] == 0, f"double-writer/stale-sweep 0x{tmo[6]:x}"
+    finally:
+        os.environ.pop("LFMQ_ENGINE", None)
+        os.environ.pop("LFMQ_ENGINE_DEBUG", None)
+        q._engine = None
+        q._engine_unsupported = False
