@@ -30,6 +30,16 @@ def _model8b2(n_layers=2, vocab=1024, max_seq=1024):
     return q, cfg
 
 
+@pytest.fixture(scope="module")
+def m8b2():
+    """One default-shape build shared by the same-geometry tests (the
+    on-GPU quantize is ~30 s; weights are read-only in every test and
+    caches/engines are rebuilt per decode run)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    return _model8b2()
+
+
 
 
 def _per_step_rel(got, ref):
@@ -80,9 +90,9 @@ def _decode_steps(model, cfg, ids, nsteps, max_seq, use_engine, seed=11,
 
 
 @needs_gpu
-def test_engine_parity_short_T():
+def test_engine_parity_short_T(m8b2):
     """T < 256 the whole run: the no-combine (nch==1) attention path."""
-    q, cfg = _model8b2()
+    q, cfg = m8b2
     ids = torch.randint(0, cfg.vocab_size, (1, 64), device="cuda")
     ref = _decode_steps(q, cfg, ids, 8, 512, use_engine=False)
     got = _decode_steps(q, cfg, ids, 8, 512, use_engine=True)
@@ -91,10 +101,10 @@ def test_engine_parity_short_T():
 
 
 @needs_gpu
-def test_engine_parity_across_combine_boundary():
+def test_engine_parity_across_combine_boundary(m8b2):
     """Prefill 250 + 12 steps: T crosses 256 so the scratch-partial +
     in-engine combine path runs (nch 1 -> 2)."""
-    q, cfg = _model8b2()
+    q, cfg = m8b2
     ids = torch.randint(0, cfg.vocab_size, (1, 250), device="cuda")
     ref = _decode_steps(q, cfg, ids, 12, 512, use_engine=False)
     got = _decode_steps(q, cfg, ids, 12, 512, use_engine=True)
@@ -103,9 +113,9 @@ def test_engine_parity_across_combine_boundary():
 
 
 @needs_gpu
-def test_engine_parity_long_T():
+def test_engine_parity_long_T(m8b2):
     """T ~ 600: 3 chunks per head, uneven unit counts across CUs."""
-    q, cfg = _model8b2(max_seq=1024)
+    q, cfg = m8b2
     ids = torch.randint(0, cfg.vocab_size, (1, 590), device="cuda")
     ref = _decode_steps(q, cfg, ids, 4, 1024, use_engine=False)
     got = _decode_steps(q, cfg, ids, 4, 1024, use_engine=True)
@@ -114,10 +124,10 @@ def test_engine_parity_long_T():
 
 
 @needs_gpu
-def test_engine_parity_many_steps():
+def test_engine_parity_many_steps(m8b2):
     """24 teacher-forced steps: per-step error must stay bounded (no
     drift or state corruption across repeated engine launches)."""
-    q, cfg = _model8b2()
+    q, cfg = m8b2
     ids = torch.randint(0, cfg.vocab_size, (1, 32), device="cuda")
     ref = _decode_steps(q, cfg, ids, 24, 512, use_engine=False)
     got = _decode_steps(q, cfg, ids, 24, 512, use_engine=True)
@@ -126,11 +136,11 @@ def test_engine_parity_many_steps():
 
 
 @needs_gpu
-def test_engine_bit_determinism():
+def test_engine_bit_determinism(m8b2):
     """Two identical engine runs from identical state must be BIT-equal:
     the step has no cross-run nondeterminism (fixed reduction orders,
     exactly-once granule publishes)."""
-    q, cfg = _model8b2()
+    q, cfg = m8b2
     ids = torch.randint(0, cfg.vocab_size, (1, 64), device="cuda")
     a = _decode_steps(q, cfg, ids, 8, 512, use_engine=True)
     b = _decode_steps(q, cfg, ids, 8, 512, use_engine=True)
@@ -138,13 +148,13 @@ def test_engine_bit_determinism():
 
 
 @needs_gpu
-def test_engine_detectors_clean():
+def test_engine_detectors_clean(m8b2):
     """Run with the in-kernel corruption detectors armed (debug 8: records
     double-read + checksummed around each dot, which also exercises the
     single-record dot path the pair fast path bypasses; debug 16: granule
     staging re-verified after every sweep).  Any ring overwrite, double
     writer, or stale sweep shows up in tmo[5]/tmo[6]."""
-    q, cfg = _model8b2()
+    q, cfg = m8b2
     ids = torch.randint(0, cfg.vocab_size, (1, 64), device="cuda")
     ref = _decode_steps(q, cfg, ids, 4, 512, use_engine=False)
     got = _decode_steps(q, cfg, ids, 4, 512, use_engine=True, debug=8 | 16)
