@@ -83,6 +83,14 @@ def _manifest(model: nn.Module) -> dict:
 
 def save(model: nn.Module, path: str, quant_cfg: QuantConfig) -> None:
     state = {k: v for k, v in model.state_dict().items()}
+    # tied embeddings alias lm_head.weight to embed_tokens.weight;
+    # safetensors refuses shared storage, so persist ONE copy and let
+    # load() re-alias (tie_embeddings is in the model-config echo).
+    lm = getattr(model, "lm_head", None)
+    emb = getattr(model, "embed_tokens", None)
+    if (isinstance(lm, nn.Linear) and emb is not None
+            and "lm_head.weight" in state and lm.weight is emb.weight):
+        del state["lm_head.weight"]
     save_packed(state, quant_cfg, model.cfg.to_dict(), path,
                 manifest=_manifest(model))
 
@@ -116,6 +124,10 @@ def load(path: str, device: str = "cpu", tp_rank: int = 0, tp_world: int = 1,
     # module tree (no CPU materialization of a second copy): a 70B W4
     # checkpoint (~35 GB packed) loads with one host pass, not two.
     state = load_packed(path, device=str(device))
+    if ("lm_head.weight" not in state and "embed_tokens.weight" in state
+            and isinstance(getattr(model, "lm_head", None), nn.Linear)):
+        # tied checkpoint stores one copy (save() dropped the alias)
+        state["lm_head.weight"] = state["embed_tokens.weight"]
     if dtype is not None:
         # assign=True would otherwise let the checkpoint dtype override an
         # explicitly requested one: cast float tensors here.

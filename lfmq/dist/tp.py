@@ -297,6 +297,17 @@ def load_tp(path: str, meta: dict, device: str, rank: int, world: int,
             fixed["lm_head.local.weight"] = t
             continue
         fixed[tname] = t
+    if (not any(k.startswith("lm_head.") for k in fixed)
+            and "embed_tokens.weight" in fixed):
+        # tied checkpoint: one stored copy (api.save dropped the alias);
+        # the embedding is replicated, the head is vocab-column-sharded —
+        # slice this rank's rows out of the replicated table
+        sl = _row_slices([(0, vocab)], rank, world)[0]
+        t = fixed["embed_tokens.weight"][sl]
+        if isinstance(model.lm_head, _GatheredLinear):
+            fixed["lm_head.local.weight"] = t
+        else:
+            fixed["lm_head.weight"] = t
     model.load_state_dict(fixed, strict=True)
     return model.to(device)
 

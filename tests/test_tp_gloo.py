@@ -273,3 +273,43 @@ def test_collective_call_sites_no_spawn(monkeypatch):
     qm(ids, qm.make_cache(1, 16))
     assert calls["all_reduce"] == 2 * cfg.n_layers, calls
     assert calls["all_gather"] == 1, calls
+
+
+def test_tied_embeddings_save_load_roundtrip(tmp_path):
+    """Tied lm_head aliases embed_tokens; safetensors refuses shared
+    storage, so save() drops the alias and load() re-ties.  Logits must
+    round-trip bit-exact and the loaded module must be truly tied."""
+    cfg = _tiny_cfg()
+    cfg.tie_embeddings = True
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=9), qcfg)
+    assert qm.lm_head.weight is qm.embed_tokens.weight
+    path = str(tmp_path / "tied")
+    lfmq.save(qm, path, qcfg)
+    m2 = lfmq.load(path)
+    assert m2.lm_head.weight is m2.embed_tokens.weight
+    ids = torch.arange(12).remainder(256).reshape(1, 12)
+    a = qm(ids, qm.make_cache(1, 32))
+    b = m2(ids, m2.make_cache(1, 32))
+    assert torch.equal(a, b)
+
+
+def test_tied_embeddings_tp_gloo(tmp_path):
+    """TP load of a tied checkpoint: the single stored copy must be
+    row-sliced into each rank's vocab shard of the gathered head."""
+    cfg = _tiny_cfg()
+    cfg.tie_embeddings = True
+    qcfg = lfmq.QuantConfig(scheme="rtn", bits=4, group_size=32)
+    qm = lfmq.quantize(lfmq.build_model(cfg, seed=9), qcfg)
+    path = str(tmp_path / "tied")
+    lfmq.save(qm, path, qcfg)
+    ids = torch.arange(12).remainder(256).reshape(1, 12)
+    ref_logits = qm(ids, qm.make_cache(1, 32))
+    ref_gen = qm.generate(ids[:, :4], max_new_tokens=4)
+    mp.spawn(_worker, args=(2, path, 29671, str(tmp_path)), nprocs=2,
+             join=True)
+    out = torch.load(str(tmp_path / "tp_out.pt"), weights_only=False)
+    rel = (out["logits"].float() - ref_logits.float()).abs().max() / \
+        ref_logits.float().abs().max()
+    assert rel < 0.05, f"tied TP logits rel err {float(rel)}"
+    assert torch.equal(out["gen"], ref_gen)
