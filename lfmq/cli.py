@@ -140,6 +140,13 @@ def eval(ckpt: str = typer.Option(...),
     meta = load_meta(ckpt)
     qm = lfmq.load(ckpt, device=dev)
     fp = lfmq.build_model(meta["model"], seed=model_seed)
+    if meta["quant"].get("fold_rms", False):
+        # the checkpoint stores norm-FOLDED projections; fold the bf16
+        # reference identically (function-preserving) so weight_mse
+        # compares like with like — unfolded-vs-folded reads ~0.5 rel
+        # MSE regardless of quantization quality
+        from lfmq.quant.fold import fold_rms_norms
+        fold_rms_norms(fp)
     fp = fp.to(dev)
     runs = [bench_ppl(fp, qm, qm.cfg.vocab_size, seq_len=seq_len,
                       n_seqs=n_seqs, seed=seed + i, device=dev)

@@ -28,6 +28,9 @@ def test_cli_quantize_eval_cpu(tmp_path):
     # quantization noise only: the matching-seed bf16 reference keeps the
     # divergence small; a wrong model_seed would give KL ~ ln(vocab)
     assert out["logit_kl"] < 1.0, out
+    # weight_mse must compare against a norm-FOLDED reference: the
+    # unfolded-vs-folded mistake reads ~0.5 rel MSE, true W4 error ~1e-2
+    assert out["weight_mse_mean"] < 0.05, out
 
 
 def test_cli_info(tmp_path):
