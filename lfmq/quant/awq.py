@@ -70,8 +70,8 @@ def search_clip(x: torch.Tensor, w: torch.Tensor, cfg: QuantConfig,
     clip ratio per row [n]."""
     xw = x.float().pow(2).mean(0)                    # per-in-channel energy
     n = w.shape[0]
-    best_err = torch.full((n,), float("inf"))
-    best_r = torch.ones(n)
+    best_err = torch.full((n,), float("inf"), device=w.device)
+    best_r = torch.ones(n, device=w.device)
     for r in grid:
         wc = torch.clamp(w, min=w.amin(dim=1, keepdim=True) * r,
                          max=w.amax(dim=1, keepdim=True) * r)

@@ -58,6 +58,22 @@ def test_gpu_quantize_gptq():
     assert torch.isfinite(logits.float()).all()
 
 
+def test_gpu_quantize_awq_clip():
+    """AWQ with the clip search ON the GPU — the clip path's tensors must
+    live on the weight's device (device-mismatch regression guard)."""
+    cfg = LlamaConfig(vocab_size=1024, hidden_size=256, intermediate_size=512,
+                      n_layers=1, n_heads=4, n_kv_heads=2, head_dim=64,
+                      max_seq=256)
+    qcfg = lfmq.QuantConfig(scheme="awq", bits=4, group_size=64,
+                            awq_clip=True, awq_grid=4,
+                            calib_n_seqs=2, calib_seq_len=64)
+    model = lfmq.build_model(cfg, seed=6).to(DEV)
+    qm = lfmq.quantize(model, qcfg, device=torch.device(DEV))
+    ids = torch.randint(0, cfg.vocab_size, (1, 8), device=DEV)
+    logits = qm(ids, qm.make_cache(1, 32))
+    assert torch.isfinite(logits.float()).all()
+
+
 def test_decode_matches_prefill_gpu():
     cfg = LlamaConfig(vocab_size=1024, hidden_size=512, intermediate_size=1024,
                       n_layers=2, n_heads=8, n_kv_heads=2, head_dim=64,
