@@ -57,6 +57,10 @@ def quantize(model: str = typer.Option("llama3-8b"),
              calib_len: int = typer.Option(512),
              resume: bool = typer.Option(False, help="resume a partial run"),
              act_order: bool = typer.Option(False),
+             sym: bool = typer.Option(False, help="symmetric quantization "
+                                      "(no zero-point)"),
+             awq_clip: bool = typer.Option(False, help="AWQ per-row weight-"
+                                           "clip search (scheme=awq only)"),
              hf_dir: Optional[str] = typer.Option(
                  None, help="local HF-transformers checkpoint dir "
                  "(safetensors shards) to load into the preset before "
@@ -69,7 +73,7 @@ def quantize(model: str = typer.Option("llama3-8b"),
     dev = _device()
     qcfg = lfmq.QuantConfig(scheme=scheme, bits=bits, group_size=group_size,
                             calib_n_seqs=calib_seqs, calib_seq_len=calib_len,
-                            act_order=act_order)
+                            act_order=act_order, sym=sym, awq_clip=awq_clip)
     with phase_timer("build model"):
         m, cfg = _build_preset(model, seed, layers)
         if hf_dir is not None:
