@@ -170,6 +170,18 @@ def eval(ckpt: str = typer.Option(...),
 
 
 @app.command()
+def export_hf(ckpt: str = typer.Option(...),
+              out: str = typer.Option(...)):
+    """Dequantize a packed checkpoint back to an HF-transformers-layout
+    bf16 safetensors dir (the inverse of quantize --hf-dir)."""
+    import lfmq
+    from lfmq.models.export_hf import export_hf as _export
+    qm = lfmq.load(ckpt)
+    fn = _export(qm, out)
+    typer.echo(f"wrote {fn}")
+
+
+@app.command()
 def bench(ckpt: Optional[str] = typer.Option(None),
           model: str = typer.Option("llama3-8b"),
           scheme: str = typer.Option("rtn"),
