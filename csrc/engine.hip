@@ -62,7 +62,6 @@ constexpr int NCU = 256;
 // tried for latency hiding but stalls in the gate_up/down ring window —
 // parked pending a fill-flow redesign; NCONS=3 is the validated geometry.
 constexpr int NCONS = 3;
-constexpr u32 ONE2_BF16 = 0x3F803F80u;
 
 struct Ctl {
   unsigned filled;       // loader: slots published
