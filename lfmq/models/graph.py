@@ -76,6 +76,15 @@ class GraphDecoder:
         [batch, n_tokens] starting with the first post-prefill token
         (each replay records the token it consumed and feeds the newly
         sampled one back into the graph's input buffer)."""
+        if self.step_count + n_tokens > self.ring.shape[1]:
+            raise RuntimeError(
+                f"token ring exhausted: {self.step_count} + {n_tokens} > "
+                f"{self.ring.shape[1]} (raise max_new_tokens at construction)")
+        if self.cache.seq_len + n_tokens > self.max_seq:
+            raise RuntimeError(
+                f"KV cache exhausted: {self.cache.seq_len} + {n_tokens} > "
+                f"{self.max_seq} (the in-graph append would write past the "
+                f"cache; raise max_seq at construction)")
         if self.graph is None:
             # capture AFTER a real prefill so shapes/addresses are final
             saved = (self.state.clone(), self.tok.clone(),

@@ -51,7 +51,8 @@ def make_app(model) -> FastAPI:
         need = ids.shape[1] + max_new + 8
         with graph_lock:
             gd = graph_decoders.get(b)
-            if gd is None or gd.max_seq < need:
+            if gd is None or gd.max_seq < need or \
+                    gd.ring.shape[1] < max_new:
                 gd = GraphDecoder(model, b, max_seq=max(1024, need),
                                   max_new_tokens=max(4096, max_new))
                 graph_decoders[b] = gd
