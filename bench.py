@@ -108,7 +108,10 @@ def main():
             "delta_ppl": delta_ppl,
             "logit_kl": logit_kl,
             "config": {
-                "model": args.model,
+                # a --layers override is a debug run, never the named
+                # config — make the JSON say so
+                "model": args.model + (f"-layers{args.layers}"
+                                       if args.layers else ""),
                 "quant": f"w{args.bits}a16-g{args.group_size}-rtn"
                          + ("" if lm_head == "bf16" else f"-{lm_head}head"),
                 "global_batch": args.batch,
