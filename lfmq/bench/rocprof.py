@@ -23,6 +23,9 @@ def profile_command(cmd: str, out_dir: str = "gpurun_out/prof",
                     name: str = "lfmq", timeout: int = 600) -> Optional[str]:
     """Run `rocprofv3 --kernel-trace --stats` around cmd; returns the
     results DB path (or None on failure)."""
+    # rocprofv3 resolves a relative -d against ITS cwd (/tmp below, the
+    # TMPDIR workaround) — absolutize so the DB lands where we glob
+    out_dir = os.path.abspath(out_dir)
     os.makedirs(out_dir, exist_ok=True)
     env = dict(os.environ, TMPDIR="/tmp")
     full = ["rocprofv3", "--kernel-trace", "--stats", "-d", out_dir,
@@ -40,6 +43,7 @@ def pmc_command(cmd: str, counters: List[str],
                 out_dir: str = "gpurun_out/pmc", name: str = "lfmq_pmc",
                 timeout: int = 600) -> Optional[str]:
     """Counter collection in its OWN pass (no trace domains)."""
+    out_dir = os.path.abspath(out_dir)
     os.makedirs(out_dir, exist_ok=True)
     env = dict(os.environ, TMPDIR="/tmp")
     full = ["rocprofv3", "--pmc", ",".join(counters), "-d", out_dir,

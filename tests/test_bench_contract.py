@@ -93,3 +93,31 @@ def test_bench_torchrun_world48_gloo(world):
     assert r["n_gpus"] == world
     assert r["config"]["parallelism"] == f"tp{world}"
     assert r["value"] > 0
+
+
+def test_rocprof_wrapper_paths(tmp_path, monkeypatch):
+    """profile_command must absolutize -d: rocprofv3 resolves a relative
+    output dir against its own cwd (/tmp), not the caller's — the DB
+    would land in /tmp and the glob would miss it.  Stub the binary and
+    check the constructed command + returned DB path."""
+    import lfmq.bench.rocprof as rp
+
+    seen = {}
+
+    def fake_run(cmd, **kw):
+        seen["cmd"] = cmd
+        d = cmd[cmd.index("-d") + 1]
+        assert os.path.isabs(d), d
+        with open(os.path.join(d, "lfmq_results.db"), "w") as f:
+            f.write("x")
+        class R: pass
+        return R()
+
+    monkeypatch.setattr(rp.subprocess, "run", fake_run)
+    monkeypatch.chdir(tmp_path)
+    db = rp.profile_command("python bench.py --steps 2", out_dir="prof")
+    assert db is not None and os.path.exists(db)
+    assert seen["cmd"][0] == "rocprofv3" and "--kernel-trace" in seen["cmd"]
+    db = rp.pmc_command("python bench.py", ["SQ_WAVES"], out_dir="pmc")
+    assert db is not None and "--pmc" in seen["cmd"]
+    assert "--kernel-trace" not in seen["cmd"]  # counters ride alone
