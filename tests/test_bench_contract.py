@@ -108,7 +108,8 @@ def test_rocprof_wrapper_paths(tmp_path, monkeypatch):
         seen["cmd"] = cmd
         d = cmd[cmd.index("-d") + 1]
         assert os.path.isabs(d), d
-        with open(os.path.join(d, "lfmq_results.db"), "w") as f:
+        name = cmd[cmd.index("-o") + 1]
+        with open(os.path.join(d, f"{name}_results.db"), "w") as f:
             f.write("x")
         class R: pass
         return R()
