@@ -122,3 +122,35 @@ def test_rocprof_wrapper_paths(tmp_path, monkeypatch):
     db = rp.pmc_command("python bench.py", ["SQ_WAVES"], out_dir="pmc")
     assert db is not None and "--pmc" in seen["cmd"]
     assert "--kernel-trace" not in seen["cmd"]  # counters ride alone
+
+
+def test_rocprof_summarize_db(tmp_path):
+    """summarize_db against a synthetic rocpd-schema DB: steady-state
+    tail filter, per-kernel aggregation, heaviest-first order."""
+    import sqlite3
+
+    from lfmq.bench.rocprof import summarize_db
+
+    db_path = str(tmp_path / "fake_results.db")
+    db = sqlite3.connect(db_path)
+    cur = db.cursor()
+    cur.execute("CREATE TABLE rocpd_kernel_dispatch_abc "
+                "(kernel_id INT, start INT, end INT)")
+    cur.execute("CREATE TABLE rocpd_info_kernel_symbol_abc "
+                "(id INT, display_name TEXT)")
+    cur.execute("INSERT INTO rocpd_info_kernel_symbol_abc VALUES (1, 'gemv')")
+    cur.execute("INSERT INTO rocpd_info_kernel_symbol_abc VALUES (2, 'attn')")
+    # warmup dispatch in the first half (must be cut by tail_frac=0.5)
+    cur.execute("INSERT INTO rocpd_kernel_dispatch_abc VALUES (1, 0, 100000)")
+    # steady state: gemv heavier than attn
+    for t in range(600000, 1000000, 100000):
+        cur.execute("INSERT INTO rocpd_kernel_dispatch_abc VALUES "
+                    f"(1, {t}, {t + 50000})")
+        cur.execute("INSERT INTO rocpd_kernel_dispatch_abc VALUES "
+                    f"(2, {t}, {t + 10000})")
+    db.commit()
+    db.close()
+    rows = summarize_db(db_path)
+    assert [r["kernel"] for r in rows] == ["gemv", "attn"]
+    assert rows[0]["calls"] == 4  # the warmup dispatch was excluded
+    assert rows[0]["total_ms"] > rows[1]["total_ms"]
