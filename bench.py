@@ -22,7 +22,7 @@ import torch
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=256)
+    ap.add_argument("--steps", type=int, default=512)
     ap.add_argument("--warmup", type=int, default=16)
     ap.add_argument("--model", default="llama3-8b")
     ap.add_argument("--batch", type=int, default=1)
