@@ -122,6 +122,7 @@ def main():
             },
         }
         print(json.dumps(out))
+    lfmq_dist.shutdown()  # a live gloo/RCCL pg at exit aborts intermittently
 
 
 if __name__ == "__main__":

@@ -90,3 +90,13 @@ def all_to_all_single(out: torch.Tensor, inp: torch.Tensor,
 def barrier() -> None:
     if get_world_size() > 1:
         dist.barrier()
+
+
+def shutdown() -> None:
+    """Tear the process group down explicitly.  Exiting the interpreter
+    with a live gloo world aborts intermittently (SIGABRT in the gloo
+    connection teardown at high rank counts) — every entry point that
+    calls init_dist() should call this on the way out."""
+    if dist.is_initialized():
+        dist.barrier()
+        dist.destroy_process_group()
