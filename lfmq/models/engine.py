@@ -52,10 +52,12 @@ def engine_supported(model, batch: int) -> bool:
     from .llama import LlamaForCausalLM
     from .qlinear import QLinear
     # OPT-IN this round: the engine is correct (bit-deterministic, parity
-    # within launched-variant noise — tests/test_gpu_engine.py) but at
-    # 183 tok/s it does not yet beat the launched path's 519; default
-    # stays on the launched kernels until the priced perf work lands
-    # (docs/persistent_decode_design.md status section).
+    # within launched-variant noise — tests/test_gpu_engine.py) but its
+    # last full-8B measurement (183 tok/s, single-record dots) trails the
+    # launched path's 522; the record-pair ILP rework (0.849 -> 0.481
+    # us/record in the ubench) landed after that measurement and projects
+    # ~250-300 — still short, so the default stays on the launched
+    # kernels (docs/persistent_decode_design.md status section).
     if os.environ.get("LFMQ_ENGINE", "0") != "1":
         return False
     if not isinstance(model, LlamaForCausalLM) or batch != 1:
