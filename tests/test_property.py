@@ -42,3 +42,30 @@ def test_group_edge_rejection_property(k, g):
     else:
         with pytest.raises(ValueError):
             ref.quant_groupwise(torch.randn(4, k), 4, g, False)
+
+
+@settings(max_examples=30, deadline=None)
+@given(n=st.integers(1, 16), G=st.integers(1, 20),
+       bits=st.sampled_from([4, 8]))
+def test_pack_zeros_padding_property(n, G, bits):
+    """pack_zeros pads G up to a whole uint32 word; the roundtrip must be
+    exact for every non-multiple G (the checkpoint stores zeros packed)."""
+    z = torch.randint(0, 1 << bits, (n, G), dtype=torch.int32)
+    assert torch.equal(ref.unpack_zeros(ref.pack_zeros(z, bits), bits, G), z)
+
+
+@settings(max_examples=25, deadline=None)
+@given(n=st.integers(1, 8), groups=st.integers(1, 4),
+       g=st.sampled_from([32, 64]), bits=st.sampled_from([4, 8]))
+def test_sym_zero_is_midpoint_property(n, groups, g, bits):
+    """Symmetric mode: the zero code is the fixed range midpoint for every
+    group (the fused kernels rely on it folding like any other zero), and
+    a symmetric tensor dequantizes with a sign-symmetric error profile."""
+    w = torch.randn(n, groups * g)
+    _, _, z = ref.quant_groupwise(w, bits, g, sym=True)
+    assert (z == (1 << bits) // 2).all()
+    codes, s, z = ref.quant_groupwise(torch.cat([w, -w], 0), bits, g, True)
+    dq = ref.dequant_groupwise(codes, s, z, g)
+    err = (torch.cat([w, -w], 0) - dq)
+    # clamping aside, err(w) == -err(-w): same scale, mirrored codes
+    assert (err[: n * 1] + err[n:]).abs().max() <= s.max() + 1e-5
