@@ -313,3 +313,30 @@ def test_tied_embeddings_tp_gloo(tmp_path):
         ref_logits.float().abs().max()
     assert rel < 0.05, f"tied TP logits rel err {float(rel)}"
     assert torch.equal(out["gen"], ref_gen)
+
+
+def test_shard_llama_8b_widths_in_memory():
+    """Shard the REAL llama3-8b layer widths (hidden 4096, inter 14336,
+    32q/8kv heads, vocab 128256, g=128, W8 head — exactly what the
+    driver's scale run builds) at world 8, the largest scale point
+    (single layer so it stays CI-light).  Catches 8B-specific
+    divisibility/shape bugs without needing 8 processes."""
+    from lfmq.bench.decode import build_quantized_llama
+    from lfmq.dist.tp import shard_llama
+    world = 8
+    qm, cfg, _ = build_quantized_llama(
+        "llama3-8b", "rtn", 4, 128, "cpu", seed=0, n_layers=1,
+        quant_lm_head="w8")
+    rank = world - 1  # the last rank has the edge-case offsets
+    shard_llama(qm, rank, world)
+    at = qm.layers[0].self_attn
+    assert at.qkv_proj.out_features == (32 // world + 2 * (8 // world)) * 128
+    assert at.o_proj.in_features == 4096 // world
+    mlp = qm.layers[0].mlp
+    assert mlp.gate_up_proj.out_features == 2 * 14336 // world
+    assert mlp.down_proj.in_features == 14336 // world
+    assert mlp.down_proj.group_size == 128  # group survives the k-shard
+    assert qm.lm_head.out_features == 128256 // world
+    # forward shapes hold at world 1 semantics (no pg: collectives no-op)
+    import lfmq.dist as d
+    assert d.get_world_size() == 1
