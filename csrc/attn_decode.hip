@@ -325,12 +325,16 @@ __global__ __launch_bounds__(TPB) void attn_decode_fused(
   }
   if (tickets == nullptr) return;
 
-  // arrival ticket: one lane drains its stores, adds, and broadcasts
-  // whether this workgroup arrived last
+  // arrival ticket.  vmcnt is a PER-WAVE counter and the partial stores
+  // above were issued by every wave of the workgroup: each wave must
+  // drain its own stores, and the barrier must order those drains before
+  // lane 0 takes the ticket — a lane-0-only waitcnt would let another
+  // CU's winner acquire while waves 1..3 still had partials in flight.
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
   __shared__ int last_sh;
   const int used = (T + FCH - 1) / FCH;
   if (threadIdx.x == 0) {
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     const unsigned old = __hip_atomic_fetch_add(
         (__attribute__((address_space(1))) unsigned*)&tickets[bh], 1u,
         __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
