@@ -219,3 +219,28 @@ def test_awq_clip_e2e():
     qm = lfmq.quantize(lfmq.build_model(cfg, seed=15), qcfg, calib=calib)
     ids = torch.randint(0, cfg.vocab_size, (1, 8))
     assert torch.isfinite(qm(ids, qm.make_cache(1, 16)).float()).all()
+
+
+def test_degenerate_weights_quantize_finite():
+    """All-zero and all-constant weight rows must quantize (scale==0
+    guard) and run finite; GPTQ must survive dead input columns (zeroed
+    Hessian diagonal gets the dead-column patch)."""
+    import lfmq
+    from lfmq.models.llama import LlamaConfig
+    cfg = LlamaConfig.tiny()
+    m = lfmq.build_model(cfg, seed=0)
+    with torch.no_grad():
+        m.layers[0].self_attn.qkv_proj.weight.zero_()
+        m.layers[1].mlp.down_proj.weight.fill_(0.25)
+    qm = lfmq.quantize(m, lfmq.QuantConfig(scheme="rtn", bits=4,
+                                           group_size=32))
+    ids = torch.arange(8).reshape(1, 8)
+    assert torch.isfinite(qm(ids, qm.make_cache(1, 16)).float()).all()
+
+    m2 = lfmq.build_model(cfg, seed=1)
+    with torch.no_grad():
+        m2.embed_tokens.weight[:, :8].zero_()
+    qm2 = lfmq.quantize(m2, lfmq.QuantConfig(
+        scheme="gptq", bits=4, group_size=32, calib_n_seqs=1,
+        calib_seq_len=32))
+    assert torch.isfinite(qm2(ids, qm2.make_cache(1, 16)).float()).all()
